@@ -1,0 +1,344 @@
+/*
+ * hook.h — the cross-language ABI of the MI355X vGPU control library.
+ *
+ * This header is the single source of truth for every shared-memory region
+ * exchanged between the node agents (Python control plane, see
+ * vgpu_manager_amd/config/) and the in-container LD_PRELOAD library
+ * (libvgpu-control.so).  The Python side mirrors these layouts with ctypes
+ * and pins them with offset/size tests against `abi_probe` (library/test/
+ * abi_probe.c), so the contract is asserted from BOTH languages.
+ *
+ * Design parity notes (reference: coldzerofear/vgpu-manager):
+ *   - region set & semantics follow library/include/hook.h:64-626 of the
+ *     reference (vgpu.config resource region with per-device seqlock,
+ *     sm_util watcher region, vmem ledger, sm_node shared token bucket),
+ *     re-designed for HIP/amd-smi on gfx950 — not copied.
+ *   - frozen headers: the first 16 bytes of every mmap'd region never change
+ *     meaning across ABI versions (magic, version, size), so a reader can
+ *     always decide whether it understands a region.
+ *   - all cross-process mutable words are accessed with __atomic builtins
+ *     only.  `_Atomic` is deliberately banned: libatomic may implement it
+ *     with a per-process lock table which is meaningless across processes.
+ *     `volatile` is not a synchronization primitive and is not used as one.
+ *   - 128-byte cachelines: MI355X host CPUs prefetch in 128B; every
+ *     independently-written record is 128B-aligned to prevent false sharing.
+ */
+#ifndef VGPU_HOOK_H
+#define VGPU_HOOK_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ------------------------------------------------------------------ */
+/* Paths (container-visible).  Layout mirrors reference Appendix B.    */
+/* ------------------------------------------------------------------ */
+#define VGPU_MANAGER_DIR       "/etc/vgpu-manager"
+#define VGPU_CONFIG_DIR        VGPU_MANAGER_DIR "/config"
+#define VGPU_CONFIG_PATH       VGPU_CONFIG_DIR "/vgpu.config"
+#define VGPU_PIDS_PATH         VGPU_CONFIG_DIR "/pids.config"
+#define VGPU_WATCHER_DIR       VGPU_MANAGER_DIR "/watcher"
+#define VGPU_UTIL_PATH         VGPU_WATCHER_DIR "/sm_util.config"
+#define VGPU_HOST_PROC_DIR     VGPU_MANAGER_DIR "/.host_proc"
+#define VGPU_REGISTRY_SOCKET   VGPU_MANAGER_DIR "/registry/socket.sock"
+#define VGPU_DEVICE_CLIENT     VGPU_MANAGER_DIR "/registry/device-client"
+#define VGPU_LOCK_DIR          "/tmp/.vgpu_lock"
+#define VGPU_VMEM_DIR          "/tmp/.vmem_node"
+#define VGPU_VMEM_PATH         VGPU_VMEM_DIR "/vmem_node.config"
+#define VGPU_SM_NODE_DIR       "/tmp/.sm_node"
+#define VGPU_SM_NODE_PATH      VGPU_SM_NODE_DIR "/sm_node.config"
+#define VGPU_SM_NODE_LOCK      VGPU_SM_NODE_DIR "/sm_node.lock"
+
+/* ------------------------------------------------------------------ */
+/* Global constants                                                    */
+/* ------------------------------------------------------------------ */
+#define MAX_DEVICE_COUNT   16      /* devices per container            */
+#define MAX_DEVICE_PIDS    1024    /* pids tracked per container       */
+#define MAX_UTIL_PROCS     64      /* per-device processes in sm_util  */
+#define MAX_VMEM_RECORDS   4096    /* ledger records per container     */
+#define UUID_LEN           48      /* NUL-padded GPU uuid string       */
+#define CACHELINE_SIZE     128
+
+/* token-bucket timing */
+#define TIME_TICK_MS       10      /* throttle sleep tick              */
+#define WATCHER_CYCLE_MS   100     /* utilization watcher base cadence */
+#define TOKEN_FACTOR       32      /* pool = threads/CU * CUs * FACTOR */
+#define SOFT_ADJUST_CYCLES 30      /* soft-limit elastic ramp interval */
+
+/* frozen header magics ("AMDV" family) */
+#define VGPU_CFG_MAGIC     0x31464356554D4441ULL  /* "AMDUVCF1"       */
+#define VGPU_UTIL_MAGIC    0x31464355554D4441ULL  /* "AMDUUCF1"       */
+#define VGPU_VMEM_MAGIC    0x31464D56554D4441ULL  /* "AMDUVMF1"       */
+#define VGPU_SMND_MAGIC    0x31444E53554D4441ULL  /* "AMDUSND1"       */
+#define VGPU_PIDS_MAGIC    0x31534449504D4441ULL  /* "AMDPIDS1"       */
+#define VGPU_ABI_VERSION   1
+
+/* ------------------------------------------------------------------ */
+/* Frozen region header — first 16 bytes of every region, immutable.   */
+/* ------------------------------------------------------------------ */
+typedef struct {
+    uint64_t magic;        /* region identity                           */
+    uint32_t abi_version;  /* bump on any layout change                 */
+    uint32_t region_size;  /* total bytes incl. header                  */
+} region_header_t;
+
+_Static_assert(sizeof(region_header_t) == 16, "frozen header is 16 bytes");
+
+/* ------------------------------------------------------------------ */
+/* vgpu.config — per-container resource quota region                   */
+/*                                                                     */
+/* Written by the device plugin / DRA driver (Python, mmap), read by   */
+/* the C library.  Each device_t carries its own seqlock so the node   */
+/* agent can mutate limits at runtime: writer bumps seq to odd, writes,*/
+/* bumps to even (release); reader retries until it observes the same  */
+/* even seq before and after the read (acquire).                       */
+/* ------------------------------------------------------------------ */
+
+/* device flags (device_t.flags) */
+#define DEV_FLAG_MEM_LIMIT    (1u << 0)  /* memory quota enforced       */
+#define DEV_FLAG_CORE_LIMIT   (1u << 1)  /* CU throttle enforced        */
+#define DEV_FLAG_OVERSOLD     (1u << 2)  /* spill to hipMallocManaged   */
+#define DEV_FLAG_SOFT_CORE    (1u << 3)  /* soft (elastic) core limit   */
+
+typedef struct {
+    uint32_t seq;             /* seqlock; even = stable                 */
+    uint32_t flags;           /* DEV_FLAG_*                             */
+    uint64_t total_memory;    /* quota bytes (the spoofed "total")      */
+    uint32_t core_limit;      /* 0-100, % of the GPU's CUs              */
+    uint32_t soft_core_limit; /* 0-100, burst ceiling when idle         */
+    int32_t  host_index;      /* host-side device index                 */
+    uint32_t _rsvd0;
+    char     uuid[UUID_LEN];  /* host GPU uuid, NUL padded              */
+    uint8_t  _pad[CACHELINE_SIZE - 80];
+} device_t;
+
+_Static_assert(sizeof(device_t) == CACHELINE_SIZE, "device_t is one cacheline");
+_Static_assert(offsetof(device_t, uuid) == 32, "device_t.uuid offset");
+
+/* compute policy (resource_data_t.compute_policy) */
+#define COMPUTE_POLICY_FIXED    0  /* hard core limit                    */
+#define COMPUTE_POLICY_BALANCE  1  /* soft limit; burst to soft ceiling  */
+#define COMPUTE_POLICY_NONE     2  /* no CU throttle                     */
+
+typedef struct {
+    region_header_t hdr;      /* VGPU_CFG_MAGIC                         */
+    char     pod_uid[64];
+    char     pod_name[128];
+    char     pod_namespace[128];
+    char     container_name[128];
+    int32_t  device_count;
+    uint32_t compute_policy;
+    uint32_t oversold;        /* container-level memory oversell switch */
+    uint32_t _rsvd0;
+    uint8_t  _pad[32];               /* pad 480 -> 512: devices aligned */
+    device_t devices[MAX_DEVICE_COUNT];
+} resource_data_t;
+
+_Static_assert(offsetof(resource_data_t, devices) == 512,
+               "devices[] cacheline aligned");
+_Static_assert(sizeof(resource_data_t) == 512 + 16 * CACHELINE_SIZE,
+               "resource_data_t frozen size");
+
+/* ------------------------------------------------------------------ */
+/* pids.config — container PID set                                     */
+/* Written by the registry server (client mode) or assembled by the    */
+/* library from the host-proc mount; read sorted for binary search.    */
+/* ------------------------------------------------------------------ */
+typedef struct {
+    region_header_t hdr;      /* VGPU_PIDS_MAGIC                        */
+    uint32_t pid_count;
+    uint32_t _rsvd0;
+    uint64_t updated_ns;      /* CLOCK_REALTIME of last write           */
+    int32_t  pids[MAX_DEVICE_PIDS];
+} pids_data_t;
+
+_Static_assert(sizeof(pids_data_t) == 32 + 4 * MAX_DEVICE_PIDS,
+               "pids_data_t frozen size");
+
+/* ------------------------------------------------------------------ */
+/* sm_util.config — host-side shared utilization watcher region        */
+/*                                                                     */
+/* One privileged host sampler (device-monitor) samples amd-smi per    */
+/* device and publishes here; containers mmap it read-only instead of  */
+/* each paying their own amd-smi query (the reference's                */
+/* SharedSMUtilizationWatcher).  Per-device seqlock, writer = host.    */
+/* ------------------------------------------------------------------ */
+typedef struct {
+    int32_t  pid;             /* host pid                               */
+    uint32_t gfx_busy_permille;   /* this process' gfx engine share     */
+    uint64_t vram_bytes;
+    uint32_t cu_occupancy;    /* CUs occupied (amd-smi)                 */
+    uint32_t _rsvd0;
+} util_proc_t;
+
+_Static_assert(sizeof(util_proc_t) == 24, "util_proc_t size");
+
+typedef struct {
+    uint32_t seq;                 /* seqlock                            */
+    uint32_t dev_busy_permille;   /* whole-GPU gfx activity             */
+    uint64_t sample_ns;           /* CLOCK_MONOTONIC of sample          */
+    uint32_t proc_count;
+    uint32_t _rsvd0;
+    uint64_t vram_used_bytes;     /* whole-GPU VRAM used                */
+    util_proc_t procs[MAX_UTIL_PROCS];
+    uint8_t  _pad[CACHELINE_SIZE - (32 + 24 * MAX_UTIL_PROCS) % CACHELINE_SIZE];
+} device_util_t;
+
+_Static_assert(sizeof(device_util_t) % CACHELINE_SIZE == 0,
+               "device_util_t cacheline multiple");
+
+typedef struct {
+    region_header_t hdr;      /* VGPU_UTIL_MAGIC                        */
+    uint32_t device_count;
+    uint32_t _rsvd0;
+    uint64_t heartbeat_ns;    /* sampler liveness                       */
+    uint8_t  _pad[CACHELINE_SIZE - 32];
+    device_util_t devices[MAX_DEVICE_COUNT];
+} util_region_t;
+
+_Static_assert(offsetof(util_region_t, devices) == CACHELINE_SIZE,
+               "util devices aligned");
+
+/* ------------------------------------------------------------------ */
+/* vmem_node.config — virtual-memory ledger                            */
+/*                                                                     */
+/* Records managed-memory (HMM) spill allocations past the HBM quota   */
+/* and async/graph-captured allocations whose size is only known at    */
+/* charge time.  Shared across all processes of the container; all     */
+/* mutations CAS-based; record slots are allocated by CAS on `state`.  */
+/* ------------------------------------------------------------------ */
+
+/* record kinds (vmem_record_t.kind) — parity with reference           */
+/* memory_node_t UVA_SYNC/UVA_ASYNC/CAPTURE/ASYNC_BRIDGE               */
+#define VMEM_KIND_SYNC          1  /* hipMallocManaged spill            */
+#define VMEM_KIND_ASYNC         2  /* hipMallocAsync charged            */
+#define VMEM_KIND_CAPTURE       3  /* graph-captured alloc              */
+#define VMEM_KIND_ASYNC_BRIDGE  4  /* async free pending completion     */
+
+#define VMEM_STATE_FREE   0
+#define VMEM_STATE_BUSY   1        /* transient during claim            */
+#define VMEM_STATE_LIVE   2
+
+typedef struct {
+    uint32_t state;           /* CAS: FREE -> BUSY -> LIVE -> FREE      */
+    uint32_t kind;
+    uint64_t dptr;            /* device pointer                         */
+    uint64_t size;
+    int32_t  pid;
+    int32_t  device;          /* container-local device index           */
+    uint64_t created_ns;
+} vmem_record_t;
+
+_Static_assert(sizeof(vmem_record_t) == 40, "vmem_record_t size");
+
+typedef struct {
+    uint64_t vmem_used;       /* managed bytes past cap (CAS)           */
+    uint64_t dev_hooked_used; /* device bytes allocated through hooks   */
+    uint8_t  _pad[CACHELINE_SIZE - 16];
+} vmem_dev_counter_t;
+
+_Static_assert(sizeof(vmem_dev_counter_t) == CACHELINE_SIZE,
+               "vmem_dev_counter_t one line");
+
+typedef struct {
+    region_header_t hdr;      /* VGPU_VMEM_MAGIC                        */
+    uint32_t record_cap;      /* MAX_VMEM_RECORDS at creation           */
+    uint32_t _rsvd0;
+    uint64_t created_ns;      /* region identity (rebuilt detection)    */
+    uint8_t  _pad[CACHELINE_SIZE - 32];
+    vmem_dev_counter_t counters[MAX_DEVICE_COUNT];
+    vmem_record_t records[MAX_VMEM_RECORDS];
+} vmem_region_t;
+
+_Static_assert(offsetof(vmem_region_t, counters) == CACHELINE_SIZE,
+               "vmem counters aligned");
+
+/* ------------------------------------------------------------------ */
+/* sm_node.config — container-wide shared CU token bucket              */
+/*                                                                     */
+/* One bucket per device shared by every process in the container      */
+/* (a per-process bucket would over-supply N-process containers).      */
+/* Token decrement on launch is a single CAS; refill is performed by   */
+/* an elected owner (election by CAS on refill_owner + staleness       */
+/* takeover on refill_ns age).  The published sample lets non-owner    */
+/* processes reuse the owner's utilization query.                      */
+/* ------------------------------------------------------------------ */
+typedef struct {
+    /* line 0: the bucket — the only word touched on the launch path   */
+    int64_t  tokens;              /* CAS; may go slightly negative      */
+    int64_t  pool_size;           /* g_total for this device            */
+    uint8_t  _pad0[CACHELINE_SIZE - 16];
+    /* line 1: refill election + controller state (owner-only writes)  */
+    int32_t  refill_owner_pid;
+    uint32_t controller_kind;     /* 1 delta, 2 aimd, 3 auto            */
+    uint64_t refill_ns;           /* CLOCK_MONOTONIC of last refill     */
+    int64_t  cur_share;           /* tokens granted per cycle           */
+    int32_t  aimd_cooldown;       /* cycles until next MD allowed       */
+    uint32_t exclusive_state;     /* auto-FSM: 0 shared, 1 exclusive    */
+    int32_t  debounce_count;      /* auto-FSM debounce counter          */
+    uint32_t soft_cycle;          /* soft-limit elastic ramp counter    */
+    uint8_t  _pad1[CACHELINE_SIZE - 40];
+    /* line 2: published utilization sample (owner writes, seqlock)     */
+    uint32_t sample_seq;
+    uint32_t util_permille;       /* container's gfx share of device    */
+    uint32_t dev_busy_permille;   /* whole-device busy                  */
+    uint32_t _rsvd0;
+    uint64_t sample_ns;
+    uint8_t  _pad2[CACHELINE_SIZE - 24];
+} sm_node_dev_t;
+
+_Static_assert(sizeof(sm_node_dev_t) == 3 * CACHELINE_SIZE,
+               "sm_node_dev_t three lines");
+
+typedef struct {
+    region_header_t hdr;      /* VGPU_SMND_MAGIC                        */
+    uint32_t device_count;
+    uint32_t _rsvd0;
+    uint64_t created_ns;      /* region identity                        */
+    uint8_t  _pad[CACHELINE_SIZE - 32];
+    sm_node_dev_t devices[MAX_DEVICE_COUNT];
+} sm_node_region_t;
+
+_Static_assert(offsetof(sm_node_region_t, devices) == CACHELINE_SIZE,
+               "sm_node devices aligned");
+
+/* ------------------------------------------------------------------ */
+/* dynamic_config_t — process-local env tunables (parsed once).        */
+/* Env names: see util.c.  Not shared memory; layout free to change.   */
+/* ------------------------------------------------------------------ */
+typedef struct {
+    int  controller;              /* 1 delta, 2 aimd, 3 auto            */
+    int  usage_threshold;         /* % busy considered "active"         */
+    int  aimd_md_divisor;         /* multiplicative-decrease divisor    */
+    int  aimd_eff_num;            /* efficiency buffer numerator (7/8)  */
+    int  aimd_eff_den;
+    int  aimd_ai_base_div;        /* additive-increase = pool/div       */
+    int  aimd_deadband_permille;  /* no-op band around target           */
+    int  aimd_md_cooldown;        /* cycles between MDs                 */
+    int  auto_debounce_cycles;
+    int  auto_ext_util_threshold; /* permille                           */
+    int  delta_ramp_floor_div;
+    int  shared_bucket;           /* use sm_node region                 */
+    int  mem_oversold;
+    int  mem_account_mode;        /* 0 ledger, 1 smi, 2 max(both)       */
+    int  uva_advise;              /* hipMemAdvise residency hints       */
+    int  gap_disable;             /* disable gap duty-cycle path        */
+    int  log_level;
+} dynamic_config_t;
+
+/* logging levels */
+#define LOG_FATAL 0
+#define LOG_ERROR 1
+#define LOG_WARN  2
+#define LOG_INFO  3
+#define LOG_DEBUG 4
+#define LOG_TRACE 5
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* VGPU_HOOK_H */

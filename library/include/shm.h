@@ -1,0 +1,74 @@
+/* shm.h — shared-memory region lifecycle + seqlock + lock helpers. */
+#ifndef VGPU_SHM_H
+#define VGPU_SHM_H
+
+#include "hook.h"
+#include <stdbool.h>
+#include <sys/types.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- logging (structured stderr, level via VGPU_LOGGER_LEVEL) ---- */
+void vgpu_log(int level, const char *fmt, ...)
+    __attribute__((format(printf, 2, 3)));
+int  vgpu_log_level(void);
+
+#define LOGGER(lvl, ...) do {                                          \
+        if ((lvl) <= vgpu_log_level()) vgpu_log((lvl), __VA_ARGS__);   \
+    } while (0)
+
+/* ---- monotonic / realtime ns ---- */
+uint64_t mono_ns(void);
+uint64_t real_ns(void);
+
+/* ---- region lifecycle ----
+ * vgpu_region_attach: open (creating if `create`) `path`, size it to
+ * `size`, mmap MAP_SHARED, and validate/initialize the frozen header.
+ * On magic/version/size mismatch: if `create`, the region is rebuilt
+ * under an exclusive flock on the file; otherwise attach fails.
+ * Returns mapped pointer or NULL.  `*out_created` reports whether this
+ * call initialized the region. */
+void *vgpu_region_attach(const char *path, size_t size, uint64_t magic,
+                         bool create, bool *out_created);
+int   vgpu_region_detach(void *ptr, size_t size);
+
+/* region identity: detect the mapped file having been replaced under us
+ * (e.g. rm -rf /tmp): compares the mapped inode with a fresh stat. */
+bool  vgpu_region_stale(const char *path, const void *ptr);
+
+/* ---- seqlock ----
+ * Writer: seq_write_begin (odd), mutate payload, seq_write_end (even).
+ * Reader: loop { s0 = seq_read_begin; copy; } until seq_read_valid.   */
+static inline uint32_t seq_load(const uint32_t *seq) {
+    return __atomic_load_n(seq, __ATOMIC_ACQUIRE);
+}
+static inline void seq_write_begin(uint32_t *seq) {
+    __atomic_store_n(seq, __atomic_load_n(seq, __ATOMIC_RELAXED) + 1,
+                     __ATOMIC_RELEASE);
+    __atomic_thread_fence(__ATOMIC_SEQ_CST);
+}
+static inline void seq_write_end(uint32_t *seq) {
+    __atomic_thread_fence(__ATOMIC_SEQ_CST);
+    __atomic_store_n(seq, __atomic_load_n(seq, __ATOMIC_RELAXED) + 1,
+                     __ATOMIC_RELEASE);
+}
+static inline bool seq_read_valid(const uint32_t *seq, uint32_t s0) {
+    __atomic_thread_fence(__ATOMIC_ACQUIRE);
+    return (s0 & 1u) == 0 && __atomic_load_n(seq, __ATOMIC_ACQUIRE) == s0;
+}
+
+/* ---- cross-process file locks (OFD fcntl with POSIX fallback) ---- */
+/* Returns an fd holding the lock; -1 on failure.  Byte-range [off,len). */
+int  vgpu_flock_acquire(const char *path, off_t off, off_t len, bool wait);
+void vgpu_flock_release(int fd);
+
+/* per-device allocation lock under VGPU_LOCK_DIR                      */
+int  lock_gpu_device(int host_index);      /* returns lock fd or -1    */
+void unlock_gpu_device(int fd);
+
+#ifdef __cplusplus
+}
+#endif
+#endif

@@ -1,0 +1,204 @@
+/* shm.c — shared-memory region lifecycle, logging, time, locks.
+ *
+ * Region discipline (parity with reference loader.c:1481-1900 attach
+ * paths, re-designed):
+ *  - every region file carries the 16-byte frozen header (hook.h);
+ *  - creation/rebuild happens under an exclusive flock on the file so
+ *    concurrent processes never observe a half-initialized region;
+ *  - a reader that does not own the region (create=false) refuses any
+ *    header mismatch instead of guessing;
+ *  - any shared-region failure must degrade to per-process behavior in
+ *    the caller — helpers here never abort the process.
+ */
+#define _GNU_SOURCE
+#include "shm.h"
+
+#include <errno.h>
+#include <fcntl.h>
+#include <stdarg.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+#include <sys/file.h>
+#include <sys/mman.h>
+#include <sys/stat.h>
+#include <time.h>
+#include <unistd.h>
+
+/* ---------------- logging ---------------- */
+
+static int g_log_level = -1;
+
+int vgpu_log_level(void) {
+    int lvl = __atomic_load_n(&g_log_level, __ATOMIC_RELAXED);
+    if (lvl < 0) {
+        const char *e = getenv("VGPU_LOGGER_LEVEL");
+        if (!e) e = getenv("LOGGER_LEVEL");
+        lvl = e ? atoi(e) : LOG_WARN;
+        if (lvl < LOG_FATAL) lvl = LOG_FATAL;
+        if (lvl > LOG_TRACE) lvl = LOG_TRACE;
+        __atomic_store_n(&g_log_level, lvl, __ATOMIC_RELAXED);
+    }
+    return lvl;
+}
+
+void vgpu_log(int level, const char *fmt, ...) {
+    static const char *names[] = {"FATAL", "ERROR", "WARN",
+                                  "INFO",  "DEBUG", "TRACE"};
+    char buf[1024];
+    va_list ap;
+    va_start(ap, fmt);
+    vsnprintf(buf, sizeof(buf), fmt, ap);
+    va_end(ap);
+    fprintf(stderr, "[vgpu-control %s pid=%d] %s\n",
+            names[level >= 0 && level <= 5 ? level : 1], (int)getpid(), buf);
+}
+
+/* ---------------- time ---------------- */
+
+uint64_t mono_ns(void) {
+    struct timespec ts;
+    clock_gettime(CLOCK_MONOTONIC, &ts);
+    return (uint64_t)ts.tv_sec * 1000000000ull + (uint64_t)ts.tv_nsec;
+}
+
+uint64_t real_ns(void) {
+    struct timespec ts;
+    clock_gettime(CLOCK_REALTIME, &ts);
+    return (uint64_t)ts.tv_sec * 1000000000ull + (uint64_t)ts.tv_nsec;
+}
+
+/* ---------------- region lifecycle ---------------- */
+
+static int header_ok(const void *ptr, size_t size, uint64_t magic) {
+    const region_header_t *h = (const region_header_t *)ptr;
+    return h->magic == magic && h->abi_version == VGPU_ABI_VERSION &&
+           h->region_size == (uint32_t)size;
+}
+
+void *vgpu_region_attach(const char *path, size_t size, uint64_t magic,
+                         bool create, bool *out_created) {
+    if (out_created) *out_created = false;
+    int flags = create ? (O_RDWR | O_CREAT) : O_RDWR;
+    int fd = open(path, flags | O_CLOEXEC, 0666);
+    if (fd < 0) {
+        LOGGER(LOG_DEBUG, "region open %s failed: %s", path, strerror(errno));
+        return NULL;
+    }
+    struct stat st;
+    if (fstat(fd, &st) != 0) { close(fd); return NULL; }
+
+    if ((size_t)st.st_size != size) {
+        if (!create) {
+            LOGGER(LOG_WARN, "region %s size %zu != expected %zu",
+                   path, (size_t)st.st_size, size);
+            close(fd);
+            return NULL;
+        }
+        /* rebuild under exclusive lock */
+        if (flock(fd, LOCK_EX) != 0) { close(fd); return NULL; }
+        if (fstat(fd, &st) != 0 || (size_t)st.st_size != size) {
+            if (ftruncate(fd, 0) != 0 || ftruncate(fd, (off_t)size) != 0) {
+                flock(fd, LOCK_UN); close(fd); return NULL;
+            }
+        }
+        flock(fd, LOCK_UN);
+    }
+
+    void *ptr = mmap(NULL, size, PROT_READ | PROT_WRITE, MAP_SHARED, fd, 0);
+    if (ptr == MAP_FAILED) { close(fd); return NULL; }
+
+    if (!header_ok(ptr, size, magic)) {
+        if (!create) {
+            LOGGER(LOG_WARN, "region %s header mismatch (magic=%llx)", path,
+                   (unsigned long long)((const region_header_t *)ptr)->magic);
+            munmap(ptr, size);
+            close(fd);
+            return NULL;
+        }
+        if (flock(fd, LOCK_EX) == 0) {
+            if (!header_ok(ptr, size, magic)) {
+                /* zero payload, then publish the header last */
+                memset((char *)ptr + sizeof(region_header_t), 0,
+                       size - sizeof(region_header_t));
+                region_header_t h = {magic, VGPU_ABI_VERSION, (uint32_t)size};
+                memcpy(ptr, &h, sizeof(h));
+                __atomic_thread_fence(__ATOMIC_SEQ_CST);
+                if (out_created) *out_created = true;
+            }
+            flock(fd, LOCK_UN);
+        }
+    }
+    close(fd); /* mapping survives */
+    return ptr;
+}
+
+int vgpu_region_detach(void *ptr, size_t size) {
+    return ptr ? munmap(ptr, size) : 0;
+}
+
+bool vgpu_region_stale(const char *path, const void *ptr) {
+    /* The region was created through `path`; if the current file at
+     * `path` is a different inode (or gone), the mapping is orphaned
+     * (e.g. rm -rf /tmp) and shared semantics are lost. */
+    (void)ptr;
+    struct stat now;
+    if (stat(path, &now) != 0) return true;
+    /* compare against /proc/self/map_files? cheaper: keep a shadow stat */
+    static __thread struct stat cached;
+    static __thread const void *cached_ptr;
+    if (cached_ptr != ptr) {
+        cached = now;
+        cached_ptr = ptr;
+        return false;
+    }
+    return cached.st_ino != now.st_ino || cached.st_dev != now.st_dev;
+}
+
+/* ---------------- locks ---------------- */
+
+/* OFD locks are per-open-file-description: safe against the classic
+ * POSIX-lock close() drop and usable from multiple threads.  Fallback
+ * to POSIX record locks where OFD is unsupported. */
+static int ofd_fcntl(int fd, int cmd_ofd, int cmd_posix, struct flock *fl) {
+    int rc = fcntl(fd, cmd_ofd, fl);
+    if (rc == -1 && (errno == EINVAL || errno == ENOTSUP))
+        rc = fcntl(fd, cmd_posix, fl);
+    return rc;
+}
+
+int vgpu_flock_acquire(const char *path, off_t off, off_t len, bool wait) {
+    int fd = open(path, O_RDWR | O_CREAT | O_CLOEXEC, 0666);
+    if (fd < 0) return -1;
+    struct flock fl = {0};
+    fl.l_type = F_WRLCK;
+    fl.l_whence = SEEK_SET;
+    fl.l_start = off;
+    fl.l_len = len;
+    int cmd_ofd = wait ? F_OFD_SETLKW : F_OFD_SETLK;
+    int cmd_posix = wait ? F_SETLKW : F_SETLK;
+    if (ofd_fcntl(fd, cmd_ofd, cmd_posix, &fl) != 0) {
+        close(fd);
+        return -1;
+    }
+    return fd;
+}
+
+void vgpu_flock_release(int fd) {
+    if (fd >= 0) close(fd); /* closing drops both OFD and POSIX locks */
+}
+
+int lock_gpu_device(int host_index) {
+    char path[256];
+    snprintf(path, sizeof(path), VGPU_LOCK_DIR "/dev_%d.lock", host_index);
+    /* bounded backoff: the allocation path must not spin unboundedly  */
+    for (int attempt = 0; attempt < 3; attempt++) {
+        int fd = vgpu_flock_acquire(path, 0, 1, /*wait=*/attempt == 2);
+        if (fd >= 0) return fd;
+        struct timespec ts = {0, 2000000L << attempt}; /* 2,4 ms */
+        nanosleep(&ts, NULL);
+    }
+    return -1;
+}
+
+void unlock_gpu_device(int fd) { vgpu_flock_release(fd); }

@@ -1,0 +1,231 @@
+/* util.c — env parsing with PID-1 fallback, tunables, pid sets.
+ * Parity: reference library/src/util.c (behavioral, not copied).      */
+#define _GNU_SOURCE
+#include "util.h"
+#include "shm.h"
+
+#include <ctype.h>
+#include <fcntl.h>
+#include <pthread.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+#include <unistd.h>
+
+/* ---------------- env ---------------- */
+
+/* Search /proc/1/environ for name=value (NUL-separated).              */
+static const char *pid1_getenv(const char *name, char *buf, size_t buflen) {
+    int fd = open("/proc/1/environ", O_RDONLY | O_CLOEXEC);
+    if (fd < 0) return NULL;
+    static __thread char envbuf[65536];
+    ssize_t n = read(fd, envbuf, sizeof(envbuf) - 1);
+    close(fd);
+    if (n <= 0) return NULL;
+    envbuf[n] = '\0';
+    size_t namelen = strlen(name);
+    for (char *p = envbuf; p < envbuf + n;) {
+        size_t l = strlen(p);
+        if (l > namelen + 1 && strncmp(p, name, namelen) == 0 &&
+            p[namelen] == '=') {
+            snprintf(buf, buflen, "%s", p + namelen + 1);
+            return buf;
+        }
+        p += l + 1;
+    }
+    return NULL;
+}
+
+const char *vgpu_getenv(const char *name, char *buf, size_t buflen) {
+    const char *v = getenv(name);
+    if (v) {
+        snprintf(buf, buflen, "%s", v);
+        return buf;
+    }
+    return pid1_getenv(name, buf, buflen);
+}
+
+/* ---------------- size parsing ---------------- */
+
+long long vgpu_parse_size(const char *s) {
+    if (!s || !*s) return -1;
+    char *end = NULL;
+    double val = strtod(s, &end);
+    if (end == s || val < 0) return -1;
+    while (*end == ' ') end++;
+    long long mult = 1;
+    char c = (char)tolower((unsigned char)*end);
+    switch (c) {
+    case 'k': mult = 1024LL; break;
+    case 'm': mult = 1024LL * 1024; break;
+    case 'g': mult = 1024LL * 1024 * 1024; break;
+    case 't': mult = 1024LL * 1024 * 1024 * 1024; break;
+    case '\0': return (long long)val;
+    default: return -1;
+    }
+    /* accept k/ki/kb/kib spellings */
+    end++;
+    if (*end && tolower((unsigned char)*end) == 'i') end++;
+    if (*end && tolower((unsigned char)*end) == 'b') end++;
+    if (*end) return -1;
+    return (long long)(val * (double)mult);
+}
+
+/* ---------------- tunables ---------------- */
+
+static dynamic_config_t g_dyncfg;
+static pthread_once_t g_dyncfg_once = PTHREAD_ONCE_INIT;
+
+static int env_int(const char *name, int def) {
+    char buf[64];
+    const char *v = vgpu_getenv(name, buf, sizeof(buf));
+    if (!v || !*v) return def;
+    return atoi(v);
+}
+
+static int env_bool(const char *name, int def) {
+    char buf[64];
+    const char *v = vgpu_getenv(name, buf, sizeof(buf));
+    if (!v || !*v) return def;
+    return (strcmp(v, "1") == 0 || strcasecmp(v, "true") == 0 ||
+            strcasecmp(v, "on") == 0)
+               ? 1
+               : 0;
+}
+
+static void dyncfg_init(void) {
+    dynamic_config_t *c = &g_dyncfg;
+    char buf[64];
+    const char *v;
+
+    c->controller = 3; /* auto */
+    v = vgpu_getenv("VGPU_CU_CONTROLLER", buf, sizeof(buf));
+    if (v) {
+        if (strcasecmp(v, "delta") == 0) c->controller = 1;
+        else if (strcasecmp(v, "aimd") == 0) c->controller = 2;
+        else if (strcasecmp(v, "auto") == 0) c->controller = 3;
+    }
+    c->usage_threshold = env_int("VGPU_CU_USAGE_THRESHOLD", 10);
+    c->aimd_md_divisor = env_int("VGPU_CU_AIMD_MD_DIVISOR", 3);
+    c->aimd_eff_num = 7;
+    c->aimd_eff_den = 8;
+    v = vgpu_getenv("VGPU_CU_AIMD_EFF_RATIO", buf, sizeof(buf));
+    if (v) {
+        int num, den;
+        if (sscanf(v, "%d/%d", &num, &den) == 2 && den > 0 && num > 0 &&
+            num <= den) {
+            c->aimd_eff_num = num;
+            c->aimd_eff_den = den;
+        }
+    }
+    c->aimd_ai_base_div = env_int("VGPU_CU_AIMD_AI_BASE_DIV", 64);
+    c->aimd_deadband_permille = env_int("VGPU_CU_AIMD_DEADBAND_PERMILLE", 20);
+    c->aimd_md_cooldown = env_int("VGPU_CU_AIMD_MD_COOLDOWN_CYCLES", 5);
+    c->auto_debounce_cycles = env_int("VGPU_CU_AUTO_DEBOUNCE_CYCLES", 3);
+    c->auto_ext_util_threshold =
+        env_int("VGPU_CU_AUTO_EXTERNAL_UTIL_THRESHOLD", 50);
+    c->delta_ramp_floor_div = env_int("VGPU_CU_DELTA_RAMP_FLOOR_DIVISOR", 10);
+    c->shared_bucket = env_bool("VGPU_CU_SHARED_BUCKET", 1);
+    c->mem_oversold = env_bool("VGPU_MEM_OVERSOLD", 0);
+    c->mem_account_mode = MEM_ACCOUNT_MAX;
+    v = vgpu_getenv("VGPU_MEM_ACCOUNT_MODE", buf, sizeof(buf));
+    if (v) {
+        if (strcasecmp(v, "ledger") == 0) c->mem_account_mode = MEM_ACCOUNT_LEDGER;
+        else if (strcasecmp(v, "smi") == 0) c->mem_account_mode = MEM_ACCOUNT_SMI;
+        else if (strcasecmp(v, "max") == 0) c->mem_account_mode = MEM_ACCOUNT_MAX;
+    }
+    c->uva_advise = env_bool("VGPU_MEM_UVA_ADVISE", 1);
+    c->gap_disable = env_bool("VGPU_GAP_DISABLE", 0);
+    c->log_level = vgpu_log_level();
+
+    LOGGER(LOG_INFO,
+           "dynconfig: controller=%d usage_thr=%d aimd(md=/%d eff=%d/%d "
+           "ai=/%d dead=%d cd=%d) auto(db=%d ext=%d) ramp=/%d shared=%d "
+           "oversold=%d acct=%d advise=%d gap_off=%d",
+           c->controller, c->usage_threshold, c->aimd_md_divisor,
+           c->aimd_eff_num, c->aimd_eff_den, c->aimd_ai_base_div,
+           c->aimd_deadband_permille, c->aimd_md_cooldown,
+           c->auto_debounce_cycles, c->auto_ext_util_threshold,
+           c->delta_ramp_floor_div, c->shared_bucket, c->mem_oversold,
+           c->mem_account_mode, c->uva_advise, c->gap_disable);
+}
+
+const dynamic_config_t *vgpu_dynconfig(void) {
+    pthread_once(&g_dyncfg_once, dyncfg_init);
+    return &g_dyncfg;
+}
+
+/* ---------------- pid sets ---------------- */
+
+static int cmp_i32(const void *a, const void *b) {
+    int32_t x = *(const int32_t *)a, y = *(const int32_t *)b;
+    return (x > y) - (x < y);
+}
+
+/* load from pids.config region if present and fresh */
+static int load_pids_from_config(pid_set_t *set) {
+    void *ptr = vgpu_region_attach(VGPU_PIDS_PATH, sizeof(pids_data_t),
+                                   VGPU_PIDS_MAGIC, false, NULL);
+    if (!ptr) return -1;
+    pids_data_t *pd = (pids_data_t *)ptr;
+    uint32_t n = __atomic_load_n(&pd->pid_count, __ATOMIC_ACQUIRE);
+    if (n > MAX_DEVICE_PIDS) n = MAX_DEVICE_PIDS;
+    memcpy(set->pids, pd->pids, n * sizeof(int32_t));
+    set->count = (int)n;
+    vgpu_region_detach(ptr, sizeof(pids_data_t));
+    return (int)n;
+}
+
+/* Fallback: walk our own cgroup's procs (container == cgroup).        */
+static int load_pids_from_cgroup(pid_set_t *set) {
+    /* cgroup v2: /sys/fs/cgroup<path>/cgroup.procs ; v1: pick cpuset  */
+    char path[640] = {0};
+    FILE *f = fopen("/proc/self/cgroup", "re");
+    if (!f) return -1;
+    char line[512];
+    while (fgets(line, sizeof(line), f)) {
+        /* v2 line: "0::/path" */
+        if (strncmp(line, "0::", 3) == 0) {
+            char *p = line + 3;
+            p[strcspn(p, "\n")] = 0;
+            snprintf(path, sizeof(path), "/sys/fs/cgroup%s/cgroup.procs", p);
+            break;
+        }
+    }
+    fclose(f);
+    if (!path[0]) return -1;
+    f = fopen(path, "re");
+    if (!f) return -1;
+    int n = 0;
+    while (n < MAX_DEVICE_PIDS && fgets(line, sizeof(line), f))
+        set->pids[n++] = (int32_t)atoi(line);
+    fclose(f);
+    set->count = n;
+    return n;
+}
+
+int vgpu_load_pid_set(pid_set_t *set) {
+    set->count = 0;
+    int n = load_pids_from_config(set);
+    if (n <= 0) n = load_pids_from_cgroup(set);
+    if (n < 0) {
+        /* last resort: just this process */
+        set->pids[0] = (int32_t)getpid();
+        set->count = 1;
+        n = 1;
+    }
+    qsort(set->pids, (size_t)set->count, sizeof(int32_t), cmp_i32);
+    set->loaded_ns = mono_ns();
+    return set->count;
+}
+
+bool vgpu_pid_set_contains(const pid_set_t *set, int32_t pid) {
+    int lo = 0, hi = set->count - 1;
+    while (lo <= hi) {
+        int mid = (lo + hi) / 2;
+        if (set->pids[mid] == pid) return true;
+        if (set->pids[mid] < pid) lo = mid + 1;
+        else hi = mid - 1;
+    }
+    return false;
+}
