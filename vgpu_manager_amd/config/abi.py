@@ -1,0 +1,212 @@
+"""ctypes mirrors of the C shared-memory ABI (library/include/hook.h).
+
+These classes ARE the node-agent side of the node<->container contract:
+the device plugin / DRA driver writes ``resource_data_t`` regions with
+them, the monitor reads ``vmem_region_t`` / ``sm_node_region_t`` for
+per-container metrics, and the host utilization sampler fills
+``util_region_t``.
+
+Layout is pinned from both languages: tests/test_abi_layout.py compiles
+library/test/abi_probe.c and asserts every offset/size against these
+definitions (the reference pins its Go mirrors against hook.h the same
+way — pkg/config/vgpu/vgpu_config_test.go).
+"""
+from __future__ import annotations
+
+import ctypes
+
+MAX_DEVICE_COUNT = 16
+MAX_DEVICE_PIDS = 1024
+MAX_UTIL_PROCS = 64
+MAX_VMEM_RECORDS = 4096
+UUID_LEN = 48
+CACHELINE_SIZE = 128
+
+VGPU_CFG_MAGIC = 0x31464356554D4441
+VGPU_UTIL_MAGIC = 0x31464355554D4441
+VGPU_VMEM_MAGIC = 0x31464D56554D4441
+VGPU_SMND_MAGIC = 0x31444E53554D4441
+VGPU_PIDS_MAGIC = 0x31534449504D4441
+VGPU_ABI_VERSION = 1
+
+# device_t.flags
+DEV_FLAG_MEM_LIMIT = 1 << 0
+DEV_FLAG_CORE_LIMIT = 1 << 1
+DEV_FLAG_OVERSOLD = 1 << 2
+DEV_FLAG_SOFT_CORE = 1 << 3
+
+# resource_data_t.compute_policy
+COMPUTE_POLICY_FIXED = 0
+COMPUTE_POLICY_BALANCE = 1
+COMPUTE_POLICY_NONE = 2
+
+# vmem record kinds / states
+VMEM_KIND_SYNC = 1
+VMEM_KIND_ASYNC = 2
+VMEM_KIND_CAPTURE = 3
+VMEM_KIND_ASYNC_BRIDGE = 4
+VMEM_STATE_FREE = 0
+VMEM_STATE_BUSY = 1
+VMEM_STATE_LIVE = 2
+
+
+class RegionHeader(ctypes.Structure):
+    _fields_ = [
+        ("magic", ctypes.c_uint64),
+        ("abi_version", ctypes.c_uint32),
+        ("region_size", ctypes.c_uint32),
+    ]
+
+
+class DeviceT(ctypes.Structure):
+    _fields_ = [
+        ("seq", ctypes.c_uint32),
+        ("flags", ctypes.c_uint32),
+        ("total_memory", ctypes.c_uint64),
+        ("core_limit", ctypes.c_uint32),
+        ("soft_core_limit", ctypes.c_uint32),
+        ("host_index", ctypes.c_int32),
+        ("_rsvd0", ctypes.c_uint32),
+        ("uuid", ctypes.c_char * UUID_LEN),
+        ("_pad", ctypes.c_uint8 * (CACHELINE_SIZE - 80)),
+    ]
+
+
+class ResourceDataT(ctypes.Structure):
+    _fields_ = [
+        ("hdr", RegionHeader),
+        ("pod_uid", ctypes.c_char * 64),
+        ("pod_name", ctypes.c_char * 128),
+        ("pod_namespace", ctypes.c_char * 128),
+        ("container_name", ctypes.c_char * 128),
+        ("device_count", ctypes.c_int32),
+        ("compute_policy", ctypes.c_uint32),
+        ("oversold", ctypes.c_uint32),
+        ("_rsvd0", ctypes.c_uint32),
+        ("_pad", ctypes.c_uint8 * 32),
+        ("devices", DeviceT * MAX_DEVICE_COUNT),
+    ]
+
+
+class PidsDataT(ctypes.Structure):
+    _fields_ = [
+        ("hdr", RegionHeader),
+        ("pid_count", ctypes.c_uint32),
+        ("_rsvd0", ctypes.c_uint32),
+        ("updated_ns", ctypes.c_uint64),
+        ("pids", ctypes.c_int32 * MAX_DEVICE_PIDS),
+    ]
+
+
+class UtilProcT(ctypes.Structure):
+    _fields_ = [
+        ("pid", ctypes.c_int32),
+        ("gfx_busy_permille", ctypes.c_uint32),
+        ("vram_bytes", ctypes.c_uint64),
+        ("cu_occupancy", ctypes.c_uint32),
+        ("_rsvd0", ctypes.c_uint32),
+    ]
+
+
+_UTIL_DEV_PAY = 32 + 24 * MAX_UTIL_PROCS
+_UTIL_DEV_PAD = CACHELINE_SIZE - (_UTIL_DEV_PAY % CACHELINE_SIZE)
+
+
+class DeviceUtilT(ctypes.Structure):
+    _fields_ = [
+        ("seq", ctypes.c_uint32),
+        ("dev_busy_permille", ctypes.c_uint32),
+        ("sample_ns", ctypes.c_uint64),
+        ("proc_count", ctypes.c_uint32),
+        ("_rsvd0", ctypes.c_uint32),
+        ("vram_used_bytes", ctypes.c_uint64),
+        ("procs", UtilProcT * MAX_UTIL_PROCS),
+        ("_pad", ctypes.c_uint8 * _UTIL_DEV_PAD),
+    ]
+
+
+class UtilRegionT(ctypes.Structure):
+    _fields_ = [
+        ("hdr", RegionHeader),
+        ("device_count", ctypes.c_uint32),
+        ("_rsvd0", ctypes.c_uint32),
+        ("heartbeat_ns", ctypes.c_uint64),
+        ("_pad", ctypes.c_uint8 * (CACHELINE_SIZE - 32)),
+        ("devices", DeviceUtilT * MAX_DEVICE_COUNT),
+    ]
+
+
+class VmemRecordT(ctypes.Structure):
+    _fields_ = [
+        ("state", ctypes.c_uint32),
+        ("kind", ctypes.c_uint32),
+        ("dptr", ctypes.c_uint64),
+        ("size", ctypes.c_uint64),
+        ("pid", ctypes.c_int32),
+        ("device", ctypes.c_int32),
+        ("created_ns", ctypes.c_uint64),
+    ]
+
+
+class VmemDevCounterT(ctypes.Structure):
+    _fields_ = [
+        ("vmem_used", ctypes.c_uint64),
+        ("dev_hooked_used", ctypes.c_uint64),
+        ("_pad", ctypes.c_uint8 * (CACHELINE_SIZE - 16)),
+    ]
+
+
+class VmemRegionT(ctypes.Structure):
+    _fields_ = [
+        ("hdr", RegionHeader),
+        ("record_cap", ctypes.c_uint32),
+        ("_rsvd0", ctypes.c_uint32),
+        ("created_ns", ctypes.c_uint64),
+        ("_pad", ctypes.c_uint8 * (CACHELINE_SIZE - 32)),
+        ("counters", VmemDevCounterT * MAX_DEVICE_COUNT),
+        ("records", VmemRecordT * MAX_VMEM_RECORDS),
+    ]
+
+
+class SmNodeDevT(ctypes.Structure):
+    _fields_ = [
+        ("tokens", ctypes.c_int64),
+        ("pool_size", ctypes.c_int64),
+        ("_pad0", ctypes.c_uint8 * (CACHELINE_SIZE - 16)),
+        ("refill_owner_pid", ctypes.c_int32),
+        ("controller_kind", ctypes.c_uint32),
+        ("refill_ns", ctypes.c_uint64),
+        ("cur_share", ctypes.c_int64),
+        ("aimd_cooldown", ctypes.c_int32),
+        ("exclusive_state", ctypes.c_uint32),
+        ("debounce_count", ctypes.c_int32),
+        ("soft_cycle", ctypes.c_uint32),
+        ("_pad1", ctypes.c_uint8 * (CACHELINE_SIZE - 40)),
+        ("sample_seq", ctypes.c_uint32),
+        ("util_permille", ctypes.c_uint32),
+        ("dev_busy_permille", ctypes.c_uint32),
+        ("_rsvd0", ctypes.c_uint32),
+        ("sample_ns", ctypes.c_uint64),
+        ("_pad2", ctypes.c_uint8 * (CACHELINE_SIZE - 24)),
+    ]
+
+
+class SmNodeRegionT(ctypes.Structure):
+    _fields_ = [
+        ("hdr", RegionHeader),
+        ("device_count", ctypes.c_uint32),
+        ("_rsvd0", ctypes.c_uint32),
+        ("created_ns", ctypes.c_uint64),
+        ("_pad", ctypes.c_uint8 * (CACHELINE_SIZE - 32)),
+        ("devices", SmNodeDevT * MAX_DEVICE_COUNT),
+    ]
+
+
+# quick self-checks (mirror of the C _Static_asserts)
+assert ctypes.sizeof(RegionHeader) == 16
+assert ctypes.sizeof(DeviceT) == CACHELINE_SIZE
+assert ResourceDataT.devices.offset == 512
+assert ctypes.sizeof(ResourceDataT) == 512 + 16 * CACHELINE_SIZE
+assert ctypes.sizeof(PidsDataT) == 32 + 4 * MAX_DEVICE_PIDS
+assert ctypes.sizeof(VmemRecordT) == 40
+assert ctypes.sizeof(SmNodeDevT) == 3 * CACHELINE_SIZE

@@ -1,0 +1,337 @@
+"""Region writers/readers — the node-agent side of the mmap ABI.
+
+Responsibilities (parity with reference pkg/config/{vgpu,watcher,vmem}):
+  * ``VgpuConfigWriter``   — create/update a container's vgpu.config,
+    including runtime limit mutation under the per-device seqlock.
+  * ``PidsWriter``         — publish the container PID set (registry).
+  * ``UtilRegionWriter``   — host utilization sampler publishes per-device
+    samples under the per-device seqlock.
+  * ``VmemRegionReader`` / ``SmNodeReader`` — monitor-side readers.
+
+All writers publish the frozen header LAST on creation so a concurrent
+C-side attach never sees a half-initialized region.
+"""
+from __future__ import annotations
+
+import ctypes
+import mmap
+import os
+import time
+from dataclasses import dataclass, field
+from typing import Iterable, Optional
+
+from .abi import (
+    CACHELINE_SIZE,
+    COMPUTE_POLICY_BALANCE,
+    COMPUTE_POLICY_FIXED,
+    COMPUTE_POLICY_NONE,
+    DEV_FLAG_CORE_LIMIT,
+    DEV_FLAG_MEM_LIMIT,
+    DEV_FLAG_OVERSOLD,
+    DEV_FLAG_SOFT_CORE,
+    MAX_DEVICE_COUNT,
+    MAX_DEVICE_PIDS,
+    MAX_UTIL_PROCS,
+    PidsDataT,
+    RegionHeader,
+    ResourceDataT,
+    SmNodeRegionT,
+    UtilRegionT,
+    VGPU_ABI_VERSION,
+    VGPU_CFG_MAGIC,
+    VGPU_PIDS_MAGIC,
+    VGPU_SMND_MAGIC,
+    VGPU_UTIL_MAGIC,
+    VGPU_VMEM_MAGIC,
+    VmemRegionT,
+)
+
+COMPUTE_POLICY_NAMES = {
+    "fixed": COMPUTE_POLICY_FIXED,
+    "balance": COMPUTE_POLICY_BALANCE,
+    "none": COMPUTE_POLICY_NONE,
+}
+
+
+@dataclass
+class DeviceLimit:
+    """One device's quota as decided by the allocator."""
+
+    uuid: str
+    host_index: int
+    memory_bytes: int = 0          # 0 = unlimited
+    core_limit: int = 0            # 0 = unlimited; else 1-100
+    soft_core_limit: int = 0
+    oversold: bool = False
+
+    def flags(self) -> int:
+        f = 0
+        if self.memory_bytes > 0:
+            f |= DEV_FLAG_MEM_LIMIT
+        if self.core_limit > 0:
+            f |= DEV_FLAG_CORE_LIMIT
+        if self.oversold:
+            f |= DEV_FLAG_OVERSOLD
+        if self.soft_core_limit > 0:
+            f |= DEV_FLAG_SOFT_CORE
+        return f
+
+
+class _MappedRegion:
+    """An mmap'd struct of ctypes type ``struct_cls`` over ``path``."""
+
+    def __init__(self, path: str, struct_cls, magic: int, create: bool):
+        self.path = path
+        self.struct_cls = struct_cls
+        self.magic = magic
+        size = ctypes.sizeof(struct_cls)
+        flags = os.O_RDWR | (os.O_CREAT if create else 0)
+        fd = os.open(path, flags, 0o666)
+        try:
+            st = os.fstat(fd)
+            if st.st_size != size:
+                if not create:
+                    raise ValueError(
+                        f"{path}: size {st.st_size} != expected {size}")
+                os.ftruncate(fd, 0)
+                os.ftruncate(fd, size)
+            self.mm = mmap.mmap(fd, size)
+        finally:
+            os.close(fd)
+        self.data = struct_cls.from_buffer(self.mm)
+        hdr: RegionHeader = self.data.hdr
+        if hdr.magic != magic or hdr.abi_version != VGPU_ABI_VERSION \
+                or hdr.region_size != size:
+            if not create:
+                raise ValueError(f"{path}: bad region header")
+            # zero payload, publish header last
+            ctypes.memset(ctypes.addressof(self.data), 0, size)
+            hdr.abi_version = VGPU_ABI_VERSION
+            hdr.region_size = size
+            hdr.magic = magic  # last word published
+
+    def close(self) -> None:
+        # drop the ctypes view before closing the mmap; if a caller still
+        # holds a sub-view the close is deferred to GC (mmap keeps the
+        # pages alive until every exported buffer is released)
+        import gc
+        if hasattr(self, "data"):
+            del self.data
+        gc.collect()
+        try:
+            self.mm.close()
+        except BufferError:
+            pass
+
+
+def _seq_write(dev_struct, mutate) -> None:
+    """Per-device seqlock write: odd -> mutate -> even."""
+    dev_struct.seq += 1  # odd: writer active
+    # CPython bytecode gives us no fence; mmap stores are ordered enough
+    # on x86 (TSO) for the seqlock protocol: the seq store is program-
+    # order before/after the payload stores.
+    mutate(dev_struct)
+    dev_struct.seq += 1  # even: stable
+
+
+class VgpuConfigWriter:
+    """Writes /etc/vgpu-manager/<pod>_<cont>/config/vgpu.config."""
+
+    def __init__(self, path: str):
+        os.makedirs(os.path.dirname(path), exist_ok=True)
+        self.region = _MappedRegion(path, ResourceDataT, VGPU_CFG_MAGIC,
+                                    create=True)
+
+    def write(self, *, pod_uid: str, pod_name: str, pod_namespace: str,
+              container_name: str, limits: Iterable[DeviceLimit],
+              compute_policy: str = "fixed", oversold: bool = False) -> None:
+        d = self.region.data
+        d.pod_uid = pod_uid.encode()[:63]
+        d.pod_name = pod_name.encode()[:127]
+        d.pod_namespace = pod_namespace.encode()[:127]
+        d.container_name = container_name.encode()[:127]
+        d.compute_policy = COMPUTE_POLICY_NAMES.get(
+            compute_policy, COMPUTE_POLICY_FIXED)
+        d.oversold = 1 if oversold else 0
+        limits = list(limits)
+        if len(limits) > MAX_DEVICE_COUNT:
+            raise ValueError("too many devices")
+        for i, lim in enumerate(limits):
+            dev = d.devices[i]
+
+            def fill(s, lim=lim):
+                s.flags = lim.flags()
+                s.total_memory = lim.memory_bytes
+                s.core_limit = lim.core_limit
+                s.soft_core_limit = lim.soft_core_limit
+                s.host_index = lim.host_index
+                s.uuid = lim.uuid.encode()[:47]
+
+            _seq_write(dev, fill)
+        d.device_count = len(limits)
+
+    def modify_device(self, index: int, *, memory_bytes: Optional[int] = None,
+                      core_limit: Optional[int] = None,
+                      soft_core_limit: Optional[int] = None) -> None:
+        """Runtime quota mutation — the C side observes it via seqlock."""
+        d = self.region.data
+        if not 0 <= index < d.device_count:
+            raise IndexError(index)
+        dev = d.devices[index]
+
+        def mutate(s):
+            if memory_bytes is not None:
+                s.total_memory = memory_bytes
+                if memory_bytes > 0:
+                    s.flags |= DEV_FLAG_MEM_LIMIT
+                else:
+                    s.flags &= ~DEV_FLAG_MEM_LIMIT
+            if core_limit is not None:
+                s.core_limit = core_limit
+                if core_limit > 0:
+                    s.flags |= DEV_FLAG_CORE_LIMIT
+                else:
+                    s.flags &= ~DEV_FLAG_CORE_LIMIT
+            if soft_core_limit is not None:
+                s.soft_core_limit = soft_core_limit
+                if soft_core_limit > 0:
+                    s.flags |= DEV_FLAG_SOFT_CORE
+                else:
+                    s.flags &= ~DEV_FLAG_SOFT_CORE
+
+        _seq_write(dev, mutate)
+
+    def close(self) -> None:
+        self.region.close()
+
+
+class VgpuConfigReader:
+    """Monitor-side read of a container's vgpu.config."""
+
+    def __init__(self, path: str):
+        self.region = _MappedRegion(path, ResourceDataT, VGPU_CFG_MAGIC,
+                                    create=False)
+
+    def snapshot(self) -> dict:
+        d = self.region.data
+        devices = []
+        for i in range(d.device_count):
+            s = d.devices[i]
+            # seqlock read loop
+            for _ in range(1000):
+                s0 = s.seq
+                snap = dict(flags=s.flags, total_memory=s.total_memory,
+                            core_limit=s.core_limit,
+                            soft_core_limit=s.soft_core_limit,
+                            host_index=s.host_index,
+                            uuid=s.uuid.decode(errors="replace"))
+                if s.seq == s0 and s0 % 2 == 0:
+                    break
+            devices.append(snap)
+        return dict(
+            pod_uid=d.pod_uid.decode(errors="replace"),
+            pod_name=d.pod_name.decode(errors="replace"),
+            pod_namespace=d.pod_namespace.decode(errors="replace"),
+            container_name=d.container_name.decode(errors="replace"),
+            compute_policy=d.compute_policy,
+            oversold=bool(d.oversold),
+            devices=devices,
+        )
+
+    def close(self) -> None:
+        self.region.close()
+
+
+class PidsWriter:
+    """Registry server publishes the container PID set."""
+
+    def __init__(self, path: str):
+        os.makedirs(os.path.dirname(path), exist_ok=True)
+        self.region = _MappedRegion(path, PidsDataT, VGPU_PIDS_MAGIC,
+                                    create=True)
+
+    def write(self, pids: Iterable[int]) -> None:
+        d = self.region.data
+        pids = sorted(set(int(p) for p in pids))[:MAX_DEVICE_PIDS]
+        for i, p in enumerate(pids):
+            d.pids[i] = p
+        d.updated_ns = time.time_ns()
+        d.pid_count = len(pids)  # published last
+
+    def close(self) -> None:
+        self.region.close()
+
+
+class UtilRegionWriter:
+    """Host sampler publishes per-device utilization/process samples."""
+
+    def __init__(self, path: str, device_count: int):
+        os.makedirs(os.path.dirname(path), exist_ok=True)
+        self.region = _MappedRegion(path, UtilRegionT, VGPU_UTIL_MAGIC,
+                                    create=True)
+        self.region.data.device_count = device_count
+
+    def publish(self, dev_index: int, *, dev_busy_permille: int,
+                vram_used_bytes: int,
+                procs: Iterable[dict]) -> None:
+        d = self.region.data.devices[dev_index]
+
+        def fill(s):
+            s.dev_busy_permille = dev_busy_permille
+            s.vram_used_bytes = vram_used_bytes
+            s.sample_ns = time.monotonic_ns()
+            plist = list(procs)[:MAX_UTIL_PROCS]
+            for i, p in enumerate(plist):
+                s.procs[i].pid = p.get("pid", 0)
+                s.procs[i].gfx_busy_permille = p.get("gfx_busy_permille", 0)
+                s.procs[i].vram_bytes = p.get("vram_bytes", 0)
+                s.procs[i].cu_occupancy = p.get("cu_occupancy", 0)
+            s.proc_count = len(plist)
+
+        _seq_write(d, fill)
+        self.region.data.heartbeat_ns = time.monotonic_ns()
+
+    def close(self) -> None:
+        self.region.close()
+
+
+class VmemRegionReader:
+    """Monitor reads a container's managed-memory ledger."""
+
+    def __init__(self, path: str):
+        self.region = _MappedRegion(path, VmemRegionT, VGPU_VMEM_MAGIC,
+                                    create=False)
+
+    def device_usage(self) -> list:
+        d = self.region.data
+        return [
+            dict(vmem_used=d.counters[i].vmem_used,
+                 dev_hooked_used=d.counters[i].dev_hooked_used)
+            for i in range(MAX_DEVICE_COUNT)
+        ]
+
+    def close(self) -> None:
+        self.region.close()
+
+
+class SmNodeReader:
+    """Monitor reads a container's shared token-bucket state."""
+
+    def __init__(self, path: str):
+        self.region = _MappedRegion(path, SmNodeRegionT, VGPU_SMND_MAGIC,
+                                    create=False)
+
+    def snapshot(self) -> list:
+        d = self.region.data
+        out = []
+        for i in range(MAX_DEVICE_COUNT):
+            s = d.devices[i]
+            out.append(dict(tokens=s.tokens, pool_size=s.pool_size,
+                            refill_owner_pid=s.refill_owner_pid,
+                            cur_share=s.cur_share,
+                            util_permille=s.util_permille,
+                            dev_busy_permille=s.dev_busy_permille))
+        return out
+
+    def close(self) -> None:
+        self.region.close()
