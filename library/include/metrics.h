@@ -1,0 +1,34 @@
+/* metrics.h — in-library observability counters (logged, not exported;
+ * the device-monitor is the Prometheus face).                         */
+#ifndef VGPU_METRICS_H
+#define VGPU_METRICS_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+enum {
+    MET_OOM = 0,
+    MET_UVA_FALLBACK,
+    MET_RATE_SLEEP,
+    MET_GAP_SLEEP,
+    MET_WATCHER_MISS,
+    MET_AIMD_MD,
+    MET_AIMD_AI,
+    MET_EXCL_FLIP,
+    MET_REFILL_TAKEOVER,
+    MET_LOCK_WAIT,
+    MET_COUNT,
+};
+
+/* increments the counter; logs a structured line at power-of-two
+ * counts so hot counters do not flood stderr.                         */
+void metrics_inc(int which);
+uint64_t metrics_get(int which);
+
+#ifdef __cplusplus
+}
+#endif
+#endif

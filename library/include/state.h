@@ -1,0 +1,182 @@
+/* state.h — internal process state of the interception library.
+ * Not part of the cross-process ABI (hook.h is).                      */
+#ifndef VGPU_STATE_H
+#define VGPU_STATE_H
+
+#define __HIP_PLATFORM_AMD__ 1
+#include <hip/hip_runtime_api.h>
+
+#include "hook.h"
+#include "util.h"
+#include <pthread.h>
+#include <stdbool.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- real HIP entry table (filled from the dlopen'd libamdhip64) --- */
+typedef struct {
+    hipError_t (*hipMalloc)(void **, size_t);
+    hipError_t (*hipExtMallocWithFlags)(void **, size_t, unsigned int);
+    hipError_t (*hipMallocManaged)(void **, size_t, unsigned int);
+    hipError_t (*hipMallocAsync)(void **, size_t, hipStream_t);
+    hipError_t (*hipMallocFromPoolAsync)(void **, size_t, hipMemPool_t,
+                                         hipStream_t);
+    hipError_t (*hipMallocPitch)(void **, size_t *, size_t, size_t);
+    hipError_t (*hipMalloc3D)(hipPitchedPtr *, hipExtent);
+    hipError_t (*hipMallocArray)(hipArray_t *, const hipChannelFormatDesc *,
+                                 size_t, size_t, unsigned int);
+    hipError_t (*hipMalloc3DArray)(hipArray_t *, const hipChannelFormatDesc *,
+                                   hipExtent, unsigned int);
+    hipError_t (*hipFree)(void *);
+    hipError_t (*hipFreeAsync)(void *, hipStream_t);
+    hipError_t (*hipFreeArray)(hipArray_t);
+    hipError_t (*hipMemGetInfo)(size_t *, size_t *);
+    hipError_t (*hipDeviceTotalMem)(size_t *, hipDevice_t);
+    hipError_t (*hipGetDevicePropertiesR0600)(hipDeviceProp_tR0600 *, int);
+    hipError_t (*hipMemAdvise)(const void *, size_t, hipMemoryAdvise, int);
+    hipError_t (*hipMemPrefetchAsync)(const void *, size_t, int, hipStream_t);
+
+    hipError_t (*hipLaunchKernel)(const void *, dim3, dim3, void **, size_t,
+                                  hipStream_t);
+    hipError_t (*hipExtLaunchKernel)(const void *, dim3, dim3, void **,
+                                     size_t, hipStream_t, hipEvent_t,
+                                     hipEvent_t, int);
+    hipError_t (*hipModuleLaunchKernel)(hipFunction_t, unsigned int,
+                                        unsigned int, unsigned int,
+                                        unsigned int, unsigned int,
+                                        unsigned int, unsigned int,
+                                        hipStream_t, void **, void **);
+    hipError_t (*hipExtModuleLaunchKernel)(hipFunction_t, uint32_t, uint32_t,
+                                           uint32_t, uint32_t, uint32_t,
+                                           uint32_t, size_t, hipStream_t,
+                                           void **, void **, hipEvent_t,
+                                           hipEvent_t, uint32_t);
+    hipError_t (*hipLaunchCooperativeKernel)(const void *, dim3, dim3,
+                                             void **, unsigned int,
+                                             hipStream_t);
+    hipError_t (*hipModuleLaunchCooperativeKernel)(hipFunction_t, unsigned int,
+                                                   unsigned int, unsigned int,
+                                                   unsigned int, unsigned int,
+                                                   unsigned int, unsigned int,
+                                                   hipStream_t, void **);
+
+    hipError_t (*hipGraphLaunch)(hipGraphExec_t, hipStream_t);
+    hipError_t (*hipGraphInstantiate)(hipGraphExec_t *, hipGraph_t,
+                                      hipGraphNode_t *, char *, size_t);
+    hipError_t (*hipGraphInstantiateWithFlags)(hipGraphExec_t *, hipGraph_t,
+                                               unsigned long long);
+    hipError_t (*hipGraphExecDestroy)(hipGraphExec_t);
+    hipError_t (*hipGraphGetNodes)(hipGraph_t, hipGraphNode_t *, size_t *);
+    hipError_t (*hipGraphNodeGetType)(hipGraphNode_t, hipGraphNodeType *);
+    hipError_t (*hipGraphKernelNodeGetParams)(hipGraphNode_t,
+                                              hipKernelNodeParams *);
+
+    hipError_t (*hipGetDevice)(int *);
+    hipError_t (*hipSetDevice)(int);
+    hipError_t (*hipGetDeviceCount)(int *);
+    hipError_t (*hipDeviceGetAttribute)(int *, hipDeviceAttribute_t, int);
+    hipError_t (*hipDeviceGetUuid)(hipUUID *, hipDevice_t);
+    hipError_t (*hipEventCreateWithFlags)(hipEvent_t *, unsigned int);
+    hipError_t (*hipEventRecord)(hipEvent_t, hipStream_t);
+    hipError_t (*hipEventSynchronize)(hipEvent_t);
+    hipError_t (*hipEventElapsedTime)(float *, hipEvent_t, hipEvent_t);
+    hipError_t (*hipEventDestroy)(hipEvent_t);
+    hipError_t (*hipStreamIsCapturing)(hipStream_t,
+                                       hipStreamCaptureStatus *);
+    hipError_t (*hipGetProcAddress)(const char *, void **, int, uint64_t,
+                                    hipDriverProcAddressQueryResult *);
+} hip_real_t;
+
+extern hip_real_t real_hip;
+
+/* ---- per-device hot state (process-local) ---- */
+typedef struct {
+    /* padded: the launch path reads flags/tokens of its own device     */
+    int64_t tokens;          /* local bucket (used when no shared one)  */
+    int64_t pool;            /* total token pool                        */
+    int64_t cur_share;       /* controller output, tokens per cycle     */
+    uint64_t last_launch_ns; /* for the GAP idle-gap detector           */
+    uint64_t last_check_ns;
+    int32_t cu_count;
+    int32_t max_threads_per_cu;
+    int32_t aimd_cooldown;
+    uint32_t excl_state;     /* auto-FSM                                */
+    int32_t debounce;
+    uint32_t soft_cycle;
+    uint32_t throttled;      /* observability                           */
+    uint32_t gap_active;
+    hipEvent_t gap_start, gap_stop;
+    pthread_mutex_t gap_mu;
+    uint64_t prev_proc_gfx_ns;  /* per-container engine-time sample     */
+    uint64_t prev_sample_ns;
+    uint8_t _pad[24];
+} dev_hot_t;
+
+/* ---- global library state ---- */
+typedef struct {
+    int initialized;
+    int disabled;             /* DISABLE_VGPU_CONTROL                   */
+    resource_data_t *cfg;     /* shared mmap or private heap (env)      */
+    bool cfg_shared;
+    vmem_region_t *vmem;      /* shared ledger or private heap          */
+    bool vmem_shared;
+    sm_node_region_t *sm_node; /* NULL unless shared bucket enabled     */
+    util_region_t *util;      /* NULL unless external watcher mounted   */
+    pid_set_t pids;
+    int device_count;         /* visible HIP devices                    */
+    dev_hot_t dev[MAX_DEVICE_COUNT];
+} vgpu_state_t;
+
+extern vgpu_state_t g_state;
+
+/* loader API */
+void *vgpu_real_dlsym(void *handle, const char *name);
+int   vgpu_ensure_init(void);        /* load_necessary_data; 0 = ok     */
+void *vgpu_lookup_hook(const char *name);  /* hook table by name        */
+
+/* config accessors (seqlock snapshot) */
+void  vgpu_device_snapshot(int dev, device_t *out);
+static inline uint32_t vgpu_device_flags(int dev) {
+    /* single relaxed load — THE hot-path check                        */
+    extern vgpu_state_t g_state;
+    return __atomic_load_n(&g_state.cfg->devices[dev].flags,
+                           __ATOMIC_RELAXED);
+}
+
+/* allocation registry (process-local ptr -> {size, kind, dev}) */
+#define ALLOC_KIND_DEVICE  0
+#define ALLOC_KIND_MANAGED 1   /* oversold spill, in vmem ledger       */
+#define ALLOC_KIND_ASYNC   2
+int  alloc_registry_add(void *ptr, size_t size, int kind, int dev,
+                        int vmem_idx);
+/* returns true and fills outputs if found (and removes the entry)     */
+bool alloc_registry_remove(void *ptr, size_t *size, int *kind, int *dev,
+                           int *vmem_idx);
+uint64_t alloc_registry_total(int dev);
+
+/* vmem ledger ops */
+int  vmem_ledger_add(int dev, uint64_t dptr, uint64_t size, int kind);
+void vmem_ledger_remove(int idx);
+uint64_t vmem_ledger_used(int dev);
+void dev_hooked_add(int dev, int64_t delta);
+uint64_t dev_hooked_used(int dev);
+
+/* container used-bytes accounting (ledger/smi/max; hip_hook.c) */
+uint64_t vgpu_account_used(int dev);
+
+/* amd-smi sampling (watcher side; dlopens libamd_smi lazily) */
+bool smi_available(void);
+/* whole-device busy (permille) + container gfx engine ns + vram bytes */
+bool smi_sample_device(int host_index, uint32_t *busy_permille,
+                       uint64_t *container_gfx_ns,
+                       uint64_t *container_vram, const pid_set_t *pids);
+uint64_t smi_container_vram(int host_index, const pid_set_t *pids);
+
+/* hook implementations (exported) live in hip_hook.c / smi_hook.c     */
+
+#ifdef __cplusplus
+}
+#endif
+#endif

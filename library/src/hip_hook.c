@@ -1,0 +1,1111 @@
+/* hip_hook.c — the hot path: HBM quota + CU token-bucket throttle.
+ *
+ * Exported hip* symbols interpose libamdhip64 at link time (LD_PRELOAD)
+ * and at dlsym time (loader.c).  Design parity with the reference's
+ * cuda_hook.c (memory gate, token bucket, delta/aimd/auto controllers,
+ * GAP duty-cycle path, graph cost accounting) re-designed for HIP and
+ * gfx950: the token pool is CUs x maxWavesPerCU-threads x FACTOR, the
+ * utilization source is amd-smi process gfx engine time (or the shared
+ * host watcher region), and the launch fast path when limiting is OFF
+ * is a single relaxed atomic load of the device's flag word.
+ */
+#define _GNU_SOURCE
+#include "state.h"
+#include "shm.h"
+#include "metrics.h"
+
+#include <pthread.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+#include <time.h>
+#include <unistd.h>
+
+#define EXPORT __attribute__((visibility("default")))
+
+/* ------------------------------------------------------------------ */
+/* current-device tracking (TLS, maintained by hipSetDevice hook)      */
+/* ------------------------------------------------------------------ */
+static __thread int tls_device = 0;
+
+static inline int cur_dev(void) {
+    int d = tls_device;
+    if (d < 0 || d >= MAX_DEVICE_COUNT) d = 0;
+    return d;
+}
+
+/* device has a config entry? (container-local positional mapping)     */
+static inline int cfg_dev(int dev) {
+    return dev < g_state.cfg->device_count ? dev : -1;
+}
+
+/* ------------------------------------------------------------------ */
+/* utilization sampling + controllers + refill (watcher thread)        */
+/* ------------------------------------------------------------------ */
+
+static pthread_once_t g_watcher_once = PTHREAD_ONCE_INIT;
+
+static void dev_hot_init(int dev) {
+    dev_hot_t *h = &g_state.dev[dev];
+    if (h->pool) return;
+    int cus = 0, thr = 0;
+    if (real_hip.hipDeviceGetAttribute) {
+        real_hip.hipDeviceGetAttribute(&cus,
+            hipDeviceAttributeMultiprocessorCount, dev);
+        real_hip.hipDeviceGetAttribute(&thr,
+            hipDeviceAttributeMaxThreadsPerMultiProcessor, dev);
+    }
+    if (cus <= 0) cus = 256;   /* MI355X default */
+    if (thr <= 0) thr = 2048;
+    h->cu_count = cus;
+    h->max_threads_per_cu = thr;
+    h->pool = (int64_t)cus * thr * TOKEN_FACTOR;
+    h->cur_share = h->pool / 100; /* conservative start: 1% per cycle  */
+    __atomic_store_n(&h->tokens, h->cur_share, __ATOMIC_RELAXED);
+    if (g_state.sm_node) {
+        sm_node_dev_t *s = &g_state.sm_node->devices[dev];
+        int64_t z = 0;
+        __atomic_compare_exchange_n(&s->pool_size, &z, h->pool, false,
+                                    __ATOMIC_ACQ_REL, __ATOMIC_RELAXED);
+    }
+}
+
+/* --- utilization sources ------------------------------------------- */
+
+/* sample container + device utilization (permille).  Returns false if
+ * no source produced a sample this cycle. */
+static bool sample_util(int dev, uint32_t *cont_permille,
+                        uint32_t *dev_permille) {
+    dev_hot_t *h = &g_state.dev[dev];
+    device_t snap;
+    vgpu_device_snapshot(dev, &snap);
+    int host_index = snap.host_index >= 0 ? snap.host_index : dev;
+
+    /* 1. external host watcher region */
+    if (g_state.util) {
+        const device_util_t *u = &g_state.util->devices[host_index < MAX_DEVICE_COUNT ? host_index : dev];
+        for (int retry = 0; retry < 100; retry++) {
+            uint32_t s0 = seq_load(&u->seq);
+            if (s0 & 1u) continue;
+            uint32_t busy = u->dev_busy_permille;
+            uint64_t ts = u->sample_ns;
+            uint32_t cont = 0;
+            uint32_t n = u->proc_count;
+            if (n > MAX_UTIL_PROCS) n = MAX_UTIL_PROCS;
+            for (uint32_t i = 0; i < n; i++)
+                if (vgpu_pid_set_contains(&g_state.pids, u->procs[i].pid))
+                    cont += u->procs[i].gfx_busy_permille;
+            if (!seq_read_valid(&u->seq, s0)) continue;
+            if (mono_ns() - ts < 1000000000ull) { /* fresh within 1 s */
+                *cont_permille = cont > 1000 ? 1000 : cont;
+                *dev_permille = busy;
+                return true;
+            }
+            break;
+        }
+    }
+
+    /* 2. shared sm_node published sample from another process's owner */
+    if (g_state.sm_node) {
+        sm_node_dev_t *s = &g_state.sm_node->devices[dev];
+        int owner = __atomic_load_n(&s->refill_owner_pid, __ATOMIC_ACQUIRE);
+        if (owner != 0 && owner != getpid()) {
+            uint64_t ts = __atomic_load_n(&s->sample_ns, __ATOMIC_ACQUIRE);
+            if (mono_ns() - ts < 500000000ull) {
+                *cont_permille = s->util_permille;
+                *dev_permille = s->dev_busy_permille;
+                return true;
+            }
+        }
+    }
+
+    /* 3. own amd-smi query */
+    uint32_t busy = 0;
+    uint64_t gfx_ns = 0, vram = 0;
+    if (smi_available() &&
+        smi_sample_device(dev, &busy, &gfx_ns, &vram, &g_state.pids)) {
+        uint64_t now = mono_ns();
+        uint32_t cont = 0;
+        if (h->prev_sample_ns && gfx_ns >= h->prev_proc_gfx_ns) {
+            uint64_t dt = now - h->prev_sample_ns;
+            if (dt > 0)
+                cont = (uint32_t)((gfx_ns - h->prev_proc_gfx_ns) * 1000 / dt);
+        }
+        h->prev_proc_gfx_ns = gfx_ns;
+        h->prev_sample_ns = now;
+        *cont_permille = cont > 1000 ? 1000 : cont;
+        *dev_permille = busy;
+        /* publish for siblings when we own the shared bucket          */
+        if (g_state.sm_node) {
+            sm_node_dev_t *s = &g_state.sm_node->devices[dev];
+            if (__atomic_load_n(&s->refill_owner_pid, __ATOMIC_ACQUIRE) ==
+                getpid()) {
+                seq_write_begin(&s->sample_seq);
+                s->util_permille = *cont_permille;
+                s->dev_busy_permille = busy;
+                __atomic_store_n(&s->sample_ns, now, __ATOMIC_RELEASE);
+                seq_write_end(&s->sample_seq);
+            }
+        }
+        return true;
+    }
+    return false;
+}
+
+/* --- controllers ---------------------------------------------------
+ * All operate on `share` = tokens granted per WATCHER_CYCLE_MS, in
+ * [share_min, pool].  target/obs in permille of whole-device busy.    */
+
+static int64_t ctl_delta(const dynamic_config_t *c, dev_hot_t *h,
+                         int64_t share, uint32_t target, uint32_t obs) {
+    /* proportional: step toward the error, floor so convergence does
+     * not stall near the target (reference delta controller).         */
+    int64_t err = (int64_t)target - (int64_t)obs;    /* permille       */
+    int64_t step = h->pool * err / 20000;            /* gain 1/20      */
+    int64_t floor = h->pool / (c->delta_ramp_floor_div * 1000);
+    if (step > 0 && step < floor) step = floor;
+    if (step < 0 && -step < floor) step = -floor;
+    share += step;
+    return share;
+}
+
+static int64_t ctl_aimd(const dynamic_config_t *c, dev_hot_t *h,
+                        int64_t share, uint32_t target, uint32_t obs) {
+    /* AIMD with deadband + MD cooldown (reference sm_controller_aimd:
+     * naive AIMD sawtooths; the deadband and cooldown remove the
+     * steady-state oscillation that cost +1/3 walltime there).        */
+    uint32_t hi = target + target * c->aimd_deadband_permille / 1000;
+    uint32_t lo = (uint32_t)((uint64_t)target * c->aimd_eff_num /
+                             c->aimd_eff_den);
+    if (h->aimd_cooldown > 0) h->aimd_cooldown--;
+    if (obs > hi) {
+        if (h->aimd_cooldown == 0) {
+            share /= c->aimd_md_divisor;
+            h->aimd_cooldown = c->aimd_md_cooldown;
+            metrics_inc(MET_AIMD_MD);
+        }
+    } else if (obs < lo) {
+        share += h->pool / (c->aimd_ai_base_div * 100);
+        metrics_inc(MET_AIMD_AI);
+    }
+    return share;
+}
+
+/* control cycle for one device; returns tokens to grant this cycle    */
+static int64_t control_cycle(int dev) {
+    const dynamic_config_t *c = vgpu_dynconfig();
+    dev_hot_t *h = &g_state.dev[dev];
+    device_t snap;
+    vgpu_device_snapshot(dev, &snap);
+    if (!(snap.flags & DEV_FLAG_CORE_LIMIT) || snap.core_limit == 0 ||
+        snap.core_limit >= 100)
+        return h->pool; /* unlimited */
+
+    uint32_t target = snap.core_limit * 10;       /* % -> permille      */
+    uint32_t cont = 0, busy = 0;
+    bool have = sample_util(dev, &cont, &busy);
+    if (!have) {
+        /* no sample: steady state — grant the nominal share           */
+        metrics_inc(MET_WATCHER_MISS);
+        return h->pool * snap.core_limit / 100 / (1000 / WATCHER_CYCLE_MS);
+    }
+
+    /* soft-limit / auto exclusivity: when nobody else uses the GPU,
+     * allow bursting to the soft ceiling (policy balance).            */
+    uint32_t eff_target = target;
+    int soft_on = (snap.flags & DEV_FLAG_SOFT_CORE) &&
+                  snap.soft_core_limit > snap.core_limit &&
+                  g_state.cfg->compute_policy == COMPUTE_POLICY_BALANCE;
+    if (soft_on || c->controller == 3) {
+        uint32_t others = busy > cont ? busy - cont : 0;
+        int exclusive = others < (uint32_t)c->auto_ext_util_threshold;
+        if (exclusive != (int)h->excl_state) {
+            if (++h->debounce >= c->auto_debounce_cycles) {
+                h->excl_state = (uint32_t)exclusive;
+                h->debounce = 0;
+                metrics_inc(MET_EXCL_FLIP);
+            }
+        } else {
+            h->debounce = 0;
+        }
+        if (h->excl_state && soft_on) {
+            /* elastic ramp toward the soft ceiling                    */
+            if (++h->soft_cycle >= SOFT_ADJUST_CYCLES) {
+                h->soft_cycle = 0;
+            }
+            eff_target = snap.soft_core_limit * 10;
+        } else {
+            h->soft_cycle = 0;
+        }
+    }
+
+    /* anti-jitter bypass: long under-utilization with a hard limit
+     * means the app is not launch-bound — keep the bucket full so the
+     * throttle adds no latency (reference hard-limit bypass).         */
+    if (cont + 100 < eff_target && busy < eff_target) {
+        h->cur_share = h->pool * eff_target / 1000 /
+                       (1000 / WATCHER_CYCLE_MS);
+        return h->cur_share * 4; /* generous refill while idle         */
+    }
+
+    int64_t share = h->cur_share;
+    int ctl = c->controller == 3 ? 2 : c->controller; /* auto -> aimd  */
+    share = ctl == 1 ? ctl_delta(c, h, share, eff_target, cont)
+                     : ctl_aimd(c, h, share, eff_target, cont);
+    int64_t smin = h->pool / 100000;
+    if (smin < 1) smin = 1;
+    int64_t smax = h->pool;
+    if (share < smin) share = smin;
+    if (share > smax) share = smax;
+    h->cur_share = share;
+    return share;
+}
+
+/* refill the bucket (shared if present, else local)                   */
+static void refill(int dev, int64_t grant) {
+    dev_hot_t *h = &g_state.dev[dev];
+    int64_t cap = h->pool;
+    if (g_state.sm_node) {
+        sm_node_dev_t *s = &g_state.sm_node->devices[dev];
+        /* refill election: one process per container refills          */
+        int me = getpid();
+        int owner = __atomic_load_n(&s->refill_owner_pid, __ATOMIC_ACQUIRE);
+        uint64_t now = mono_ns();
+        if (owner != me) {
+            uint64_t last = __atomic_load_n(&s->refill_ns, __ATOMIC_ACQUIRE);
+            int stale = owner == 0 ||
+                        now - last > 3ull * WATCHER_CYCLE_MS * 1000000ull;
+            if (!stale) return;
+            if (!__atomic_compare_exchange_n(&s->refill_owner_pid, &owner,
+                                             me, false, __ATOMIC_ACQ_REL,
+                                             __ATOMIC_RELAXED))
+                return;
+            metrics_inc(MET_REFILL_TAKEOVER);
+        }
+        __atomic_store_n(&s->refill_ns, now, __ATOMIC_RELEASE);
+        s->cur_share = grant;
+        for (;;) {
+            int64_t cur = __atomic_load_n(&s->tokens, __ATOMIC_RELAXED);
+            int64_t next = cur + grant;
+            if (next > cap) next = cap;
+            if (next == cur) break;
+            if (__atomic_compare_exchange_n(&s->tokens, &cur, next, true,
+                                            __ATOMIC_ACQ_REL,
+                                            __ATOMIC_RELAXED))
+                break;
+        }
+    } else {
+        for (;;) {
+            int64_t cur = __atomic_load_n(&h->tokens, __ATOMIC_RELAXED);
+            int64_t next = cur + grant;
+            if (next > cap) next = cap;
+            if (next == cur) break;
+            if (__atomic_compare_exchange_n(&h->tokens, &cur, next, true,
+                                            __ATOMIC_ACQ_REL,
+                                            __ATOMIC_RELAXED))
+                break;
+        }
+    }
+}
+
+static void *watcher_main(void *arg) {
+    (void)arg;
+    /* absolute-time cadence: drift-free 100ms grid, overrun floor     */
+    uint64_t next = mono_ns();
+    for (;;) {
+        next += (uint64_t)WATCHER_CYCLE_MS * 1000000ull;
+        for (int dev = 0; dev < g_state.device_count; dev++) {
+            if (cfg_dev(dev) < 0) continue;
+            uint32_t flags = vgpu_device_flags(dev);
+            if (!(flags & DEV_FLAG_CORE_LIMIT)) continue;
+            dev_hot_init(dev);
+            int64_t grant = control_cycle(dev);
+            refill(dev, grant);
+        }
+        uint64_t now = mono_ns();
+        if (next <= now + 10000000ull) /* 10ms overrun floor           */
+            next = now + 10000000ull;
+        struct timespec ts = {(time_t)((next - now) / 1000000000ull),
+                              (long)((next - now) % 1000000000ull)};
+        nanosleep(&ts, NULL);
+    }
+    return NULL;
+}
+
+static void start_watcher(void) {
+    pthread_t t;
+    pthread_attr_t a;
+    pthread_attr_init(&a);
+    pthread_attr_setdetachstate(&a, PTHREAD_CREATE_DETACHED);
+    if (pthread_create(&t, &a, watcher_main, NULL) != 0)
+        LOGGER(LOG_ERROR, "failed to start utilization watcher");
+    pthread_attr_destroy(&a);
+}
+
+/* ------------------------------------------------------------------ */
+/* rate limiter (launch path)                                          */
+/* ------------------------------------------------------------------ */
+static void rate_limiter(int dev, int64_t grids) {
+    dev_hot_t *h = &g_state.dev[dev];
+    pthread_once(&g_watcher_once, start_watcher);
+    dev_hot_init(dev);
+    if (grids > h->pool) grids = h->pool;
+    int64_t *bucket = g_state.sm_node
+                          ? &g_state.sm_node->devices[dev].tokens
+                          : &h->tokens;
+    for (;;) {
+        int64_t cur = __atomic_load_n(bucket, __ATOMIC_RELAXED);
+        if (cur <= 0) {
+            __atomic_store_n(&h->throttled, 1u, __ATOMIC_RELAXED);
+            metrics_inc(MET_RATE_SLEEP);
+            struct timespec ts = {0, TIME_TICK_MS * 1000000L};
+            nanosleep(&ts, NULL);
+            continue;
+        }
+        if (__atomic_compare_exchange_n(bucket, &cur, cur - grids, true,
+                                        __ATOMIC_ACQ_REL, __ATOMIC_RELAXED))
+            break;
+    }
+    __atomic_store_n(&h->throttled, 0u, __ATOMIC_RELAXED);
+}
+
+/* ------------------------------------------------------------------ */
+/* GAP duty-cycle path: big synchronous kernels evade the token bucket
+ * (one launch, long runtime).  If launches are sparse (>200ms gap),
+ * measure the kernel with events and inject sleep = gpu_ms*(100/dc-1).
+ * The events/sync are only paid in the sparse-launch regime.          */
+/* ------------------------------------------------------------------ */
+#define GAP_IDLE_NS 200000000ull
+
+static int gap_begin(int dev, hipStream_t stream, uint32_t core_limit) {
+    if (vgpu_dynconfig()->gap_disable || core_limit == 0 ||
+        core_limit >= 100)
+        return 0;
+    dev_hot_t *h = &g_state.dev[dev];
+    uint64_t now = mono_ns();
+    uint64_t last = h->last_launch_ns;
+    h->last_launch_ns = now;
+    if (last != 0 && now - last < GAP_IDLE_NS) return 0;
+    if (pthread_mutex_trylock(&h->gap_mu) != 0) return 0;
+    if (!h->gap_start) {
+        if (real_hip.hipEventCreateWithFlags(&h->gap_start, 0) != hipSuccess ||
+            real_hip.hipEventCreateWithFlags(&h->gap_stop, 0) != hipSuccess) {
+            h->gap_start = h->gap_stop = NULL;
+            pthread_mutex_unlock(&h->gap_mu);
+            return 0;
+        }
+    }
+    if (real_hip.hipEventRecord(h->gap_start, stream) != hipSuccess) {
+        pthread_mutex_unlock(&h->gap_mu);
+        return 0;
+    }
+    return 1;
+}
+
+static void gap_end(int dev, hipStream_t stream, uint32_t core_limit) {
+    dev_hot_t *h = &g_state.dev[dev];
+    float ms = 0.f;
+    if (real_hip.hipEventRecord(h->gap_stop, stream) == hipSuccess &&
+        real_hip.hipEventSynchronize(h->gap_stop) == hipSuccess &&
+        real_hip.hipEventElapsedTime(&ms, h->gap_start, h->gap_stop) ==
+            hipSuccess &&
+        ms > 1.0f) {
+        /* kernel ran ms on GPU; duty cycle dc% => sleep ms*(100/dc-1).
+         * The sleep happens OUTSIDE any lock (reference gap design:
+         * holding a lock across the sleep would serialize siblings).  */
+        double sleep_ms = (double)ms * (100.0 / core_limit - 1.0);
+        if (sleep_ms > 5000.0) sleep_ms = 5000.0; /* bound single stall */
+        pthread_mutex_unlock(&h->gap_mu);
+        metrics_inc(MET_GAP_SLEEP);
+        struct timespec ts = {(time_t)(sleep_ms / 1000.0),
+                              (long)((uint64_t)(sleep_ms * 1e6) % 1000000000ull)};
+        nanosleep(&ts, NULL);
+        h->last_launch_ns = mono_ns();
+        return;
+    }
+    pthread_mutex_unlock(&h->gap_mu);
+}
+
+/* common launch gate                                                  */
+static inline int launch_gate(hipStream_t stream, int64_t grids,
+                              uint32_t *core_limit_out, int *dev_out) {
+    if (vgpu_ensure_init() != 0 || g_state.disabled) return 0;
+    int dev = cur_dev();
+    if (cfg_dev(dev) < 0) return 0;
+    uint32_t flags = vgpu_device_flags(dev); /* THE fast-path load     */
+    if (!(flags & DEV_FLAG_CORE_LIMIT)) return 0;
+    device_t snap;
+    vgpu_device_snapshot(dev, &snap);
+    rate_limiter(dev, grids);
+    *core_limit_out = snap.core_limit;
+    *dev_out = dev;
+    return gap_begin(dev, stream, snap.core_limit) ? 2 : 1;
+}
+
+/* ------------------------------------------------------------------ */
+/* memory gate                                                         */
+/* ------------------------------------------------------------------ */
+
+static uint64_t account_used(int dev);
+uint64_t vgpu_account_used(int dev) { return account_used(dev); }
+
+static uint64_t account_used(int dev) {
+    const dynamic_config_t *c = vgpu_dynconfig();
+    uint64_t ledger = dev_hooked_used(dev);
+    uint64_t used = ledger;
+    if (c->mem_account_mode != MEM_ACCOUNT_LEDGER && smi_available()) {
+        uint64_t smi = smi_container_vram(dev, &g_state.pids);
+        if (c->mem_account_mode == MEM_ACCOUNT_SMI) used = smi;
+        else used = smi > ledger ? smi : ledger;
+    }
+    return used + vmem_ledger_used(dev);
+}
+
+/* returns: 0 allow device alloc; 1 route to managed; <0 = OOM.
+ * On success *lockfd holds the per-device cross-container lock —
+ * release with vgpu_malloc_done after the real allocation.            */
+static int malloc_gate(int dev, size_t size, int *lockfd) {
+    *lockfd = -1;
+    if (vgpu_ensure_init() != 0 || g_state.disabled) return 0;
+    if (cfg_dev(dev) < 0) return 0;
+    uint32_t flags = vgpu_device_flags(dev);
+    if (!(flags & DEV_FLAG_MEM_LIMIT)) return 0;
+    device_t snap;
+    vgpu_device_snapshot(dev, &snap);
+    int host_index = snap.host_index >= 0 ? snap.host_index : dev;
+    int fd = lock_gpu_device(host_index);
+    uint64_t used = account_used(dev);
+    if (used + size > snap.total_memory) {
+        if (fd >= 0) unlock_gpu_device(fd);
+        int oversold = (snap.flags & DEV_FLAG_OVERSOLD) || g_state.cfg->oversold;
+        if (oversold) {
+            metrics_inc(MET_UVA_FALLBACK);
+            return 1;
+        }
+        metrics_inc(MET_OOM);
+        LOGGER(LOG_INFO,
+               "OOM: dev=%d req=%zu used=%llu quota=%llu", dev, size,
+               (unsigned long long)used,
+               (unsigned long long)snap.total_memory);
+        return -1;
+    }
+    *lockfd = fd;
+    return 0;
+}
+
+static void malloc_done(int lockfd) {
+    if (lockfd >= 0) unlock_gpu_device(lockfd);
+}
+
+/* managed spill: allocate HMM memory past the HBM cap + ledger record */
+static hipError_t managed_spill(int dev, void **ptr, size_t size, int kind) {
+    hipError_t rc =
+        real_hip.hipMallocManaged(ptr, size, hipMemAttachGlobal);
+    if (rc != hipSuccess) return rc;
+    if (vgpu_dynconfig()->uva_advise && real_hip.hipMemAdvise) {
+        /* prefer host residency for spilled ranges: the quota exists
+         * because HBM is contended; keep spill out of HBM until used. */
+        real_hip.hipMemAdvise(*ptr, size, hipMemAdviseSetPreferredLocation,
+                              hipCpuDeviceId);
+        real_hip.hipMemAdvise(*ptr, size, hipMemAdviseSetAccessedBy, dev);
+    }
+    int idx = vmem_ledger_add(dev, (uint64_t)(uintptr_t)*ptr, size, kind);
+    alloc_registry_add(*ptr, size, ALLOC_KIND_MANAGED, dev, idx);
+    return hipSuccess;
+}
+
+/* ------------------------------------------------------------------ */
+/* exported memory hooks                                               */
+/* ------------------------------------------------------------------ */
+
+EXPORT hipError_t hipMalloc(void **ptr, size_t size) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    if (size == 0 || g_state.disabled) return real_hip.hipMalloc(ptr, size);
+    int dev = cur_dev();
+    int lockfd;
+    int route = malloc_gate(dev, size, &lockfd);
+    if (route < 0) return hipErrorOutOfMemory;
+    if (route == 1) return managed_spill(dev, ptr, size, VMEM_KIND_SYNC);
+    hipError_t rc = real_hip.hipMalloc(ptr, size);
+    if (rc == hipErrorOutOfMemory &&
+        (g_state.cfg->oversold ||
+         (cfg_dev(dev) >= 0 &&
+          (vgpu_device_flags(dev) & DEV_FLAG_OVERSOLD)))) {
+        /* real HBM exhausted: spill (reference driver-OOM fallback)   */
+        malloc_done(lockfd);
+        return managed_spill(dev, ptr, size, VMEM_KIND_SYNC);
+    }
+    if (rc == hipSuccess) {
+        dev_hooked_add(dev, (int64_t)size);
+        alloc_registry_add(*ptr, size, ALLOC_KIND_DEVICE, dev, -1);
+    }
+    malloc_done(lockfd);
+    return rc;
+}
+
+EXPORT hipError_t hipExtMallocWithFlags(void **ptr, size_t size,
+                                        unsigned int flags) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    if (size == 0 || g_state.disabled)
+        return real_hip.hipExtMallocWithFlags(ptr, size, flags);
+    int dev = cur_dev();
+    int lockfd;
+    int route = malloc_gate(dev, size, &lockfd);
+    if (route < 0) return hipErrorOutOfMemory;
+    if (route == 1) return managed_spill(dev, ptr, size, VMEM_KIND_SYNC);
+    hipError_t rc = real_hip.hipExtMallocWithFlags(ptr, size, flags);
+    if (rc == hipSuccess) {
+        dev_hooked_add(dev, (int64_t)size);
+        alloc_registry_add(*ptr, size, ALLOC_KIND_DEVICE, dev, -1);
+    }
+    malloc_done(lockfd);
+    return rc;
+}
+
+EXPORT hipError_t hipMallocManaged(void **ptr, size_t size,
+                                   unsigned int flags) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    if (size == 0 || g_state.disabled)
+        return real_hip.hipMallocManaged(ptr, size, flags);
+    /* managed memory counts against the quota only while resident on
+     * device; we charge it to the vmem ledger (spoofed memGetInfo
+     * excludes it from the device-quota used).                        */
+    int dev = cur_dev();
+    hipError_t rc = real_hip.hipMallocManaged(ptr, size, flags);
+    if (rc == hipSuccess) {
+        int idx = vmem_ledger_add(dev, (uint64_t)(uintptr_t)*ptr, size,
+                                  VMEM_KIND_SYNC);
+        alloc_registry_add(*ptr, size, ALLOC_KIND_MANAGED, dev, idx);
+    }
+    return rc;
+}
+
+EXPORT hipError_t hipMallocAsync(void **ptr, size_t size,
+                                 hipStream_t stream) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    if (size == 0 || g_state.disabled)
+        return real_hip.hipMallocAsync(ptr, size, stream);
+    int dev = cur_dev();
+    /* capture-aware: during stream capture the allocation happens at
+     * graph launch; record kind CAPTURE so the ledger stays truthful. */
+    int kind = ALLOC_KIND_ASYNC;
+    int vkind = VMEM_KIND_ASYNC;
+    hipStreamCaptureStatus cap = hipStreamCaptureStatusNone;
+    if (real_hip.hipStreamIsCapturing)
+        real_hip.hipStreamIsCapturing(stream, &cap);
+    if (cap == hipStreamCaptureStatusActive) vkind = VMEM_KIND_CAPTURE;
+    int lockfd;
+    int route = malloc_gate(dev, size, &lockfd);
+    if (route < 0) return hipErrorOutOfMemory;
+    if (route == 1) {
+        /* async spill degrades to sync managed alloc                  */
+        return managed_spill(dev, ptr, size, vkind);
+    }
+    hipError_t rc = real_hip.hipMallocAsync(ptr, size, stream);
+    if (rc == hipSuccess) {
+        dev_hooked_add(dev, (int64_t)size);
+        alloc_registry_add(*ptr, size, kind, dev, -1);
+    }
+    malloc_done(lockfd);
+    return rc;
+}
+
+EXPORT hipError_t hipMallocFromPoolAsync(void **ptr, size_t size,
+                                         hipMemPool_t pool,
+                                         hipStream_t stream) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    if (size == 0 || g_state.disabled)
+        return real_hip.hipMallocFromPoolAsync(ptr, size, pool, stream);
+    int dev = cur_dev();
+    int lockfd;
+    int route = malloc_gate(dev, size, &lockfd);
+    if (route < 0) return hipErrorOutOfMemory;
+    if (route == 1) return managed_spill(dev, ptr, size, VMEM_KIND_ASYNC);
+    hipError_t rc = real_hip.hipMallocFromPoolAsync(ptr, size, pool, stream);
+    if (rc == hipSuccess) {
+        dev_hooked_add(dev, (int64_t)size);
+        alloc_registry_add(*ptr, size, ALLOC_KIND_ASYNC, dev, -1);
+    }
+    malloc_done(lockfd);
+    return rc;
+}
+
+EXPORT hipError_t hipMallocPitch(void **ptr, size_t *pitch, size_t width,
+                                 size_t height) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    if (g_state.disabled)
+        return real_hip.hipMallocPitch(ptr, pitch, width, height);
+    int dev = cur_dev();
+    size_t est = ((width + 255) & ~(size_t)255) * height; /* pitch est. */
+    int lockfd;
+    int route = malloc_gate(dev, est, &lockfd);
+    if (route < 0) return hipErrorOutOfMemory;
+    if (route == 1) {
+        /* pitched layout cannot spill; fail the quota                 */
+        metrics_inc(MET_OOM);
+        return hipErrorOutOfMemory;
+    }
+    hipError_t rc = real_hip.hipMallocPitch(ptr, pitch, width, height);
+    if (rc == hipSuccess) {
+        size_t real_size = *pitch * height;
+        dev_hooked_add(dev, (int64_t)real_size);
+        alloc_registry_add(*ptr, real_size, ALLOC_KIND_DEVICE, dev, -1);
+    }
+    malloc_done(lockfd);
+    return rc;
+}
+
+EXPORT hipError_t hipMalloc3D(hipPitchedPtr *p, hipExtent extent) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    if (g_state.disabled) return real_hip.hipMalloc3D(p, extent);
+    int dev = cur_dev();
+    size_t est = ((extent.width + 255) & ~(size_t)255) * extent.height *
+                 extent.depth;
+    int lockfd;
+    int route = malloc_gate(dev, est, &lockfd);
+    if (route != 0) {
+        if (lockfd >= 0) malloc_done(lockfd);
+        metrics_inc(MET_OOM);
+        return hipErrorOutOfMemory;
+    }
+    hipError_t rc = real_hip.hipMalloc3D(p, extent);
+    if (rc == hipSuccess) {
+        size_t real_size = p->pitch * extent.height * extent.depth;
+        dev_hooked_add(dev, (int64_t)real_size);
+        alloc_registry_add(p->ptr, real_size, ALLOC_KIND_DEVICE, dev, -1);
+    }
+    malloc_done(lockfd);
+    return rc;
+}
+
+static size_t channel_bytes(const hipChannelFormatDesc *d) {
+    return (size_t)(d->x + d->y + d->z + d->w) / 8;
+}
+
+EXPORT hipError_t hipMallocArray(hipArray_t *array,
+                                 const hipChannelFormatDesc *desc,
+                                 size_t width, size_t height,
+                                 unsigned int flags) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    if (g_state.disabled)
+        return real_hip.hipMallocArray(array, desc, width, height, flags);
+    int dev = cur_dev();
+    size_t est = width * (height ? height : 1) * channel_bytes(desc);
+    int lockfd;
+    int route = malloc_gate(dev, est, &lockfd);
+    if (route != 0) {
+        if (lockfd >= 0) malloc_done(lockfd);
+        metrics_inc(MET_OOM);
+        return hipErrorOutOfMemory;
+    }
+    hipError_t rc = real_hip.hipMallocArray(array, desc, width, height, flags);
+    if (rc == hipSuccess) {
+        dev_hooked_add(dev, (int64_t)est);
+        alloc_registry_add((void *)*array, est, ALLOC_KIND_DEVICE, dev, -1);
+    }
+    malloc_done(lockfd);
+    return rc;
+}
+
+EXPORT hipError_t hipMalloc3DArray(hipArray_t *array,
+                                   const hipChannelFormatDesc *desc,
+                                   hipExtent extent, unsigned int flags) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    if (g_state.disabled)
+        return real_hip.hipMalloc3DArray(array, desc, extent, flags);
+    int dev = cur_dev();
+    size_t est = extent.width * (extent.height ? extent.height : 1) *
+                 (extent.depth ? extent.depth : 1) * channel_bytes(desc);
+    int lockfd;
+    int route = malloc_gate(dev, est, &lockfd);
+    if (route != 0) {
+        if (lockfd >= 0) malloc_done(lockfd);
+        metrics_inc(MET_OOM);
+        return hipErrorOutOfMemory;
+    }
+    hipError_t rc = real_hip.hipMalloc3DArray(array, desc, extent, flags);
+    if (rc == hipSuccess) {
+        dev_hooked_add(dev, (int64_t)est);
+        alloc_registry_add((void *)*array, est, ALLOC_KIND_DEVICE, dev, -1);
+    }
+    malloc_done(lockfd);
+    return rc;
+}
+
+static void release_tracking(void *ptr) {
+    size_t size;
+    int kind, dev, vmem_idx;
+    if (!alloc_registry_remove(ptr, &size, &kind, &dev, &vmem_idx)) return;
+    if (kind == ALLOC_KIND_MANAGED) {
+        if (vmem_idx >= 0) vmem_ledger_remove(vmem_idx);
+    } else {
+        dev_hooked_add(dev, -(int64_t)size);
+    }
+}
+
+EXPORT hipError_t hipFree(void *ptr) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    hipError_t rc = real_hip.hipFree(ptr);
+    if (rc == hipSuccess && !g_state.disabled) release_tracking(ptr);
+    return rc;
+}
+
+EXPORT hipError_t hipFreeAsync(void *ptr, hipStream_t stream) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    hipError_t rc = real_hip.hipFreeAsync(ptr, stream);
+    /* async free completes later; the ledger retires now — the quota
+     * is conservative by at most the in-flight frees (reference
+     * ASYNC_BRIDGE semantics collapsed: HIP pools return memory to the
+     * pool, so the charge stays until pool trim anyway).              */
+    if (rc == hipSuccess && !g_state.disabled) release_tracking(ptr);
+    return rc;
+}
+
+EXPORT hipError_t hipFreeArray(hipArray_t array) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    hipError_t rc = real_hip.hipFreeArray(array);
+    if (rc == hipSuccess && !g_state.disabled)
+        release_tracking((void *)array);
+    return rc;
+}
+
+/* ---- view spoofing ---- */
+
+EXPORT hipError_t hipMemGetInfo(size_t *free_out, size_t *total_out) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    hipError_t rc = real_hip.hipMemGetInfo(free_out, total_out);
+    if (rc != hipSuccess || g_state.disabled) return rc;
+    int dev = cur_dev();
+    if (cfg_dev(dev) < 0 || !(vgpu_device_flags(dev) & DEV_FLAG_MEM_LIMIT))
+        return rc;
+    device_t snap;
+    vgpu_device_snapshot(dev, &snap);
+    uint64_t used = account_used(dev);
+    uint64_t quota = snap.total_memory;
+    if (total_out) *total_out = (size_t)quota;
+    if (free_out) *free_out = used >= quota ? 0 : (size_t)(quota - used);
+    return hipSuccess;
+}
+
+EXPORT hipError_t hipDeviceTotalMem(size_t *bytes, hipDevice_t device) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    hipError_t rc = real_hip.hipDeviceTotalMem(bytes, device);
+    if (rc != hipSuccess || g_state.disabled) return rc;
+    int dev = (int)device;
+    if (cfg_dev(dev) < 0 || !(vgpu_device_flags(dev) & DEV_FLAG_MEM_LIMIT))
+        return rc;
+    device_t snap;
+    vgpu_device_snapshot(dev, &snap);
+    *bytes = (size_t)snap.total_memory;
+    return hipSuccess;
+}
+
+EXPORT hipError_t hipGetDevicePropertiesR0600(hipDeviceProp_tR0600 *prop,
+                                              int device) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    hipError_t rc = real_hip.hipGetDevicePropertiesR0600(prop, device);
+    if (rc != hipSuccess || g_state.disabled) return rc;
+    if (cfg_dev(device) >= 0 &&
+        (vgpu_device_flags(device) & DEV_FLAG_MEM_LIMIT)) {
+        device_t snap;
+        vgpu_device_snapshot(device, &snap);
+        prop->totalGlobalMem = (size_t)snap.total_memory;
+    }
+    return rc;
+}
+
+/* ---- device tracking ---- */
+
+EXPORT hipError_t hipSetDevice(int device) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    hipError_t rc = real_hip.hipSetDevice(device);
+    if (rc == hipSuccess) tls_device = device;
+    return rc;
+}
+
+/* ------------------------------------------------------------------ */
+/* exported launch hooks                                               */
+/* ------------------------------------------------------------------ */
+
+EXPORT hipError_t hipLaunchKernel(const void *function_address,
+                                  dim3 numBlocks, dim3 dimBlocks,
+                                  void **args, size_t sharedMemBytes,
+                                  hipStream_t stream) {
+    uint32_t cl = 0;
+    int dev = 0;
+    int g = launch_gate(stream,
+                        (int64_t)numBlocks.x * numBlocks.y * numBlocks.z,
+                        &cl, &dev);
+    hipError_t rc = real_hip.hipLaunchKernel(function_address, numBlocks,
+                                             dimBlocks, args, sharedMemBytes,
+                                             stream);
+    if (g == 2) gap_end(dev, stream, cl);
+    return rc;
+}
+
+EXPORT hipError_t hipExtLaunchKernel(const void *function_address,
+                                     dim3 numBlocks, dim3 dimBlocks,
+                                     void **args, size_t sharedMemBytes,
+                                     hipStream_t stream, hipEvent_t startEvent,
+                                     hipEvent_t stopEvent, int flags) {
+    uint32_t cl = 0;
+    int dev = 0;
+    int g = launch_gate(stream,
+                        (int64_t)numBlocks.x * numBlocks.y * numBlocks.z,
+                        &cl, &dev);
+    hipError_t rc = real_hip.hipExtLaunchKernel(
+        function_address, numBlocks, dimBlocks, args, sharedMemBytes, stream,
+        startEvent, stopEvent, flags);
+    if (g == 2) gap_end(dev, stream, cl);
+    return rc;
+}
+
+EXPORT hipError_t hipModuleLaunchKernel(
+    hipFunction_t f, unsigned int gridDimX, unsigned int gridDimY,
+    unsigned int gridDimZ, unsigned int blockDimX, unsigned int blockDimY,
+    unsigned int blockDimZ, unsigned int sharedMemBytes, hipStream_t stream,
+    void **kernelParams, void **extra) {
+    uint32_t cl = 0;
+    int dev = 0;
+    int g = launch_gate(stream, (int64_t)gridDimX * gridDimY * gridDimZ, &cl,
+                        &dev);
+    hipError_t rc = real_hip.hipModuleLaunchKernel(
+        f, gridDimX, gridDimY, gridDimZ, blockDimX, blockDimY, blockDimZ,
+        sharedMemBytes, stream, kernelParams, extra);
+    if (g == 2) gap_end(dev, stream, cl);
+    return rc;
+}
+
+EXPORT hipError_t hipExtModuleLaunchKernel(
+    hipFunction_t f, uint32_t globalWorkSizeX, uint32_t globalWorkSizeY,
+    uint32_t globalWorkSizeZ, uint32_t localWorkSizeX, uint32_t localWorkSizeY,
+    uint32_t localWorkSizeZ, size_t sharedMemBytes, hipStream_t stream,
+    void **kernelParams, void **extra, hipEvent_t startEvent,
+    hipEvent_t stopEvent, uint32_t flags) {
+    /* global work size, not grid: convert to workgroup count          */
+    int64_t bx = localWorkSizeX ? globalWorkSizeX / localWorkSizeX : 1;
+    int64_t by = localWorkSizeY ? globalWorkSizeY / localWorkSizeY : 1;
+    int64_t bz = localWorkSizeZ ? globalWorkSizeZ / localWorkSizeZ : 1;
+    uint32_t cl = 0;
+    int dev = 0;
+    int g = launch_gate(stream, bx * by * bz, &cl, &dev);
+    hipError_t rc = real_hip.hipExtModuleLaunchKernel(
+        f, globalWorkSizeX, globalWorkSizeY, globalWorkSizeZ, localWorkSizeX,
+        localWorkSizeY, localWorkSizeZ, sharedMemBytes, stream, kernelParams,
+        extra, startEvent, stopEvent, flags);
+    if (g == 2) gap_end(dev, stream, cl);
+    return rc;
+}
+
+EXPORT hipError_t hipLaunchCooperativeKernel(const void *f, dim3 gridDim,
+                                             dim3 blockDimX, void **kernelParams,
+                                             unsigned int sharedMemBytes,
+                                             hipStream_t stream) {
+    uint32_t cl = 0;
+    int dev = 0;
+    int g = launch_gate(stream, (int64_t)gridDim.x * gridDim.y * gridDim.z,
+                        &cl, &dev);
+    hipError_t rc = real_hip.hipLaunchCooperativeKernel(
+        f, gridDim, blockDimX, kernelParams, sharedMemBytes, stream);
+    if (g == 2) gap_end(dev, stream, cl);
+    return rc;
+}
+
+EXPORT hipError_t hipModuleLaunchCooperativeKernel(
+    hipFunction_t f, unsigned int gridDimX, unsigned int gridDimY,
+    unsigned int gridDimZ, unsigned int blockDimX, unsigned int blockDimY,
+    unsigned int blockDimZ, unsigned int sharedMemBytes, hipStream_t stream,
+    void **kernelParams) {
+    uint32_t cl = 0;
+    int dev = 0;
+    int g = launch_gate(stream, (int64_t)gridDimX * gridDimY * gridDimZ, &cl,
+                        &dev);
+    hipError_t rc = real_hip.hipModuleLaunchCooperativeKernel(
+        f, gridDimX, gridDimY, gridDimZ, blockDimX, blockDimY, blockDimZ,
+        sharedMemBytes, stream, kernelParams);
+    if (g == 2) gap_end(dev, stream, cl);
+    return rc;
+}
+
+/* ------------------------------------------------------------------ */
+/* graph cost accounting                                               */
+/* ------------------------------------------------------------------ */
+#define GRAPH_MAP_SLOTS 256
+
+typedef struct {
+    hipGraphExec_t exec;
+    int64_t grids;
+} graph_cost_t;
+
+static graph_cost_t g_graph_cost[GRAPH_MAP_SLOTS];
+static pthread_mutex_t g_graph_mu = PTHREAD_MUTEX_INITIALIZER;
+
+static int64_t graph_count_grids(hipGraph_t graph) {
+    size_t n = 0;
+    if (!real_hip.hipGraphGetNodes ||
+        real_hip.hipGraphGetNodes(graph, NULL, &n) != hipSuccess || n == 0)
+        return 1;
+    hipGraphNode_t *nodes = malloc(n * sizeof(*nodes));
+    if (!nodes) return 1;
+    int64_t total = 0;
+    if (real_hip.hipGraphGetNodes(graph, nodes, &n) == hipSuccess) {
+        for (size_t i = 0; i < n; i++) {
+            hipGraphNodeType t;
+            if (real_hip.hipGraphNodeGetType(nodes[i], &t) != hipSuccess ||
+                t != hipGraphNodeTypeKernel)
+                continue;
+            hipKernelNodeParams p;
+            memset(&p, 0, sizeof(p));
+            if (real_hip.hipGraphKernelNodeGetParams(nodes[i], &p) ==
+                hipSuccess)
+                total += (int64_t)p.gridDim.x * p.gridDim.y * p.gridDim.z;
+        }
+    }
+    free(nodes);
+    return total > 0 ? total : 1;
+}
+
+static void graph_cost_set(hipGraphExec_t exec, int64_t grids) {
+    pthread_mutex_lock(&g_graph_mu);
+    for (int i = 0; i < GRAPH_MAP_SLOTS; i++) {
+        if (g_graph_cost[i].exec == NULL || g_graph_cost[i].exec == exec) {
+            g_graph_cost[i].exec = exec;
+            g_graph_cost[i].grids = grids;
+            break;
+        }
+    }
+    pthread_mutex_unlock(&g_graph_mu);
+}
+
+static int64_t graph_cost_get(hipGraphExec_t exec) {
+    pthread_mutex_lock(&g_graph_mu);
+    int64_t g = 1;
+    for (int i = 0; i < GRAPH_MAP_SLOTS; i++) {
+        if (g_graph_cost[i].exec == exec) {
+            g = g_graph_cost[i].grids;
+            break;
+        }
+    }
+    pthread_mutex_unlock(&g_graph_mu);
+    return g;
+}
+
+static void graph_cost_del(hipGraphExec_t exec) {
+    pthread_mutex_lock(&g_graph_mu);
+    for (int i = 0; i < GRAPH_MAP_SLOTS; i++)
+        if (g_graph_cost[i].exec == exec) g_graph_cost[i].exec = (hipGraphExec_t)1;
+    pthread_mutex_unlock(&g_graph_mu);
+}
+
+EXPORT hipError_t hipGraphInstantiate(hipGraphExec_t *pGraphExec,
+                                      hipGraph_t graph,
+                                      hipGraphNode_t *pErrorNode,
+                                      char *pLogBuffer, size_t bufferSize) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    hipError_t rc = real_hip.hipGraphInstantiate(pGraphExec, graph,
+                                                 pErrorNode, pLogBuffer,
+                                                 bufferSize);
+    if (rc == hipSuccess && !g_state.disabled)
+        graph_cost_set(*pGraphExec, graph_count_grids(graph));
+    return rc;
+}
+
+EXPORT hipError_t hipGraphInstantiateWithFlags(hipGraphExec_t *pGraphExec,
+                                               hipGraph_t graph,
+                                               unsigned long long flags) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    hipError_t rc =
+        real_hip.hipGraphInstantiateWithFlags(pGraphExec, graph, flags);
+    if (rc == hipSuccess && !g_state.disabled)
+        graph_cost_set(*pGraphExec, graph_count_grids(graph));
+    return rc;
+}
+
+EXPORT hipError_t hipGraphExecDestroy(hipGraphExec_t exec) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    hipError_t rc = real_hip.hipGraphExecDestroy(exec);
+    if (rc == hipSuccess) graph_cost_del(exec);
+    return rc;
+}
+
+EXPORT hipError_t hipGraphLaunch(hipGraphExec_t exec, hipStream_t stream) {
+    uint32_t cl = 0;
+    int dev = 0;
+    int64_t grids = 1;
+    if (vgpu_ensure_init() == 0 && !g_state.disabled)
+        grids = graph_cost_get(exec);
+    int g = launch_gate(stream, grids, &cl, &dev);
+    hipError_t rc = real_hip.hipGraphLaunch(exec, stream);
+    if (g == 2) gap_end(dev, stream, cl);
+    return rc;
+}
+
+/* ------------------------------------------------------------------ */
+/* hipGetProcAddress routing                                           */
+/* ------------------------------------------------------------------ */
+EXPORT hipError_t hipGetProcAddress(const char *symbol, void **pfn,
+                                    int hipVersion, uint64_t flags,
+                                    hipDriverProcAddressQueryResult *res) {
+    if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
+    hipError_t rc = real_hip.hipGetProcAddress
+                        ? real_hip.hipGetProcAddress(symbol, pfn, hipVersion,
+                                                     flags, res)
+                        : hipErrorNotSupported;
+    if (rc == hipSuccess && !g_state.disabled && pfn && *pfn) {
+        void *hook = vgpu_lookup_hook(symbol);
+        if (hook) {
+            LOGGER(LOG_TRACE, "hipGetProcAddress(%s) -> hook", symbol);
+            *pfn = hook;
+        }
+    }
+    return rc;
+}
+
+/* ------------------------------------------------------------------ */
+/* hook table for dlsym / hipGetProcAddress routing                    */
+/* ------------------------------------------------------------------ */
+extern void *vgpu_smi_lookup_hook(const char *name); /* smi_hook.c     */
+
+typedef struct {
+    const char *name;
+    void *fn;
+} hook_entry_t;
+
+static const hook_entry_t g_hooks[] = {
+    {"hipMalloc", (void *)hipMalloc},
+    {"hipExtMallocWithFlags", (void *)hipExtMallocWithFlags},
+    {"hipMallocManaged", (void *)hipMallocManaged},
+    {"hipMallocAsync", (void *)hipMallocAsync},
+    {"hipMallocFromPoolAsync", (void *)hipMallocFromPoolAsync},
+    {"hipMallocPitch", (void *)hipMallocPitch},
+    {"hipMalloc3D", (void *)hipMalloc3D},
+    {"hipMallocArray", (void *)hipMallocArray},
+    {"hipMalloc3DArray", (void *)hipMalloc3DArray},
+    {"hipFree", (void *)hipFree},
+    {"hipFreeAsync", (void *)hipFreeAsync},
+    {"hipFreeArray", (void *)hipFreeArray},
+    {"hipMemGetInfo", (void *)hipMemGetInfo},
+    {"hipDeviceTotalMem", (void *)hipDeviceTotalMem},
+    {"hipGetDeviceProperties", (void *)hipGetDevicePropertiesR0600},
+    {"hipGetDevicePropertiesR0600", (void *)hipGetDevicePropertiesR0600},
+    {"hipSetDevice", (void *)hipSetDevice},
+    {"hipLaunchKernel", (void *)hipLaunchKernel},
+    {"hipExtLaunchKernel", (void *)hipExtLaunchKernel},
+    {"hipModuleLaunchKernel", (void *)hipModuleLaunchKernel},
+    {"hipExtModuleLaunchKernel", (void *)hipExtModuleLaunchKernel},
+    {"hipLaunchCooperativeKernel", (void *)hipLaunchCooperativeKernel},
+    {"hipModuleLaunchCooperativeKernel",
+     (void *)hipModuleLaunchCooperativeKernel},
+    {"hipGraphInstantiate", (void *)hipGraphInstantiate},
+    {"hipGraphInstantiateWithFlags", (void *)hipGraphInstantiateWithFlags},
+    {"hipGraphExecDestroy", (void *)hipGraphExecDestroy},
+    {"hipGraphLaunch", (void *)hipGraphLaunch},
+    {"hipGetProcAddress", (void *)hipGetProcAddress},
+    {NULL, NULL},
+};
+
+void *vgpu_lookup_hook(const char *name) {
+    for (const hook_entry_t *e = g_hooks; e->name; e++)
+        if (strcmp(e->name, name) == 0) return e->fn;
+    return vgpu_smi_lookup_hook(name);
+}

@@ -1,0 +1,520 @@
+/* loader.c — interposition bootstrap for libvgpu-control.so.
+ *
+ * Responsibilities (design parity with reference library/src/loader.c,
+ * re-architected for ROCm):
+ *   - real-dlsym bootstrap (dlvsym over the glibc version ladder);
+ *   - the exported `dlsym` hook routing hip / amdsmi / rsmi lookups to
+ *     our hook table (ctypes/dlopen users), everything else passthrough;
+ *   - dlopen of the real libamdhip64 and the typed real-function table;
+ *   - attach of the four shared regions + env-var bootstrap fallback;
+ *   - the process-local allocation registry and vmem-ledger ops;
+ *   - fork handling (child re-registers its pid, drops local buckets).
+ *
+ * MI355X-native divergence from the reference: ROCm applications link
+ * libamdhip64 directly, so LD_PRELOAD interposes at symbol-bind time
+ * and NO passthrough stubs of the full API surface are needed (the
+ * reference ships ~6k LoC of generated originals for dlopen'd libcuda;
+ * here un-hooked symbols never enter our library at all).
+ */
+#define _GNU_SOURCE
+#include "state.h"
+#include "shm.h"
+
+#include <dlfcn.h>
+#include <errno.h>
+#include <pthread.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+#include <unistd.h>
+
+hip_real_t real_hip;
+vgpu_state_t g_state;
+
+/* ------------------------------------------------------------------ */
+/* real dlsym bootstrap                                                */
+/* ------------------------------------------------------------------ */
+typedef void *(*dlsym_fn_t)(void *, const char *);
+static dlsym_fn_t g_real_dlsym;
+
+static void init_real_dlsym(void) {
+    /* glibc's dlsym lives at a versioned symbol; dlvsym is NOT
+     * interposed by us so it is safe to call.  Try the version ladder
+     * newest-first (2.34 = post-libdl-merge), then the x86_64 baseline. */
+    static const char *vers[] = {"GLIBC_2.34", "GLIBC_2.2.5", "GLIBC_2.17"};
+    for (size_t i = 0; i < sizeof(vers) / sizeof(vers[0]); i++) {
+        void *p = dlvsym(RTLD_NEXT, "dlsym", vers[i]);
+        if (p) {
+            g_real_dlsym = (dlsym_fn_t)p;
+            return;
+        }
+    }
+    /* last resort: default lookup may find ourselves; guard against
+     * recursion in the dlsym hook below. */
+    g_real_dlsym = NULL;
+    LOGGER(LOG_ERROR, "failed to resolve real dlsym via dlvsym");
+}
+
+void *vgpu_real_dlsym(void *handle, const char *name) {
+    static pthread_once_t once = PTHREAD_ONCE_INIT;
+    pthread_once(&once, init_real_dlsym);
+    if (!g_real_dlsym) return NULL;
+    return g_real_dlsym(handle, name);
+}
+
+/* ------------------------------------------------------------------ */
+/* exported dlsym hook                                                 */
+/* ------------------------------------------------------------------ */
+__attribute__((visibility("default")))
+void *dlsym(void *handle, const char *name) {
+    /* recursion guard: a hook-table miss must not loop back here      */
+    static __thread int in_dlsym;
+    if (in_dlsym) return vgpu_real_dlsym(handle, name);
+    in_dlsym = 1;
+    void *ret = NULL;
+    const char *nm = name; /* glibc marks name nonnull; be defensive   */
+    if (nm && (strncmp(name, "hip", 3) == 0 ||
+                 strncmp(name, "amdsmi_", 7) == 0 ||
+                 strncmp(name, "rsmi_", 5) == 0)) {
+        ret = vgpu_lookup_hook(name);
+        if (ret) {
+            LOGGER(LOG_TRACE, "dlsym(%s) -> hook", name);
+            in_dlsym = 0;
+            return ret;
+        }
+    }
+    ret = vgpu_real_dlsym(handle, name);
+    in_dlsym = 0;
+    return ret;
+}
+
+/* ------------------------------------------------------------------ */
+/* real libamdhip64 loading + entry table                              */
+/* ------------------------------------------------------------------ */
+static void *g_hip_handle;
+
+static void *hip_sym(const char *name) {
+    void *p = vgpu_real_dlsym(g_hip_handle, name);
+    if (!p) LOGGER(LOG_DEBUG, "libamdhip64: missing symbol %s", name);
+    return p;
+}
+
+static int load_real_hip(void) {
+    const char *paths[] = {
+        getenv("VGPU_REAL_HIP_PATH"),
+        "libamdhip64.so.7",
+        "libamdhip64.so",
+        "/opt/rocm/lib/libamdhip64.so",
+    };
+    for (size_t i = 0; i < sizeof(paths) / sizeof(paths[0]); i++) {
+        if (!paths[i]) continue;
+        g_hip_handle = dlopen(paths[i], RTLD_LAZY | RTLD_LOCAL);
+        if (g_hip_handle) break;
+    }
+    if (!g_hip_handle) {
+        LOGGER(LOG_ERROR, "cannot dlopen real libamdhip64: %s", dlerror());
+        return -1;
+    }
+#define LOAD(sym) real_hip.sym = (__typeof__(real_hip.sym))hip_sym(#sym)
+    LOAD(hipMalloc);
+    LOAD(hipExtMallocWithFlags);
+    LOAD(hipMallocManaged);
+    LOAD(hipMallocAsync);
+    LOAD(hipMallocFromPoolAsync);
+    LOAD(hipMallocPitch);
+    LOAD(hipMalloc3D);
+    LOAD(hipMallocArray);
+    LOAD(hipMalloc3DArray);
+    LOAD(hipFree);
+    LOAD(hipFreeAsync);
+    LOAD(hipFreeArray);
+    LOAD(hipMemGetInfo);
+    LOAD(hipDeviceTotalMem);
+    LOAD(hipGetDevicePropertiesR0600);
+    LOAD(hipMemAdvise);
+    LOAD(hipMemPrefetchAsync);
+    LOAD(hipLaunchKernel);
+    LOAD(hipExtLaunchKernel);
+    LOAD(hipModuleLaunchKernel);
+    LOAD(hipExtModuleLaunchKernel);
+    LOAD(hipLaunchCooperativeKernel);
+    LOAD(hipModuleLaunchCooperativeKernel);
+    LOAD(hipGraphLaunch);
+    LOAD(hipGraphInstantiate);
+    LOAD(hipGraphInstantiateWithFlags);
+    LOAD(hipGraphExecDestroy);
+    LOAD(hipGraphGetNodes);
+    LOAD(hipGraphNodeGetType);
+    LOAD(hipGraphKernelNodeGetParams);
+    LOAD(hipGetDevice);
+    LOAD(hipSetDevice);
+    LOAD(hipGetDeviceCount);
+    LOAD(hipDeviceGetAttribute);
+    LOAD(hipDeviceGetUuid);
+    LOAD(hipEventCreateWithFlags);
+    LOAD(hipEventRecord);
+    LOAD(hipEventSynchronize);
+    LOAD(hipEventElapsedTime);
+    LOAD(hipEventDestroy);
+    LOAD(hipStreamIsCapturing);
+    LOAD(hipGetProcAddress);
+#undef LOAD
+    return real_hip.hipMalloc && real_hip.hipLaunchKernel ? 0 : -1;
+}
+
+/* ------------------------------------------------------------------ */
+/* env bootstrap (no vgpu.config mounted; dev/test path)               */
+/* ------------------------------------------------------------------ */
+static void env_bootstrap_config(resource_data_t *cfg) {
+    char buf[128], name[64];
+    cfg->hdr.magic = VGPU_CFG_MAGIC;
+    cfg->hdr.abi_version = VGPU_ABI_VERSION;
+    cfg->hdr.region_size = sizeof(*cfg);
+    const char *v;
+    if ((v = vgpu_getenv("VGPU_POD_UID", buf, sizeof(buf))))
+        snprintf(cfg->pod_uid, sizeof(cfg->pod_uid), "%s", v);
+    if ((v = vgpu_getenv("VGPU_POD_NAME", buf, sizeof(buf))))
+        snprintf(cfg->pod_name, sizeof(cfg->pod_name), "%s", v);
+    if ((v = vgpu_getenv("VGPU_POD_NAMESPACE", buf, sizeof(buf))))
+        snprintf(cfg->pod_namespace, sizeof(cfg->pod_namespace), "%s", v);
+    if ((v = vgpu_getenv("VGPU_CONTAINER_NAME", buf, sizeof(buf))))
+        snprintf(cfg->container_name, sizeof(cfg->container_name), "%s", v);
+
+    cfg->compute_policy = COMPUTE_POLICY_FIXED;
+    if ((v = vgpu_getenv("VGPU_COMPUTE_POLICY", buf, sizeof(buf)))) {
+        if (strcasecmp(v, "balance") == 0)
+            cfg->compute_policy = COMPUTE_POLICY_BALANCE;
+        else if (strcasecmp(v, "none") == 0)
+            cfg->compute_policy = COMPUTE_POLICY_NONE;
+    }
+    int oversold = 0;
+    if ((v = vgpu_getenv("VGPU_MEM_OVERSOLD", buf, sizeof(buf))))
+        oversold = atoi(v) != 0;
+    cfg->oversold = (uint32_t)oversold;
+
+    int n = 0;
+    for (int i = 0; i < MAX_DEVICE_COUNT; i++) {
+        device_t *d = &cfg->devices[i];
+        int any = 0;
+        snprintf(name, sizeof(name), "VGPU_MEM_LIMIT_%d", i);
+        if ((v = vgpu_getenv(name, buf, sizeof(buf)))) {
+            long long b = vgpu_parse_size(v);
+            if (b > 0) {
+                d->total_memory = (uint64_t)b;
+                d->flags |= DEV_FLAG_MEM_LIMIT;
+                any = 1;
+            }
+        }
+        snprintf(name, sizeof(name), "VGPU_CORE_LIMIT_%d", i);
+        if ((v = vgpu_getenv(name, buf, sizeof(buf)))) {
+            int c = atoi(v);
+            if (c > 0 && c <= 100) {
+                d->core_limit = (uint32_t)c;
+                d->flags |= DEV_FLAG_CORE_LIMIT;
+                any = 1;
+            }
+        }
+        snprintf(name, sizeof(name), "VGPU_CORE_SOFT_LIMIT_%d", i);
+        if ((v = vgpu_getenv(name, buf, sizeof(buf)))) {
+            int c = atoi(v);
+            if (c > 0 && c <= 100) {
+                d->soft_core_limit = (uint32_t)c;
+                d->flags |= DEV_FLAG_SOFT_CORE;
+                any = 1;
+            }
+        }
+        if (oversold) d->flags |= DEV_FLAG_OVERSOLD;
+        d->host_index = i;
+        if (any) n = i + 1;
+    }
+    cfg->device_count = n;
+    LOGGER(LOG_INFO, "env bootstrap: %d device limits", n);
+}
+
+/* ------------------------------------------------------------------ */
+/* load_necessary_data                                                 */
+/* ------------------------------------------------------------------ */
+static pthread_once_t g_init_once = PTHREAD_ONCE_INIT;
+static int g_init_rc = -1;
+
+void vgpu_device_snapshot(int dev, device_t *out) {
+    const device_t *d = &g_state.cfg->devices[dev];
+    if (!g_state.cfg_shared) {
+        memcpy(out, d, sizeof(*out));
+        return;
+    }
+    for (;;) {
+        uint32_t s0 = seq_load(&d->seq);
+        if (s0 & 1u) continue;
+        memcpy(out, d, sizeof(*out));
+        if (seq_read_valid(&d->seq, s0)) return;
+    }
+}
+
+static void fork_child_handler(void) {
+    /* child: local buckets are stale; pid set must be reloaded.
+     * Shared mappings survive fork and stay valid. */
+    for (int i = 0; i < MAX_DEVICE_COUNT; i++) {
+        g_state.dev[i].tokens = 0;
+        g_state.dev[i].gap_start = NULL;
+        g_state.dev[i].gap_stop = NULL;
+        pthread_mutex_init(&g_state.dev[i].gap_mu, NULL);
+    }
+    vgpu_load_pid_set(&g_state.pids);
+}
+
+extern void vgpu_register_client(void); /* register.c */
+
+static void do_init(void) {
+    char buf[64];
+    const char *v = vgpu_getenv("DISABLE_VGPU_CONTROL", buf, sizeof(buf));
+    if (v && (*v == '1' || strcasecmp(v, "true") == 0)) {
+        g_state.disabled = 1;
+        if (load_real_hip() != 0) return;
+        g_init_rc = 0;
+        LOGGER(LOG_INFO, "vgpu control DISABLED by env");
+        return;
+    }
+
+    if (load_real_hip() != 0) return;
+
+    /* 1. vgpu.config (shared) or env bootstrap (private).
+     * Paths are overridable by env so a single box can host several
+     * fake "containers" in tests (production uses the fixed mounts). */
+    char pbuf[512];
+    const char *cfg_path = vgpu_getenv("VGPU_CONFIG_PATH_OVERRIDE", pbuf,
+                                       sizeof(pbuf));
+    if (!cfg_path) cfg_path = VGPU_CONFIG_PATH;
+    resource_data_t *cfg = vgpu_region_attach(
+        cfg_path, sizeof(resource_data_t), VGPU_CFG_MAGIC,
+        /*create=*/false, NULL);
+    if (cfg) {
+        g_state.cfg = cfg;
+        g_state.cfg_shared = true;
+    } else {
+        g_state.cfg = calloc(1, sizeof(resource_data_t));
+        env_bootstrap_config(g_state.cfg);
+        g_state.cfg_shared = false;
+    }
+
+    /* 2. container pid set */
+    vgpu_register_client(); /* client mode: best-effort registration    */
+    vgpu_load_pid_set(&g_state.pids);
+
+    /* 3. vmem ledger (shared across container processes when the node
+     *    agent mounted /tmp/.vmem_node; else private) */
+    vmem_region_t *vm = NULL;
+    char vbuf[512];
+    const char *vmem_path = vgpu_getenv("VGPU_VMEM_PATH_OVERRIDE", vbuf,
+                                        sizeof(vbuf));
+    if (vmem_path)
+        vm = vgpu_region_attach(vmem_path, sizeof(vmem_region_t),
+                                VGPU_VMEM_MAGIC, true, NULL);
+    else if (access(VGPU_VMEM_DIR, W_OK) == 0)
+        vm = vgpu_region_attach(VGPU_VMEM_PATH, sizeof(vmem_region_t),
+                                VGPU_VMEM_MAGIC, true, NULL);
+    if (vm) {
+        g_state.vmem = vm;
+        g_state.vmem_shared = true;
+        if (vm->created_ns == 0) {
+            uint64_t z = 0;
+            __atomic_compare_exchange_n(&vm->created_ns, &z, real_ns(), false,
+                                        __ATOMIC_ACQ_REL, __ATOMIC_RELAXED);
+            vm->record_cap = MAX_VMEM_RECORDS;
+        }
+    } else {
+        g_state.vmem = calloc(1, sizeof(vmem_region_t));
+        g_state.vmem->record_cap = MAX_VMEM_RECORDS;
+        g_state.vmem_shared = false;
+    }
+
+    /* 4. shared token bucket (optional) */
+    char sbuf[512];
+    const char *smn_path = vgpu_getenv("VGPU_SM_NODE_PATH_OVERRIDE", sbuf,
+                                       sizeof(sbuf));
+    if (vgpu_dynconfig()->shared_bucket &&
+        (smn_path || access(VGPU_SM_NODE_DIR, W_OK) == 0)) {
+        g_state.sm_node = vgpu_region_attach(
+            smn_path ? smn_path : VGPU_SM_NODE_PATH,
+            sizeof(sm_node_region_t), VGPU_SMND_MAGIC, true, NULL);
+        if (g_state.sm_node && g_state.sm_node->created_ns == 0) {
+            uint64_t z = 0;
+            __atomic_compare_exchange_n(&g_state.sm_node->created_ns, &z,
+                                        real_ns(), false, __ATOMIC_ACQ_REL,
+                                        __ATOMIC_RELAXED);
+        }
+    }
+
+    /* 5. external utilization watcher region (read-only, optional) */
+    char ubuf[512];
+    const char *util_path = vgpu_getenv("VGPU_UTIL_PATH_OVERRIDE", ubuf,
+                                        sizeof(ubuf));
+    g_state.util = vgpu_region_attach(util_path ? util_path : VGPU_UTIL_PATH,
+                                      sizeof(util_region_t),
+                                      VGPU_UTIL_MAGIC, false, NULL);
+
+    /* 6. visible device count (container view) */
+    int n = 0;
+    if (real_hip.hipGetDeviceCount &&
+        real_hip.hipGetDeviceCount(&n) == hipSuccess)
+        g_state.device_count = n > MAX_DEVICE_COUNT ? MAX_DEVICE_COUNT : n;
+
+    for (int i = 0; i < MAX_DEVICE_COUNT; i++)
+        pthread_mutex_init(&g_state.dev[i].gap_mu, NULL);
+
+    pthread_atfork(NULL, NULL, fork_child_handler);
+
+    g_state.initialized = 1;
+    g_init_rc = 0;
+    LOGGER(LOG_INFO,
+           "vgpu-control init: cfg=%s devices=%d vmem=%s sm_node=%s util=%s",
+           g_state.cfg_shared ? "shared" : "env", g_state.cfg->device_count,
+           g_state.vmem_shared ? "shared" : "private",
+           g_state.sm_node ? "shared" : "off", g_state.util ? "on" : "off");
+}
+
+int vgpu_ensure_init(void) {
+    pthread_once(&g_init_once, do_init);
+    return g_init_rc;
+}
+
+/* ------------------------------------------------------------------ */
+/* allocation registry: process-local ptr -> {size, kind, dev, vmem}   */
+/* open-addressing hash; grows never (fixed 1<<16 slots ~ 64k allocs)  */
+/* ------------------------------------------------------------------ */
+#define REG_BITS 16
+#define REG_SLOTS (1u << REG_BITS)
+
+typedef struct {
+    uint64_t ptr;   /* 0 = empty, 1 = tombstone */
+    uint64_t size;
+    int32_t kind;
+    int32_t dev;
+    int32_t vmem_idx;
+    uint32_t _pad;
+} reg_entry_t;
+
+static reg_entry_t g_reg[REG_SLOTS];
+static pthread_mutex_t g_reg_mu = PTHREAD_MUTEX_INITIALIZER;
+static uint64_t g_reg_dev_total[MAX_DEVICE_COUNT];
+
+static inline uint32_t reg_hash(uint64_t p) {
+    p ^= p >> 33;
+    p *= 0xff51afd7ed558ccdULL;
+    p ^= p >> 33;
+    return (uint32_t)p & (REG_SLOTS - 1);
+}
+
+int alloc_registry_add(void *ptr, size_t size, int kind, int dev,
+                       int vmem_idx) {
+    uint64_t p = (uint64_t)(uintptr_t)ptr;
+    if (!p) return -1;
+    pthread_mutex_lock(&g_reg_mu);
+    uint32_t i = reg_hash(p);
+    for (uint32_t probe = 0; probe < REG_SLOTS; probe++, i = (i + 1) & (REG_SLOTS - 1)) {
+        if (g_reg[i].ptr == 0 || g_reg[i].ptr == 1 || g_reg[i].ptr == p) {
+            g_reg[i].ptr = p;
+            g_reg[i].size = size;
+            g_reg[i].kind = kind;
+            g_reg[i].dev = dev;
+            g_reg[i].vmem_idx = vmem_idx;
+            if (dev >= 0 && dev < MAX_DEVICE_COUNT)
+                g_reg_dev_total[dev] += size;
+            pthread_mutex_unlock(&g_reg_mu);
+            return (int)i;
+        }
+    }
+    pthread_mutex_unlock(&g_reg_mu);
+    LOGGER(LOG_WARN, "allocation registry full");
+    return -1;
+}
+
+bool alloc_registry_remove(void *ptr, size_t *size, int *kind, int *dev,
+                           int *vmem_idx) {
+    uint64_t p = (uint64_t)(uintptr_t)ptr;
+    if (!p) return false;
+    pthread_mutex_lock(&g_reg_mu);
+    uint32_t i = reg_hash(p);
+    for (uint32_t probe = 0; probe < REG_SLOTS; probe++, i = (i + 1) & (REG_SLOTS - 1)) {
+        if (g_reg[i].ptr == 0) break;
+        if (g_reg[i].ptr == p) {
+            if (size) *size = g_reg[i].size;
+            if (kind) *kind = g_reg[i].kind;
+            if (dev) *dev = g_reg[i].dev;
+            if (vmem_idx) *vmem_idx = g_reg[i].vmem_idx;
+            if (g_reg[i].dev >= 0 && g_reg[i].dev < MAX_DEVICE_COUNT)
+                g_reg_dev_total[g_reg[i].dev] -= g_reg[i].size;
+            g_reg[i].ptr = 1; /* tombstone */
+            pthread_mutex_unlock(&g_reg_mu);
+            return true;
+        }
+    }
+    pthread_mutex_unlock(&g_reg_mu);
+    return false;
+}
+
+uint64_t alloc_registry_total(int dev) {
+    if (dev < 0 || dev >= MAX_DEVICE_COUNT) return 0;
+    pthread_mutex_lock(&g_reg_mu);
+    uint64_t t = g_reg_dev_total[dev];
+    pthread_mutex_unlock(&g_reg_mu);
+    return t;
+}
+
+/* ------------------------------------------------------------------ */
+/* vmem ledger ops (shared CAS slots; see hook.h)                      */
+/* ------------------------------------------------------------------ */
+int vmem_ledger_add(int dev, uint64_t dptr, uint64_t size, int kind) {
+    vmem_region_t *r = g_state.vmem;
+    for (uint32_t i = 0; i < MAX_VMEM_RECORDS; i++) {
+        uint32_t st = VMEM_STATE_FREE;
+        if (__atomic_compare_exchange_n(&r->records[i].state, &st,
+                                        VMEM_STATE_BUSY, true,
+                                        __ATOMIC_ACQ_REL, __ATOMIC_RELAXED)) {
+            vmem_record_t *rec = &r->records[i];
+            rec->kind = (uint32_t)kind;
+            rec->dptr = dptr;
+            rec->size = size;
+            rec->pid = (int32_t)getpid();
+            rec->device = dev;
+            rec->created_ns = mono_ns();
+            __atomic_fetch_add(&r->counters[dev].vmem_used, size,
+                               __ATOMIC_ACQ_REL);
+            __atomic_store_n(&rec->state, VMEM_STATE_LIVE, __ATOMIC_RELEASE);
+            return (int)i;
+        }
+    }
+    LOGGER(LOG_WARN, "vmem ledger full");
+    return -1;
+}
+
+void vmem_ledger_remove(int idx) {
+    if (idx < 0 || idx >= (int)MAX_VMEM_RECORDS) return;
+    vmem_region_t *r = g_state.vmem;
+    vmem_record_t *rec = &r->records[idx];
+    uint32_t st = VMEM_STATE_LIVE;
+    if (!__atomic_compare_exchange_n(&rec->state, &st, VMEM_STATE_BUSY, true,
+                                     __ATOMIC_ACQ_REL, __ATOMIC_RELAXED))
+        return;
+    __atomic_fetch_sub(&r->counters[rec->device].vmem_used, rec->size,
+                       __ATOMIC_ACQ_REL);
+    __atomic_store_n(&rec->state, VMEM_STATE_FREE, __ATOMIC_RELEASE);
+}
+
+uint64_t vmem_ledger_used(int dev) {
+    if (dev < 0 || dev >= MAX_DEVICE_COUNT) return 0;
+    return __atomic_load_n(&g_state.vmem->counters[dev].vmem_used,
+                           __ATOMIC_ACQUIRE);
+}
+
+void dev_hooked_add(int dev, int64_t delta) {
+    if (dev < 0 || dev >= MAX_DEVICE_COUNT) return;
+    __atomic_fetch_add(&g_state.vmem->counters[dev].dev_hooked_used,
+                       (uint64_t)delta, __ATOMIC_ACQ_REL);
+}
+
+uint64_t dev_hooked_used(int dev) {
+    if (dev < 0 || dev >= MAX_DEVICE_COUNT) return 0;
+    return __atomic_load_n(&g_state.vmem->counters[dev].dev_hooked_used,
+                           __ATOMIC_ACQUIRE);
+}

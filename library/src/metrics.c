@@ -1,0 +1,26 @@
+/* metrics.c — power-of-two-logged counters (reference metrics.c idea). */
+#include "metrics.h"
+#include "shm.h"
+
+static const char *g_names[MET_COUNT] = {
+    "oom",          "uva_fallback",   "rate_limit_sleep", "gap_sleep",
+    "watcher_miss", "aimd_md",        "aimd_ai",          "excl_flip",
+    "refill_takeover", "lock_wait",
+};
+
+static uint64_t g_counters[MET_COUNT];
+
+void metrics_inc(int which) {
+    if (which < 0 || which >= MET_COUNT) return;
+    uint64_t v =
+        __atomic_add_fetch(&g_counters[which], 1, __ATOMIC_RELAXED);
+    /* log at 1,2,4,8,... so the signal scales with the noise          */
+    if ((v & (v - 1)) == 0)
+        LOGGER(LOG_INFO, "metric %s=%llu", g_names[which],
+               (unsigned long long)v);
+}
+
+uint64_t metrics_get(int which) {
+    if (which < 0 || which >= MET_COUNT) return 0;
+    return __atomic_load_n(&g_counters[which], __ATOMIC_RELAXED);
+}

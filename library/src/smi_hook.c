@@ -1,0 +1,318 @@
+/* smi_hook.c — amd-smi / rocm-smi integration.
+ *
+ * Two roles (reference nvml_hook.c + the NVML sampling half of
+ * cuda_hook.c, re-designed for AMD-SMI):
+ *  1. SAMPLING: the utilization watcher queries whole-GPU gfx activity
+ *     and the per-process gfx engine time / VRAM of the container's
+ *     pids (amdsmi_get_gpu_process_list), via a lazily dlopen'd
+ *     libamd_smi — the library itself never links it.
+ *  2. SPOOFING: in-container `amd-smi`/`rocm-smi` see the *quota* view:
+ *     memory-total returns the container quota, memory-usage returns
+ *     the container's accounted usage.  Interposed both at link time
+ *     and through the dlsym hook (ctypes users).
+ */
+#define _GNU_SOURCE
+#include "state.h"
+#include "shm.h"
+
+#include <amd_smi/amdsmi.h>
+#include <dlfcn.h>
+#include <pthread.h>
+#include <stdlib.h>
+#include <string.h>
+
+#define EXPORT __attribute__((visibility("default")))
+
+/* ---- real libamd_smi table ---- */
+typedef struct {
+    amdsmi_status_t (*amdsmi_init)(uint64_t);
+    amdsmi_status_t (*amdsmi_get_socket_handles)(uint32_t *,
+                                                 amdsmi_socket_handle *);
+    amdsmi_status_t (*amdsmi_get_processor_handles)(
+        amdsmi_socket_handle, uint32_t *, amdsmi_processor_handle *);
+    amdsmi_status_t (*amdsmi_get_gpu_activity)(amdsmi_processor_handle,
+                                               amdsmi_engine_usage_t *);
+    amdsmi_status_t (*amdsmi_get_gpu_process_list)(amdsmi_processor_handle,
+                                                   uint32_t *,
+                                                   amdsmi_proc_info_t *);
+    amdsmi_status_t (*amdsmi_get_gpu_memory_total)(amdsmi_processor_handle,
+                                                   amdsmi_memory_type_t,
+                                                   uint64_t *);
+    amdsmi_status_t (*amdsmi_get_gpu_memory_usage)(amdsmi_processor_handle,
+                                                   amdsmi_memory_type_t,
+                                                   uint64_t *);
+    amdsmi_status_t (*amdsmi_get_gpu_vram_usage)(amdsmi_processor_handle,
+                                                 amdsmi_vram_usage_t *);
+} smi_real_t;
+
+static smi_real_t real_smi;
+static void *g_smi_handle;
+static int g_smi_ok = -1; /* -1 unknown, 0 no, 1 yes */
+static pthread_mutex_t g_smi_mu = PTHREAD_MUTEX_INITIALIZER;
+
+#define MAX_SMI_DEVS 64
+static amdsmi_processor_handle g_smi_handles[MAX_SMI_DEVS];
+static int g_smi_count;
+
+static int smi_load_locked(void) {
+    if (g_smi_ok >= 0) return g_smi_ok;
+    g_smi_ok = 0;
+    const char *paths[] = {getenv("VGPU_REAL_SMI_PATH"), "libamd_smi.so.26",
+                           "libamd_smi.so", "/opt/rocm/lib/libamd_smi.so"};
+    for (size_t i = 0; i < sizeof(paths) / sizeof(paths[0]); i++) {
+        if (!paths[i]) continue;
+        g_smi_handle = dlopen(paths[i], RTLD_LAZY | RTLD_LOCAL);
+        if (g_smi_handle) break;
+    }
+    if (!g_smi_handle) {
+        LOGGER(LOG_DEBUG, "libamd_smi unavailable: %s", dlerror());
+        return 0;
+    }
+#define LOADS(sym) \
+    real_smi.sym = (__typeof__(real_smi.sym))vgpu_real_dlsym(g_smi_handle, #sym)
+    LOADS(amdsmi_init);
+    LOADS(amdsmi_get_socket_handles);
+    LOADS(amdsmi_get_processor_handles);
+    LOADS(amdsmi_get_gpu_activity);
+    LOADS(amdsmi_get_gpu_process_list);
+    LOADS(amdsmi_get_gpu_memory_total);
+    LOADS(amdsmi_get_gpu_memory_usage);
+    LOADS(amdsmi_get_gpu_vram_usage);
+#undef LOADS
+    if (!real_smi.amdsmi_init || !real_smi.amdsmi_get_socket_handles)
+        return 0;
+    if (real_smi.amdsmi_init(AMDSMI_INIT_AMD_GPUS) != AMDSMI_STATUS_SUCCESS)
+        return 0;
+    /* enumerate processor handles in socket order; positional index
+     * equals the container-local device index (same drm visibility)   */
+    uint32_t nsock = 0;
+    if (real_smi.amdsmi_get_socket_handles(&nsock, NULL) !=
+            AMDSMI_STATUS_SUCCESS || nsock == 0)
+        return 0;
+    amdsmi_socket_handle socks[MAX_SMI_DEVS];
+    if (nsock > MAX_SMI_DEVS) nsock = MAX_SMI_DEVS;
+    if (real_smi.amdsmi_get_socket_handles(&nsock, socks) !=
+        AMDSMI_STATUS_SUCCESS)
+        return 0;
+    for (uint32_t s = 0; s < nsock && g_smi_count < MAX_SMI_DEVS; s++) {
+        uint32_t np = 0;
+        if (real_smi.amdsmi_get_processor_handles(socks[s], &np, NULL) !=
+                AMDSMI_STATUS_SUCCESS || np == 0)
+            continue;
+        amdsmi_processor_handle procs[MAX_SMI_DEVS];
+        if (np > MAX_SMI_DEVS) np = MAX_SMI_DEVS;
+        if (real_smi.amdsmi_get_processor_handles(socks[s], &np, procs) !=
+            AMDSMI_STATUS_SUCCESS)
+            continue;
+        for (uint32_t p = 0; p < np && g_smi_count < MAX_SMI_DEVS; p++)
+            g_smi_handles[g_smi_count++] = procs[p];
+    }
+    LOGGER(LOG_INFO, "amd-smi sampler: %d devices", g_smi_count);
+    g_smi_ok = g_smi_count > 0;
+    return g_smi_ok;
+}
+
+bool smi_available(void) {
+    pthread_mutex_lock(&g_smi_mu);
+    int ok = smi_load_locked();
+    pthread_mutex_unlock(&g_smi_mu);
+    return ok == 1;
+}
+
+static amdsmi_processor_handle handle_for(int dev) {
+    if (dev < 0 || dev >= g_smi_count) return NULL;
+    return g_smi_handles[dev];
+}
+
+bool smi_sample_device(int dev, uint32_t *busy_permille,
+                       uint64_t *container_gfx_ns, uint64_t *container_vram,
+                       const pid_set_t *pids) {
+    if (!smi_available()) return false;
+    amdsmi_processor_handle h = handle_for(dev);
+    if (!h) return false;
+    amdsmi_engine_usage_t eng;
+    memset(&eng, 0, sizeof(eng));
+    if (real_smi.amdsmi_get_gpu_activity(h, &eng) != AMDSMI_STATUS_SUCCESS)
+        return false;
+    *busy_permille = eng.gfx_activity * 10;
+    uint64_t gfx = 0, vram = 0;
+    uint32_t n = 128;
+    amdsmi_proc_info_t list[128];
+    memset(list, 0, sizeof(list));
+    amdsmi_status_t st = real_smi.amdsmi_get_gpu_process_list(h, &n, list);
+    if (st == AMDSMI_STATUS_SUCCESS || st == AMDSMI_STATUS_OUT_OF_RESOURCES) {
+        if (n > 128) n = 128;
+        for (uint32_t i = 0; i < n; i++) {
+            if (!vgpu_pid_set_contains(pids, (int32_t)list[i].pid)) continue;
+            gfx += list[i].engine_usage.gfx;
+            vram += list[i].memory_usage.vram_mem;
+        }
+    }
+    *container_gfx_ns = gfx;
+    *container_vram = vram;
+    return true;
+}
+
+uint64_t smi_container_vram(int dev, const pid_set_t *pids) {
+    if (!smi_available()) return 0;
+    amdsmi_processor_handle h = handle_for(dev);
+    if (!h) return 0;
+    uint64_t vram = 0;
+    uint32_t n = 128;
+    amdsmi_proc_info_t list[128];
+    memset(list, 0, sizeof(list));
+    amdsmi_status_t st = real_smi.amdsmi_get_gpu_process_list(h, &n, list);
+    if (st == AMDSMI_STATUS_SUCCESS || st == AMDSMI_STATUS_OUT_OF_RESOURCES) {
+        if (n > 128) n = 128;
+        for (uint32_t i = 0; i < n; i++)
+            if (vgpu_pid_set_contains(pids, (int32_t)list[i].pid))
+                vram += list[i].memory_usage.vram_mem;
+    }
+    return vram;
+}
+
+/* ------------------------------------------------------------------ */
+/* spoofing exports                                                    */
+/* ------------------------------------------------------------------ */
+
+/* handle -> container device index via the recorded enumeration; the
+ * caller's handles come from the SAME libamd_smi enumeration (we pass
+ * through amdsmi_get_processor_handles), so pointer identity holds.   */
+static int index_for_handle(amdsmi_processor_handle h) {
+    for (int i = 0; i < g_smi_count; i++)
+        if (g_smi_handles[i] == h) return i;
+    return -1;
+}
+
+static int spoof_dev(amdsmi_processor_handle h, device_t *snap) {
+    if (vgpu_ensure_init() != 0 || g_state.disabled) return -1;
+    if (!smi_available()) return -1;
+    int dev = index_for_handle(h);
+    if (dev < 0 || dev >= g_state.cfg->device_count) return -1;
+    if (!(vgpu_device_flags(dev) & DEV_FLAG_MEM_LIMIT)) return -1;
+    vgpu_device_snapshot(dev, snap);
+    return dev;
+}
+
+EXPORT amdsmi_status_t amdsmi_get_gpu_memory_total(
+    amdsmi_processor_handle h, amdsmi_memory_type_t type, uint64_t *total) {
+    if (!smi_available() || !real_smi.amdsmi_get_gpu_memory_total)
+        return AMDSMI_STATUS_NOT_INIT;
+    amdsmi_status_t st = real_smi.amdsmi_get_gpu_memory_total(h, type, total);
+    if (st != AMDSMI_STATUS_SUCCESS) return st;
+    device_t snap;
+    if ((type == AMDSMI_MEM_TYPE_VRAM || type == AMDSMI_MEM_TYPE_VIS_VRAM) &&
+        spoof_dev(h, &snap) >= 0)
+        *total = snap.total_memory;
+    return st;
+}
+
+EXPORT amdsmi_status_t amdsmi_get_gpu_memory_usage(
+    amdsmi_processor_handle h, amdsmi_memory_type_t type, uint64_t *used) {
+    if (!smi_available() || !real_smi.amdsmi_get_gpu_memory_usage)
+        return AMDSMI_STATUS_NOT_INIT;
+    amdsmi_status_t st = real_smi.amdsmi_get_gpu_memory_usage(h, type, used);
+    if (st != AMDSMI_STATUS_SUCCESS) return st;
+    device_t snap;
+    int dev;
+    if ((type == AMDSMI_MEM_TYPE_VRAM || type == AMDSMI_MEM_TYPE_VIS_VRAM) &&
+        (dev = spoof_dev(h, &snap)) >= 0)
+        *used = vgpu_account_used(dev);
+    return st;
+}
+
+EXPORT amdsmi_status_t amdsmi_get_gpu_vram_usage(amdsmi_processor_handle h,
+                                                 amdsmi_vram_usage_t *info) {
+    if (!smi_available() || !real_smi.amdsmi_get_gpu_vram_usage)
+        return AMDSMI_STATUS_NOT_INIT;
+    amdsmi_status_t st = real_smi.amdsmi_get_gpu_vram_usage(h, info);
+    if (st != AMDSMI_STATUS_SUCCESS) return st;
+    device_t snap;
+    int dev;
+    if ((dev = spoof_dev(h, &snap)) >= 0) {
+        info->vram_total = (uint32_t)(snap.total_memory >> 20);
+        info->vram_used = (uint32_t)(vgpu_account_used(dev) >> 20);
+    }
+    return st;
+}
+
+/* rocm-smi (rsmi) spoofs: index-addressed                             */
+typedef int rsmi_status_t; /* 0 == RSMI_STATUS_SUCCESS                 */
+typedef int rsmi_memory_type_t; /* 0 == RSMI_MEM_TYPE_VRAM             */
+
+static void *g_rsmi_handle;
+static rsmi_status_t (*real_rsmi_total)(uint32_t, rsmi_memory_type_t,
+                                        uint64_t *);
+static rsmi_status_t (*real_rsmi_usage)(uint32_t, rsmi_memory_type_t,
+                                        uint64_t *);
+
+static int rsmi_load(void) {
+    static int ok = -1;
+    if (ok >= 0) return ok;
+    ok = 0;
+    const char *paths[] = {"librocm_smi64.so.7", "librocm_smi64.so",
+                           "/opt/rocm/lib/librocm_smi64.so"};
+    for (size_t i = 0; i < sizeof(paths) / sizeof(paths[0]); i++) {
+        g_rsmi_handle = dlopen(paths[i], RTLD_LAZY | RTLD_LOCAL);
+        if (g_rsmi_handle) break;
+    }
+    if (!g_rsmi_handle) return 0;
+    real_rsmi_total = (__typeof__(real_rsmi_total))vgpu_real_dlsym(
+        g_rsmi_handle, "rsmi_dev_memory_total_get");
+    real_rsmi_usage = (__typeof__(real_rsmi_usage))vgpu_real_dlsym(
+        g_rsmi_handle, "rsmi_dev_memory_usage_get");
+    ok = real_rsmi_total && real_rsmi_usage;
+    return ok;
+}
+
+static int rsmi_spoof_dev(uint32_t dv_ind, device_t *snap) {
+    if (vgpu_ensure_init() != 0 || g_state.disabled) return -1;
+    if ((int)dv_ind >= g_state.cfg->device_count) return -1;
+    if (!(vgpu_device_flags((int)dv_ind) & DEV_FLAG_MEM_LIMIT)) return -1;
+    vgpu_device_snapshot((int)dv_ind, snap);
+    return (int)dv_ind;
+}
+
+EXPORT rsmi_status_t rsmi_dev_memory_total_get(uint32_t dv_ind,
+                                               rsmi_memory_type_t type,
+                                               uint64_t *total) {
+    if (!rsmi_load()) return 1;
+    rsmi_status_t st = real_rsmi_total(dv_ind, type, total);
+    device_t snap;
+    if (st == 0 && type == 0 && rsmi_spoof_dev(dv_ind, &snap) >= 0)
+        *total = snap.total_memory;
+    return st;
+}
+
+EXPORT rsmi_status_t rsmi_dev_memory_usage_get(uint32_t dv_ind,
+                                               rsmi_memory_type_t type,
+                                               uint64_t *used) {
+    if (!rsmi_load()) return 1;
+    rsmi_status_t st = real_rsmi_usage(dv_ind, type, used);
+    device_t snap;
+    int dev;
+    if (st == 0 && type == 0 && (dev = rsmi_spoof_dev(dv_ind, &snap)) >= 0)
+        *used = vgpu_account_used(dev);
+    return st;
+}
+
+/* ---- dlsym routing table for the smi family ---- */
+typedef struct {
+    const char *name;
+    void *fn;
+} smi_hook_entry_t;
+
+static const smi_hook_entry_t g_smi_hooks[] = {
+    {"amdsmi_get_gpu_memory_total", (void *)amdsmi_get_gpu_memory_total},
+    {"amdsmi_get_gpu_memory_usage", (void *)amdsmi_get_gpu_memory_usage},
+    {"amdsmi_get_gpu_vram_usage", (void *)amdsmi_get_gpu_vram_usage},
+    {"rsmi_dev_memory_total_get", (void *)rsmi_dev_memory_total_get},
+    {"rsmi_dev_memory_usage_get", (void *)rsmi_dev_memory_usage_get},
+    {NULL, NULL},
+};
+
+void *vgpu_smi_lookup_hook(const char *name) {
+    for (const smi_hook_entry_t *e = g_smi_hooks; e->name; e++)
+        if (strcmp(e->name, name) == 0) return e->fn;
+    return NULL;
+}

@@ -1,0 +1,152 @@
+/* hip_stub.c — a malloc-backed fake libamdhip64 for CPU-only testing.
+ *
+ * Built as libamdhip64.so.7 (stub soname); tests point the shim at it
+ * with VGPU_REAL_HIP_PATH and link against it directly, so the whole
+ * interception path (link-time interposition -> shim -> "real" lib)
+ * runs exactly as on a GPU box, minus the GPU.
+ *
+ * Counters are exported (stub_*) so tests can assert which real entry
+ * points were reached behind the shim.
+ */
+#define __HIP_PLATFORM_AMD__ 1
+#include <hip/hip_runtime_api.h>
+
+#include <stdint.h>
+#include <stdlib.h>
+#include <string.h>
+
+#define EXPORT __attribute__((visibility("default")))
+
+static int g_device = 0;
+
+/* observable counters */
+static uint64_t c_malloc, c_free, c_managed, c_launch, c_async;
+EXPORT uint64_t stub_count_malloc(void) { return c_malloc; }
+EXPORT uint64_t stub_count_free(void) { return c_free; }
+EXPORT uint64_t stub_count_managed(void) { return c_managed; }
+EXPORT uint64_t stub_count_launch(void) { return c_launch; }
+EXPORT uint64_t stub_count_async(void) { return c_async; }
+
+EXPORT hipError_t hipGetDeviceCount(int *n) {
+    *n = 2;
+    return hipSuccess;
+}
+EXPORT hipError_t hipSetDevice(int d) {
+    g_device = d;
+    return hipSuccess;
+}
+EXPORT hipError_t hipGetDevice(int *d) {
+    *d = g_device;
+    return hipSuccess;
+}
+EXPORT hipError_t hipDeviceGetAttribute(int *v, hipDeviceAttribute_t a,
+                                        int dev) {
+    (void)dev;
+    if (a == hipDeviceAttributeMultiprocessorCount) *v = 256;
+    else if (a == hipDeviceAttributeMaxThreadsPerMultiProcessor) *v = 2048;
+    else *v = 0;
+    return hipSuccess;
+}
+
+EXPORT hipError_t hipMalloc(void **p, size_t sz) {
+    *p = malloc(sz);
+    __atomic_fetch_add(&c_malloc, 1, __ATOMIC_RELAXED);
+    return *p ? hipSuccess : hipErrorOutOfMemory;
+}
+EXPORT hipError_t hipExtMallocWithFlags(void **p, size_t sz,
+                                        unsigned int flags) {
+    (void)flags;
+    return hipMalloc(p, sz);
+}
+EXPORT hipError_t hipMallocManaged(void **p, size_t sz, unsigned int f) {
+    (void)f;
+    *p = malloc(sz);
+    __atomic_fetch_add(&c_managed, 1, __ATOMIC_RELAXED);
+    return *p ? hipSuccess : hipErrorOutOfMemory;
+}
+EXPORT hipError_t hipMallocAsync(void **p, size_t sz, hipStream_t s) {
+    (void)s;
+    __atomic_fetch_add(&c_async, 1, __ATOMIC_RELAXED);
+    *p = malloc(sz);
+    return *p ? hipSuccess : hipErrorOutOfMemory;
+}
+EXPORT hipError_t hipFree(void *p) {
+    free(p);
+    __atomic_fetch_add(&c_free, 1, __ATOMIC_RELAXED);
+    return hipSuccess;
+}
+EXPORT hipError_t hipFreeAsync(void *p, hipStream_t s) {
+    (void)s;
+    return hipFree(p);
+}
+EXPORT hipError_t hipMemGetInfo(size_t *fr, size_t *total) {
+    if (fr) *fr = 200ull << 30;
+    if (total) *total = 288ull << 30;
+    return hipSuccess;
+}
+EXPORT hipError_t hipDeviceTotalMem(size_t *b, hipDevice_t d) {
+    (void)d;
+    *b = 288ull << 30;
+    return hipSuccess;
+}
+EXPORT hipError_t hipGetDevicePropertiesR0600(hipDeviceProp_tR0600 *p,
+                                              int dev) {
+    (void)dev;
+    memset(p, 0, sizeof(*p));
+    strcpy(p->name, "stub-MI355X");
+    p->totalGlobalMem = 288ull << 30;
+    return hipSuccess;
+}
+EXPORT hipError_t hipMemAdvise(const void *p, size_t n, hipMemoryAdvise a,
+                               int d) {
+    (void)p; (void)n; (void)a; (void)d;
+    return hipSuccess;
+}
+EXPORT hipError_t hipMemPrefetchAsync(const void *p, size_t n, int d,
+                                      hipStream_t s) {
+    (void)p; (void)n; (void)d; (void)s;
+    return hipSuccess;
+}
+EXPORT hipError_t hipLaunchKernel(const void *f, dim3 g, dim3 b, void **a,
+                                  size_t shm, hipStream_t s) {
+    (void)f; (void)g; (void)b; (void)a; (void)shm; (void)s;
+    __atomic_fetch_add(&c_launch, 1, __ATOMIC_RELAXED);
+    return hipSuccess;
+}
+EXPORT hipError_t hipModuleLaunchKernel(hipFunction_t f, unsigned gx,
+                                        unsigned gy, unsigned gz, unsigned bx,
+                                        unsigned by, unsigned bz, unsigned shm,
+                                        hipStream_t s, void **kp, void **ex) {
+    (void)f; (void)gx; (void)gy; (void)gz; (void)bx; (void)by; (void)bz;
+    (void)shm; (void)s; (void)kp; (void)ex;
+    __atomic_fetch_add(&c_launch, 1, __ATOMIC_RELAXED);
+    return hipSuccess;
+}
+EXPORT hipError_t hipEventCreateWithFlags(hipEvent_t *e, unsigned f) {
+    (void)f;
+    *e = (hipEvent_t)malloc(8);
+    return hipSuccess;
+}
+EXPORT hipError_t hipEventRecord(hipEvent_t e, hipStream_t s) {
+    (void)e; (void)s;
+    return hipSuccess;
+}
+EXPORT hipError_t hipEventSynchronize(hipEvent_t e) {
+    (void)e;
+    return hipSuccess;
+}
+EXPORT hipError_t hipEventElapsedTime(float *ms, hipEvent_t a, hipEvent_t b) {
+    (void)a; (void)b;
+    *ms = 0.f;
+    return hipSuccess;
+}
+EXPORT hipError_t hipEventDestroy(hipEvent_t e) {
+    free((void *)e);
+    return hipSuccess;
+}
+EXPORT hipError_t hipStreamIsCapturing(hipStream_t s,
+                                       hipStreamCaptureStatus *st) {
+    (void)s;
+    *st = hipStreamCaptureStatusNone;
+    return hipSuccess;
+}

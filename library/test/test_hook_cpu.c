@@ -1,0 +1,143 @@
+/* test_hook_cpu.c — end-to-end hook behavior against the stub runtime.
+ *
+ * Linked against the stub libamdhip64; run under LD_PRELOAD of the
+ * shim with VGPU_REAL_HIP_PATH pointing at the stub.  Covers:
+ *   - quota gate: allocs within quota pass, over quota -> OOM
+ *   - free retires the charge; realloc then passes
+ *   - hipMemGetInfo / hipDeviceTotalMem / props quota spoofing
+ *   - oversold mode routes over-quota allocs to hipMallocManaged
+ *   - launch path passes through (and is counted by the stub)
+ * Scenario selected by argv[1]; exit 0 = pass.
+ */
+#define __HIP_PLATFORM_AMD__ 1
+#include <hip/hip_runtime_api.h>
+
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+#include <time.h>
+
+extern uint64_t stub_count_malloc(void);
+extern uint64_t stub_count_managed(void);
+extern uint64_t stub_count_launch(void);
+
+#define CHECK(cond)                                                    \
+    do {                                                               \
+        if (!(cond)) {                                                 \
+            fprintf(stderr, "FAIL %s:%d: %s\n", __FILE__, __LINE__,    \
+                    #cond);                                            \
+            return 1;                                                  \
+        }                                                              \
+    } while (0)
+
+static int scenario_quota(void) {
+    /* env: VGPU_MEM_LIMIT_0=1m */
+    void *a = NULL, *b = NULL, *c = NULL;
+    CHECK(hipMalloc(&a, 512 * 1024) == hipSuccess);        /* 512K ok   */
+    CHECK(hipMalloc(&b, 1024 * 1024) == hipErrorOutOfMemory); /* over   */
+    CHECK(hipMalloc(&b, 400 * 1024) == hipSuccess);        /* 912K ok   */
+    CHECK(hipMalloc(&c, 200 * 1024) == hipErrorOutOfMemory);
+    CHECK(hipFree(a) == hipSuccess);
+    CHECK(hipMalloc(&c, 500 * 1024) == hipSuccess);        /* freed     */
+
+    size_t freeb = 0, total = 0;
+    CHECK(hipMemGetInfo(&freeb, &total) == hipSuccess);
+    CHECK(total == 1024 * 1024);
+    CHECK(freeb == total - 900 * 1024);
+
+    size_t tb = 0;
+    CHECK(hipDeviceTotalMem(&tb, 0) == hipSuccess);
+    CHECK(tb == 1024 * 1024);
+
+    hipDeviceProp_tR0600 prop;
+    CHECK(hipGetDevicePropertiesR0600(&prop, 0) == hipSuccess);
+    CHECK(prop.totalGlobalMem == 1024 * 1024);
+
+    /* device 1 has no limit: passthrough view                          */
+    CHECK(hipSetDevice(1) == hipSuccess);
+    CHECK(hipMalloc(&b, 8 * 1024 * 1024) == hipSuccess);
+    CHECK(hipMemGetInfo(&freeb, &total) == hipSuccess);
+    CHECK(total == 288ull << 30);
+    printf("PASS quota\n");
+    return 0;
+}
+
+static int scenario_oversold(void) {
+    /* env: VGPU_MEM_LIMIT_0=1m VGPU_MEM_OVERSOLD=1 */
+    void *a = NULL, *b = NULL;
+    CHECK(hipMalloc(&a, 900 * 1024) == hipSuccess);
+    uint64_t managed_before = stub_count_managed();
+    CHECK(hipMalloc(&b, 500 * 1024) == hipSuccess); /* spills            */
+    CHECK(stub_count_managed() == managed_before + 1);
+    /* spill must NOT consume device quota                              */
+    size_t freeb = 0, total = 0;
+    CHECK(hipMemGetInfo(&freeb, &total) == hipSuccess);
+    CHECK(total == 1024 * 1024);
+    /* used = 900K device + 500K vmem -> free clamps to 0? no:
+     * vmem counts toward used in the spoofed view                      */
+    CHECK(freeb == 0);
+    CHECK(hipFree(b) == hipSuccess);
+    CHECK(hipMemGetInfo(&freeb, &total) == hipSuccess);
+    CHECK(freeb == total - 900 * 1024);
+    printf("PASS oversold\n");
+    return 0;
+}
+
+static int scenario_launch(void) {
+    /* env: VGPU_CORE_LIMIT_0=50 (throttle on, no GPU busy data)        */
+    dim3 grid = {64, 1, 1}, block = {256, 1, 1};
+    for (int i = 0; i < 2000; i++)
+        CHECK(hipLaunchKernel((void *)scenario_launch, grid, block, NULL, 0,
+                              NULL) == hipSuccess);
+    CHECK(stub_count_launch() == 2000);
+    printf("PASS launch\n");
+    return 0;
+}
+
+static int scenario_nolimit(void) {
+    /* no env limits: everything passes through untouched               */
+    void *a = NULL;
+    CHECK(hipMalloc(&a, 64 * 1024 * 1024) == hipSuccess);
+    size_t freeb, total;
+    CHECK(hipMemGetInfo(&freeb, &total) == hipSuccess);
+    CHECK(total == 288ull << 30);
+    dim3 g = {1, 1, 1}, b = {64, 1, 1};
+    CHECK(hipLaunchKernel((void *)scenario_nolimit, g, b, NULL, 0, NULL) ==
+          hipSuccess);
+    CHECK(hipFree(a) == hipSuccess);
+    printf("PASS nolimit\n");
+    return 0;
+}
+
+static int scenario_throttle(void) {
+    /* env: VGPU_CORE_LIMIT_0=50.  Launch far more grid-blocks than the
+     * initial bucket holds; the rate limiter must stall us across
+     * several watcher refill cycles (>=0.2s wall for this workload)
+     * while still completing.                                         */
+    struct timespec t0, t1;
+    clock_gettime(CLOCK_MONOTONIC, &t0);
+    dim3 grid = {16384, 1, 1}, block = {256, 1, 1};
+    for (int i = 0; i < 200; i++)
+        CHECK(hipLaunchKernel((void *)scenario_throttle, grid, block, NULL,
+                              0, NULL) == hipSuccess);
+    clock_gettime(CLOCK_MONOTONIC, &t1);
+    double el = (double)(t1.tv_sec - t0.tv_sec) +
+                (double)(t1.tv_nsec - t0.tv_nsec) / 1e9;
+    CHECK(stub_count_launch() == 200);
+    CHECK(el >= 0.15); /* bucket depletion forced >=1 refill wait      */
+    printf("PASS throttle (%.2fs for 3.2M grid tokens)\n", el);
+    return 0;
+}
+
+int main(int argc, char **argv) {
+    if (argc < 2) {
+        fprintf(stderr, "usage: %s quota|oversold|launch|nolimit\n", argv[0]);
+        return 2;
+    }
+    if (strcmp(argv[1], "quota") == 0) return scenario_quota();
+    if (strcmp(argv[1], "oversold") == 0) return scenario_oversold();
+    if (strcmp(argv[1], "launch") == 0) return scenario_launch();
+    if (strcmp(argv[1], "throttle") == 0) return scenario_throttle();
+    if (strcmp(argv[1], "nolimit") == 0) return scenario_nolimit();
+    return 2;
+}
