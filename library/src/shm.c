@@ -191,13 +191,31 @@ void vgpu_flock_release(int fd) {
 int lock_gpu_device(int host_index) {
     char path[256];
     snprintf(path, sizeof(path), VGPU_LOCK_DIR "/dev_%d.lock", host_index);
-    /* bounded backoff: the allocation path must not spin unboundedly  */
+    /* fast path: the lock file must open; if the dir is missing create
+     * it once, and if the filesystem is unusable degrade to lock-free
+     * IMMEDIATELY (a missing lock must never tax the alloc path).     */
+    int fd = open(path, O_RDWR | O_CREAT | O_CLOEXEC, 0666);
+    if (fd < 0) {
+        if (errno == ENOENT && mkdir(VGPU_LOCK_DIR, 0777) == 0)
+            fd = open(path, O_RDWR | O_CREAT | O_CLOEXEC, 0666);
+        if (fd < 0) return -1;
+    }
+    struct flock fl = {0};
+    fl.l_type = F_WRLCK;
+    fl.l_whence = SEEK_SET;
+    fl.l_start = 0;
+    fl.l_len = 1;
+    /* two non-blocking tries with short backoff (contention), then
+     * block: bounded added latency under contention, none without.    */
     for (int attempt = 0; attempt < 3; attempt++) {
-        int fd = vgpu_flock_acquire(path, 0, 1, /*wait=*/attempt == 2);
-        if (fd >= 0) return fd;
-        struct timespec ts = {0, 2000000L << attempt}; /* 2,4 ms */
+        int cmd_ofd = attempt == 2 ? F_OFD_SETLKW : F_OFD_SETLK;
+        int cmd_posix = attempt == 2 ? F_SETLKW : F_SETLK;
+        if (ofd_fcntl(fd, cmd_ofd, cmd_posix, &fl) == 0) return fd;
+        if (errno != EAGAIN && errno != EACCES && errno != EINTR) break;
+        struct timespec ts = {0, 500000L << attempt}; /* 0.5, 1 ms     */
         nanosleep(&ts, NULL);
     }
+    close(fd);
     return -1;
 }
 

@@ -27,6 +27,41 @@ EXPORT uint64_t stub_count_managed(void) { return c_managed; }
 EXPORT uint64_t stub_count_launch(void) { return c_launch; }
 EXPORT uint64_t stub_count_async(void) { return c_async; }
 
+/* fatbin registration entry points used by hipcc-compiled objects     */
+EXPORT void **__hipRegisterFatBinary(const void *data) {
+    (void)data;
+    static void *dummy;
+    return &dummy;
+}
+EXPORT void __hipRegisterFunction(void **modules, const void *hostFunction,
+                                  char *deviceFunction,
+                                  const char *deviceName,
+                                  unsigned int threadLimit, void *tid,
+                                  void *bid, void *blockDim, void *gridDim,
+                                  int *wSize) {
+    (void)modules; (void)hostFunction; (void)deviceFunction;
+    (void)deviceName; (void)threadLimit; (void)tid; (void)bid;
+    (void)blockDim; (void)gridDim; (void)wSize;
+}
+EXPORT void __hipUnregisterFatBinary(void **modules) { (void)modules; }
+EXPORT hipError_t __hipPushCallConfiguration(dim3 gridDim, dim3 blockDim,
+                                             size_t sharedMem,
+                                             hipStream_t stream) {
+    (void)gridDim; (void)blockDim; (void)sharedMem; (void)stream;
+    return hipSuccess;
+}
+EXPORT hipError_t __hipPopCallConfiguration(dim3 *gridDim, dim3 *blockDim,
+                                            size_t *sharedMem,
+                                            hipStream_t *stream) {
+    if (gridDim) { gridDim->x = gridDim->y = gridDim->z = 1; }
+    if (blockDim) { blockDim->x = blockDim->y = blockDim->z = 1; }
+    if (sharedMem) *sharedMem = 0;
+    if (stream) *stream = NULL;
+    return hipSuccess;
+}
+EXPORT hipError_t hipGetLastError(void) { return hipSuccess; }
+EXPORT hipError_t hipDeviceSynchronize(void) { return hipSuccess; }
+
 EXPORT hipError_t hipGetDeviceCount(int *n) {
     *n = 2;
     return hipSuccess;
