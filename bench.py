@@ -111,17 +111,20 @@ def inner_bare(args):
 
 
 def inner_pod(args):
-    """A quota'd pod: allocate chunks until OOM, report achieved bytes."""
+    """A quota'd pod: fill to the quota (chunk shrinks on rejection so
+    the achieved figure reflects true enforcement granularity)."""
     wk = load_workload(args.device)
     got = 0
     ptrs = []
-    while got < POD_QUOTA_BYTES * 2:
-        p = wk.wk_malloc(POD_CHUNK)
+    chunk = POD_CHUNK
+    while chunk >= (1 << 20) and got < POD_QUOTA_BYTES * 2:
+        p = wk.wk_malloc(chunk)
         if not p:
-            break
+            chunk //= 2
+            continue
         ptrs.append(p)
-        got += POD_CHUNK
-        wk.wk_touch(ctypes.c_void_p(p), POD_CHUNK // 4)
+        got += chunk
+        wk.wk_touch(ctypes.c_void_p(p), min(chunk, 16 << 20) // 4)
     wk.wk_sync()
     for p in ptrs:
         wk.wk_free(ctypes.c_void_p(p))

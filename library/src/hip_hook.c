@@ -60,7 +60,14 @@ static void dev_hot_init(int dev) {
     h->cu_count = cus;
     h->max_threads_per_cu = thr;
     h->pool = (int64_t)cus * thr * TOKEN_FACTOR;
-    h->cur_share = h->pool / 100; /* conservative start: 1% per cycle  */
+    /* start at a limit-proportional share so a hard limit bites from
+     * the first cycles instead of waiting for MD convergence          */
+    device_t snap;
+    vgpu_device_snapshot(dev, &snap);
+    uint32_t lim = (snap.flags & DEV_FLAG_CORE_LIMIT) && snap.core_limit
+                       ? snap.core_limit : 100;
+    h->cur_share = h->pool * lim / 100 / (1000 / WATCHER_CYCLE_MS) / 4;
+    if (h->cur_share < 1) h->cur_share = 1;
     __atomic_store_n(&h->tokens, h->cur_share, __ATOMIC_RELAXED);
     if (g_state.sm_node) {
         sm_node_dev_t *s = &g_state.sm_node->devices[dev];
