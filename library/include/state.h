@@ -111,7 +111,9 @@ typedef struct {
     pthread_mutex_t gap_mu;
     uint64_t prev_proc_gfx_ns;  /* per-container engine-time sample     */
     uint64_t prev_sample_ns;
-    uint8_t _pad[24];
+    uint64_t launch_count;      /* launches gated on this device        */
+    uint64_t prev_launch_count;
+    uint8_t _pad[8];
 } dev_hot_t;
 
 /* ---- global library state ---- */
@@ -171,7 +173,8 @@ bool smi_available(void);
 /* whole-device busy (permille) + container gfx engine ns + vram bytes */
 bool smi_sample_device(int host_index, uint32_t *busy_permille,
                        uint64_t *container_gfx_ns,
-                       uint64_t *container_vram, const pid_set_t *pids);
+                       uint64_t *container_vram, uint32_t *container_cus,
+                       const pid_set_t *pids);
 uint64_t smi_container_vram(int host_index, const pid_set_t *pids);
 
 /* hook implementations (exported) live in hip_hook.c / smi_hook.c     */

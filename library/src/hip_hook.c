@@ -126,11 +126,16 @@ static bool sample_util(int dev, uint32_t *cont_permille,
         }
     }
 
-    /* 3. own amd-smi query */
-    uint32_t busy = 0;
+    /* 3. own amd-smi query.  Container-share fallback ladder (parity
+     * with the reference's NVML strategy ladder, util_adapter.go):
+     *   a. per-process gfx engine-time delta (best),
+     *   b. per-process CU occupancy / total CUs,
+     *   c. whole-device busy, if WE launched since the last cycle —
+     *      right for a single-tenant device, conservative otherwise.  */
+    uint32_t busy = 0, cus = 0;
     uint64_t gfx_ns = 0, vram = 0;
     if (smi_available() &&
-        smi_sample_device(dev, &busy, &gfx_ns, &vram, &g_state.pids)) {
+        smi_sample_device(dev, &busy, &gfx_ns, &vram, &cus, &g_state.pids)) {
         uint64_t now = mono_ns();
         uint32_t cont = 0;
         if (h->prev_sample_ns && gfx_ns >= h->prev_proc_gfx_ns) {
@@ -140,6 +145,16 @@ static bool sample_util(int dev, uint32_t *cont_permille,
         }
         h->prev_proc_gfx_ns = gfx_ns;
         h->prev_sample_ns = now;
+        if (cont == 0 && cus > 0 && h->cu_count > 0)
+            cont = cus * 1000u / (uint32_t)h->cu_count;
+        if (cont == 0) {
+            uint64_t launches =
+                __atomic_load_n(&h->launch_count, __ATOMIC_RELAXED);
+            if (launches != h->prev_launch_count &&
+                busy >= (uint32_t)vgpu_dynconfig()->usage_threshold * 10)
+                cont = busy;
+            h->prev_launch_count = launches;
+        }
         *cont_permille = cont > 1000 ? 1000 : cont;
         *dev_permille = busy;
         /* publish for siblings when we own the shared bucket          */
@@ -315,11 +330,27 @@ static void refill(int dev, int64_t grant) {
     }
 }
 
+/* process-teardown guard: a detached thread touching amd-smi while
+ * exit() runs library destructors segfaults; the destructor flips the
+ * flag and briefly waits for the watcher to park.                     */
+static int g_shutdown;
+static int g_watcher_parked;
+
+__attribute__((destructor)) static void hook_fini(void) {
+    __atomic_store_n(&g_shutdown, 1, __ATOMIC_RELEASE);
+    for (int i = 0; i < 50; i++) { /* <=500ms grace                    */
+        if (__atomic_load_n(&g_watcher_parked, __ATOMIC_ACQUIRE)) break;
+        struct timespec ts = {0, 10000000L};
+        nanosleep(&ts, NULL);
+    }
+}
+
 static void *watcher_main(void *arg) {
     (void)arg;
     /* absolute-time cadence: drift-free 100ms grid, overrun floor     */
     uint64_t next = mono_ns();
     for (;;) {
+        if (__atomic_load_n(&g_shutdown, __ATOMIC_ACQUIRE)) break;
         next += (uint64_t)WATCHER_CYCLE_MS * 1000000ull;
         for (int dev = 0; dev < g_state.device_count; dev++) {
             if (cfg_dev(dev) < 0) continue;
@@ -336,6 +367,7 @@ static void *watcher_main(void *arg) {
                               (long)((next - now) % 1000000000ull)};
         nanosleep(&ts, NULL);
     }
+    __atomic_store_n(&g_watcher_parked, 1, __ATOMIC_RELEASE);
     return NULL;
 }
 
@@ -356,11 +388,13 @@ static void rate_limiter(int dev, int64_t grids) {
     dev_hot_t *h = &g_state.dev[dev];
     pthread_once(&g_watcher_once, start_watcher);
     dev_hot_init(dev);
+    __atomic_fetch_add(&h->launch_count, 1, __ATOMIC_RELAXED);
     if (grids > h->pool) grids = h->pool;
     int64_t *bucket = g_state.sm_node
                           ? &g_state.sm_node->devices[dev].tokens
                           : &h->tokens;
     for (;;) {
+        if (__atomic_load_n(&g_shutdown, __ATOMIC_ACQUIRE)) break;
         int64_t cur = __atomic_load_n(bucket, __ATOMIC_RELAXED);
         if (cur <= 0) {
             __atomic_store_n(&h->throttled, 1u, __ATOMIC_RELAXED);

@@ -126,7 +126,7 @@ static amdsmi_processor_handle handle_for(int dev) {
 
 bool smi_sample_device(int dev, uint32_t *busy_permille,
                        uint64_t *container_gfx_ns, uint64_t *container_vram,
-                       const pid_set_t *pids) {
+                       uint32_t *container_cus, const pid_set_t *pids) {
     if (!smi_available()) return false;
     amdsmi_processor_handle h = handle_for(dev);
     if (!h) return false;
@@ -134,8 +134,11 @@ bool smi_sample_device(int dev, uint32_t *busy_permille,
     memset(&eng, 0, sizeof(eng));
     if (real_smi.amdsmi_get_gpu_activity(h, &eng) != AMDSMI_STATUS_SUCCESS)
         return false;
-    *busy_permille = eng.gfx_activity * 10;
+    /* gfx_activity can read >100 on some stacks; clamp                */
+    uint32_t act = eng.gfx_activity > 100 ? 100 : eng.gfx_activity;
+    *busy_permille = act * 10;
     uint64_t gfx = 0, vram = 0;
+    uint32_t cus = 0;
     uint32_t n = 128;
     amdsmi_proc_info_t list[128];
     memset(list, 0, sizeof(list));
@@ -146,10 +149,12 @@ bool smi_sample_device(int dev, uint32_t *busy_permille,
             if (!vgpu_pid_set_contains(pids, (int32_t)list[i].pid)) continue;
             gfx += list[i].engine_usage.gfx;
             vram += list[i].memory_usage.vram_mem;
+            cus += list[i].cu_occupancy;
         }
     }
     *container_gfx_ns = gfx;
     *container_vram = vram;
+    *container_cus = cus;
     return true;
 }
 
