@@ -132,7 +132,7 @@ def test_throttle_slows_launch_storm():
 # warm up
 wk.wk_launch_busy(16, 512, 256, 20000); wk.wk_sync()
 t0 = time.perf_counter()
-wk.wk_launch_busy(600, 2048, 256, 60000)
+wk.wk_launch_busy(1600, 2048, 256, 60000)   # ~4.4s unthrottled
 wk.wk_sync()
 print(json.dumps({"elapsed": time.perf_counter() - t0}))
 """
