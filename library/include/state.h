@@ -37,6 +37,9 @@ typedef struct {
     hipError_t (*hipGetDevicePropertiesR0600)(hipDeviceProp_tR0600 *, int);
     hipError_t (*hipMemAdvise)(const void *, size_t, hipMemoryAdvise, int);
     hipError_t (*hipMemPrefetchAsync)(const void *, size_t, int, hipStream_t);
+    hipError_t (*hipHostMalloc)(void **, size_t, unsigned int);
+    hipError_t (*hipHostFree)(void *);
+    hipError_t (*hipHostGetDevicePointer)(void **, void *, unsigned int);
 
     hipError_t (*hipLaunchKernel)(const void *, dim3, dim3, void **, size_t,
                                   hipStream_t);
@@ -148,14 +151,15 @@ static inline uint32_t vgpu_device_flags(int dev) {
 }
 
 /* allocation registry (process-local ptr -> {size, kind, dev}) */
-#define ALLOC_KIND_DEVICE  0
-#define ALLOC_KIND_MANAGED 1   /* oversold spill, in vmem ledger       */
-#define ALLOC_KIND_ASYNC   2
+#define ALLOC_KIND_DEVICE   0
+#define ALLOC_KIND_MANAGED  1  /* oversold spill (HMM), in vmem ledger */
+#define ALLOC_KIND_ASYNC    2
+#define ALLOC_KIND_HOSTSPILL 3 /* oversold spill via mapped host mem   */
 int  alloc_registry_add(void *ptr, size_t size, int kind, int dev,
-                        int vmem_idx);
+                        int vmem_idx, void *host_ptr);
 /* returns true and fills outputs if found (and removes the entry)     */
 bool alloc_registry_remove(void *ptr, size_t *size, int *kind, int *dev,
-                           int *vmem_idx);
+                           int *vmem_idx, void **host_ptr);
 uint64_t alloc_registry_total(int dev);
 
 /* vmem ledger ops */

@@ -226,10 +226,22 @@ static int64_t control_cycle(int dev) {
     uint32_t target = snap.core_limit * 10;       /* % -> permille      */
     uint32_t cont = 0, busy = 0;
     bool have = sample_util(dev, &cont, &busy);
+    static uint32_t s_cycle;
+    if ((s_cycle++ % 10) == 0 || !have)
+        LOGGER(LOG_DEBUG,
+               "ctl dev=%d have=%d cont=%u busy=%u target=%u share=%lld "
+               "tok=%lld",
+               dev, (int)have, cont, busy, target, (long long)h->cur_share,
+               (long long)__atomic_load_n(
+                   g_state.sm_node ? &g_state.sm_node->devices[dev].tokens
+                                   : &h->tokens, __ATOMIC_RELAXED));
     if (!have) {
-        /* no sample: steady state — grant the nominal share           */
+        /* no sample at all: fail SAFE for isolation — grant only the
+         * limit-proportional share of the *initial* calibration (the
+         * reference's no-data behavior is its initial share too).     */
         metrics_inc(MET_WATCHER_MISS);
-        return h->pool * snap.core_limit / 100 / (1000 / WATCHER_CYCLE_MS);
+        return h->pool * snap.core_limit / 100 /
+               (1000 / WATCHER_CYCLE_MS) / 4;
     }
 
     /* soft-limit / auto exclusivity: when nobody else uses the GPU,
@@ -538,20 +550,43 @@ static void malloc_done(int lockfd) {
     if (lockfd >= 0) unlock_gpu_device(lockfd);
 }
 
-/* managed spill: allocate HMM memory past the HBM cap + ledger record */
+/* managed spill: allocate HMM memory past the HBM cap + ledger record.
+ * gfx950 boxes with XNACK disabled reject hipMallocManaged; spill then
+ * degrades to pinned mapped HOST memory (hipHostMalloc) — still usable
+ * from the GPU, still outside the HBM quota (the whole point).        */
 static hipError_t managed_spill(int dev, void **ptr, size_t size, int kind) {
-    hipError_t rc =
-        real_hip.hipMallocManaged(ptr, size, hipMemAttachGlobal);
-    if (rc != hipSuccess) return rc;
-    if (vgpu_dynconfig()->uva_advise && real_hip.hipMemAdvise) {
-        /* prefer host residency for spilled ranges: the quota exists
-         * because HBM is contended; keep spill out of HBM until used. */
-        real_hip.hipMemAdvise(*ptr, size, hipMemAdviseSetPreferredLocation,
-                              hipCpuDeviceId);
-        real_hip.hipMemAdvise(*ptr, size, hipMemAdviseSetAccessedBy, dev);
+    hipError_t rc = real_hip.hipMallocManaged
+                        ? real_hip.hipMallocManaged(ptr, size,
+                                                    hipMemAttachGlobal)
+                        : hipErrorNotSupported;
+    if (rc == hipSuccess) {
+        if (vgpu_dynconfig()->uva_advise && real_hip.hipMemAdvise) {
+            /* prefer host residency for spilled ranges: the quota
+             * exists because HBM is contended                          */
+            real_hip.hipMemAdvise(*ptr, size,
+                                  hipMemAdviseSetPreferredLocation,
+                                  hipCpuDeviceId);
+            real_hip.hipMemAdvise(*ptr, size, hipMemAdviseSetAccessedBy,
+                                  dev);
+        }
+        int idx = vmem_ledger_add(dev, (uint64_t)(uintptr_t)*ptr, size,
+                                  kind);
+        alloc_registry_add(*ptr, size, ALLOC_KIND_MANAGED, dev, idx, NULL);
+        return hipSuccess;
     }
-    int idx = vmem_ledger_add(dev, (uint64_t)(uintptr_t)*ptr, size, kind);
-    alloc_registry_add(*ptr, size, ALLOC_KIND_MANAGED, dev, idx);
+    /* host-mapped fallback */
+    void *hptr = NULL;
+    if (!real_hip.hipHostMalloc ||
+        real_hip.hipHostMalloc(&hptr, size, hipHostMallocMapped) !=
+            hipSuccess)
+        return rc; /* original managed error */
+    void *dptr = hptr;
+    if (real_hip.hipHostGetDevicePointer)
+        real_hip.hipHostGetDevicePointer(&dptr, hptr, 0);
+    *ptr = dptr;
+    int idx = vmem_ledger_add(dev, (uint64_t)(uintptr_t)dptr, size, kind);
+    alloc_registry_add(dptr, size, ALLOC_KIND_HOSTSPILL, dev, idx, hptr);
+    LOGGER(LOG_DEBUG, "spill %zu bytes to mapped host memory", size);
     return hipSuccess;
 }
 
@@ -578,7 +613,7 @@ EXPORT hipError_t hipMalloc(void **ptr, size_t size) {
     }
     if (rc == hipSuccess) {
         dev_hooked_add(dev, (int64_t)size);
-        alloc_registry_add(*ptr, size, ALLOC_KIND_DEVICE, dev, -1);
+        alloc_registry_add(*ptr, size, ALLOC_KIND_DEVICE, dev, -1, NULL);
     }
     malloc_done(lockfd);
     return rc;
@@ -597,7 +632,7 @@ EXPORT hipError_t hipExtMallocWithFlags(void **ptr, size_t size,
     hipError_t rc = real_hip.hipExtMallocWithFlags(ptr, size, flags);
     if (rc == hipSuccess) {
         dev_hooked_add(dev, (int64_t)size);
-        alloc_registry_add(*ptr, size, ALLOC_KIND_DEVICE, dev, -1);
+        alloc_registry_add(*ptr, size, ALLOC_KIND_DEVICE, dev, -1, NULL);
     }
     malloc_done(lockfd);
     return rc;
@@ -616,7 +651,7 @@ EXPORT hipError_t hipMallocManaged(void **ptr, size_t size,
     if (rc == hipSuccess) {
         int idx = vmem_ledger_add(dev, (uint64_t)(uintptr_t)*ptr, size,
                                   VMEM_KIND_SYNC);
-        alloc_registry_add(*ptr, size, ALLOC_KIND_MANAGED, dev, idx);
+        alloc_registry_add(*ptr, size, ALLOC_KIND_MANAGED, dev, idx, NULL);
     }
     return rc;
 }
@@ -645,7 +680,7 @@ EXPORT hipError_t hipMallocAsync(void **ptr, size_t size,
     hipError_t rc = real_hip.hipMallocAsync(ptr, size, stream);
     if (rc == hipSuccess) {
         dev_hooked_add(dev, (int64_t)size);
-        alloc_registry_add(*ptr, size, kind, dev, -1);
+        alloc_registry_add(*ptr, size, kind, dev, -1, NULL);
     }
     malloc_done(lockfd);
     return rc;
@@ -665,7 +700,7 @@ EXPORT hipError_t hipMallocFromPoolAsync(void **ptr, size_t size,
     hipError_t rc = real_hip.hipMallocFromPoolAsync(ptr, size, pool, stream);
     if (rc == hipSuccess) {
         dev_hooked_add(dev, (int64_t)size);
-        alloc_registry_add(*ptr, size, ALLOC_KIND_ASYNC, dev, -1);
+        alloc_registry_add(*ptr, size, ALLOC_KIND_ASYNC, dev, -1, NULL);
     }
     malloc_done(lockfd);
     return rc;
@@ -690,7 +725,7 @@ EXPORT hipError_t hipMallocPitch(void **ptr, size_t *pitch, size_t width,
     if (rc == hipSuccess) {
         size_t real_size = *pitch * height;
         dev_hooked_add(dev, (int64_t)real_size);
-        alloc_registry_add(*ptr, real_size, ALLOC_KIND_DEVICE, dev, -1);
+        alloc_registry_add(*ptr, real_size, ALLOC_KIND_DEVICE, dev, -1, NULL);
     }
     malloc_done(lockfd);
     return rc;
@@ -713,7 +748,7 @@ EXPORT hipError_t hipMalloc3D(hipPitchedPtr *p, hipExtent extent) {
     if (rc == hipSuccess) {
         size_t real_size = p->pitch * extent.height * extent.depth;
         dev_hooked_add(dev, (int64_t)real_size);
-        alloc_registry_add(p->ptr, real_size, ALLOC_KIND_DEVICE, dev, -1);
+        alloc_registry_add(p->ptr, real_size, ALLOC_KIND_DEVICE, dev, -1, NULL);
     }
     malloc_done(lockfd);
     return rc;
@@ -742,7 +777,7 @@ EXPORT hipError_t hipMallocArray(hipArray_t *array,
     hipError_t rc = real_hip.hipMallocArray(array, desc, width, height, flags);
     if (rc == hipSuccess) {
         dev_hooked_add(dev, (int64_t)est);
-        alloc_registry_add((void *)*array, est, ALLOC_KIND_DEVICE, dev, -1);
+        alloc_registry_add((void *)*array, est, ALLOC_KIND_DEVICE, dev, -1, NULL);
     }
     malloc_done(lockfd);
     return rc;
@@ -767,46 +802,57 @@ EXPORT hipError_t hipMalloc3DArray(hipArray_t *array,
     hipError_t rc = real_hip.hipMalloc3DArray(array, desc, extent, flags);
     if (rc == hipSuccess) {
         dev_hooked_add(dev, (int64_t)est);
-        alloc_registry_add((void *)*array, est, ALLOC_KIND_DEVICE, dev, -1);
+        alloc_registry_add((void *)*array, est, ALLOC_KIND_DEVICE, dev, -1, NULL);
     }
     malloc_done(lockfd);
     return rc;
 }
 
-static void release_tracking(void *ptr) {
+/* remove tracking for ptr; returns the real-free dispatch kind        */
+static int release_tracking(void *ptr, void **host_ptr_out) {
     size_t size;
     int kind, dev, vmem_idx;
-    if (!alloc_registry_remove(ptr, &size, &kind, &dev, &vmem_idx)) return;
-    if (kind == ALLOC_KIND_MANAGED) {
+    void *hptr = NULL;
+    if (!alloc_registry_remove(ptr, &size, &kind, &dev, &vmem_idx, &hptr))
+        return ALLOC_KIND_DEVICE;
+    if (kind == ALLOC_KIND_MANAGED || kind == ALLOC_KIND_HOSTSPILL) {
         if (vmem_idx >= 0) vmem_ledger_remove(vmem_idx);
     } else {
         dev_hooked_add(dev, -(int64_t)size);
     }
+    if (host_ptr_out) *host_ptr_out = hptr;
+    return kind;
 }
 
 EXPORT hipError_t hipFree(void *ptr) {
     if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
-    hipError_t rc = real_hip.hipFree(ptr);
-    if (rc == hipSuccess && !g_state.disabled) release_tracking(ptr);
-    return rc;
+    if (g_state.disabled || !ptr) return real_hip.hipFree(ptr);
+    void *hptr = NULL;
+    int kind = release_tracking(ptr, &hptr);
+    if (kind == ALLOC_KIND_HOSTSPILL && hptr)
+        return real_hip.hipHostFree(hptr);
+    return real_hip.hipFree(ptr);
 }
 
 EXPORT hipError_t hipFreeAsync(void *ptr, hipStream_t stream) {
     if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
-    hipError_t rc = real_hip.hipFreeAsync(ptr, stream);
+    if (g_state.disabled || !ptr) return real_hip.hipFreeAsync(ptr, stream);
     /* async free completes later; the ledger retires now — the quota
      * is conservative by at most the in-flight frees (reference
      * ASYNC_BRIDGE semantics collapsed: HIP pools return memory to the
      * pool, so the charge stays until pool trim anyway).              */
-    if (rc == hipSuccess && !g_state.disabled) release_tracking(ptr);
-    return rc;
+    void *hptr = NULL;
+    int kind = release_tracking(ptr, &hptr);
+    if (kind == ALLOC_KIND_HOSTSPILL && hptr)
+        return real_hip.hipHostFree(hptr);
+    return real_hip.hipFreeAsync(ptr, stream);
 }
 
 EXPORT hipError_t hipFreeArray(hipArray_t array) {
     if (vgpu_ensure_init() != 0) return hipErrorNotInitialized;
     hipError_t rc = real_hip.hipFreeArray(array);
     if (rc == hipSuccess && !g_state.disabled)
-        release_tracking((void *)array);
+        release_tracking((void *)array, NULL);
     return rc;
 }
 

@@ -133,6 +133,9 @@ static int load_real_hip(void) {
     LOAD(hipGetDevicePropertiesR0600);
     LOAD(hipMemAdvise);
     LOAD(hipMemPrefetchAsync);
+    LOAD(hipHostMalloc);
+    LOAD(hipHostFree);
+    LOAD(hipHostGetDevicePointer);
     LOAD(hipLaunchKernel);
     LOAD(hipExtLaunchKernel);
     LOAD(hipModuleLaunchKernel);
@@ -388,6 +391,7 @@ int vgpu_ensure_init(void) {
 typedef struct {
     uint64_t ptr;   /* 0 = empty, 1 = tombstone */
     uint64_t size;
+    uint64_t host_ptr; /* backing host alloc for HOSTSPILL entries */
     int32_t kind;
     int32_t dev;
     int32_t vmem_idx;
@@ -406,7 +410,7 @@ static inline uint32_t reg_hash(uint64_t p) {
 }
 
 int alloc_registry_add(void *ptr, size_t size, int kind, int dev,
-                       int vmem_idx) {
+                       int vmem_idx, void *host_ptr) {
     uint64_t p = (uint64_t)(uintptr_t)ptr;
     if (!p) return -1;
     pthread_mutex_lock(&g_reg_mu);
@@ -418,6 +422,7 @@ int alloc_registry_add(void *ptr, size_t size, int kind, int dev,
             g_reg[i].kind = kind;
             g_reg[i].dev = dev;
             g_reg[i].vmem_idx = vmem_idx;
+            g_reg[i].host_ptr = (uint64_t)(uintptr_t)host_ptr;
             if (dev >= 0 && dev < MAX_DEVICE_COUNT)
                 g_reg_dev_total[dev] += size;
             pthread_mutex_unlock(&g_reg_mu);
@@ -430,7 +435,7 @@ int alloc_registry_add(void *ptr, size_t size, int kind, int dev,
 }
 
 bool alloc_registry_remove(void *ptr, size_t *size, int *kind, int *dev,
-                           int *vmem_idx) {
+                           int *vmem_idx, void **host_ptr) {
     uint64_t p = (uint64_t)(uintptr_t)ptr;
     if (!p) return false;
     pthread_mutex_lock(&g_reg_mu);
@@ -442,6 +447,8 @@ bool alloc_registry_remove(void *ptr, size_t *size, int *kind, int *dev,
             if (kind) *kind = g_reg[i].kind;
             if (dev) *dev = g_reg[i].dev;
             if (vmem_idx) *vmem_idx = g_reg[i].vmem_idx;
+            if (host_ptr)
+                *host_ptr = (void *)(uintptr_t)g_reg[i].host_ptr;
             if (g_reg[i].dev >= 0 && g_reg[i].dev < MAX_DEVICE_COUNT)
                 g_reg_dev_total[g_reg[i].dev] -= g_reg[i].size;
             g_reg[i].ptr = 1; /* tombstone */
