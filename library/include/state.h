@@ -116,7 +116,9 @@ typedef struct {
     uint64_t prev_sample_ns;
     uint64_t launch_count;      /* launches gated on this device        */
     uint64_t prev_launch_count;
-    uint8_t _pad[8];
+    int64_t  waiting;           /* launchers parked in the rate limiter */
+    uint32_t low_cycles;        /* idle-bypass hysteresis               */
+    uint32_t _rsvd1;
 } dev_hot_t;
 
 /* ---- global library state ---- */

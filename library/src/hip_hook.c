@@ -147,14 +147,6 @@ static bool sample_util(int dev, uint32_t *cont_permille,
         h->prev_sample_ns = now;
         if (cont == 0 && cus > 0 && h->cu_count > 0)
             cont = cus * 1000u / (uint32_t)h->cu_count;
-        if (cont == 0) {
-            uint64_t launches =
-                __atomic_load_n(&h->launch_count, __ATOMIC_RELAXED);
-            if (launches != h->prev_launch_count &&
-                busy >= (uint32_t)vgpu_dynconfig()->usage_threshold * 10)
-                cont = busy;
-            h->prev_launch_count = launches;
-        }
         *cont_permille = cont > 1000 ? 1000 : cont;
         *dev_permille = busy;
         /* publish for siblings when we own the shared bucket          */
@@ -226,12 +218,28 @@ static int64_t control_cycle(int dev) {
     uint32_t target = snap.core_limit * 10;       /* % -> permille      */
     uint32_t cont = 0, busy = 0;
     bool have = sample_util(dev, &cont, &busy);
+
+    /* activity: launches gated since last cycle, or launchers parked
+     * in the rate limiter right now.  This is what distinguishes "the
+     * app is idle" from "we throttled the app into idleness".         */
+    uint64_t launches = __atomic_load_n(&h->launch_count, __ATOMIC_RELAXED);
+    int64_t waiting = __atomic_load_n(&h->waiting, __ATOMIC_RELAXED);
+    int active = launches != h->prev_launch_count || waiting > 0;
+    h->prev_launch_count = launches;
+
+    /* observed container share: per-process data when the platform
+     * provides it; whole-device busy while WE are active otherwise
+     * (exact for a single tenant, conservative for co-tenants).       */
+    uint32_t obs = cont;
+    if (obs == 0 && active) obs = busy;
+
     static uint32_t s_cycle;
     if ((s_cycle++ % 10) == 0 || !have)
         LOGGER(LOG_DEBUG,
-               "ctl dev=%d have=%d cont=%u busy=%u target=%u share=%lld "
-               "tok=%lld",
-               dev, (int)have, cont, busy, target, (long long)h->cur_share,
+               "ctl dev=%d have=%d cont=%u busy=%u obs=%u act=%d target=%u "
+               "share=%lld tok=%lld",
+               dev, (int)have, cont, busy, obs, active, target,
+               (long long)h->cur_share,
                (long long)__atomic_load_n(
                    g_state.sm_node ? &g_state.sm_node->devices[dev].tokens
                                    : &h->tokens, __ATOMIC_RELAXED));
@@ -250,8 +258,8 @@ static int64_t control_cycle(int dev) {
     int soft_on = (snap.flags & DEV_FLAG_SOFT_CORE) &&
                   snap.soft_core_limit > snap.core_limit &&
                   g_state.cfg->compute_policy == COMPUTE_POLICY_BALANCE;
-    if (soft_on || c->controller == 3) {
-        uint32_t others = busy > cont ? busy - cont : 0;
+    if (soft_on) {
+        uint32_t others = busy > obs ? busy - obs : 0;
         int exclusive = others < (uint32_t)c->auto_ext_util_threshold;
         if (exclusive != (int)h->excl_state) {
             if (++h->debounce >= c->auto_debounce_cycles) {
@@ -262,30 +270,28 @@ static int64_t control_cycle(int dev) {
         } else {
             h->debounce = 0;
         }
-        if (h->excl_state && soft_on) {
-            /* elastic ramp toward the soft ceiling                    */
-            if (++h->soft_cycle >= SOFT_ADJUST_CYCLES) {
-                h->soft_cycle = 0;
-            }
+        if (h->excl_state)
             eff_target = snap.soft_core_limit * 10;
-        } else {
-            h->soft_cycle = 0;
-        }
     }
 
-    /* anti-jitter bypass: long under-utilization with a hard limit
-     * means the app is not launch-bound — keep the bucket full so the
-     * throttle adds no latency (reference hard-limit bypass).         */
-    if (cont + 100 < eff_target && busy < eff_target) {
-        h->cur_share = h->pool * eff_target / 1000 /
-                       (1000 / WATCHER_CYCLE_MS);
-        return h->cur_share * 4; /* generous refill while idle         */
+    /* idle bypass with hysteresis: ONLY when the app itself is idle
+     * (no gated launches, nobody parked) for several cycles do we keep
+     * the bucket topped up so the next burst starts latency-free.  A
+     * throttled-into-idleness app never qualifies (active==1 while
+     * anyone sits in the rate limiter).                               */
+    if (!active && obs == 0) {
+        if (++h->low_cycles >= 5) {
+            h->low_cycles = 5; /* saturate */
+            return h->pool * eff_target / 1000 / (1000 / WATCHER_CYCLE_MS);
+        }
+    } else {
+        h->low_cycles = 0;
     }
 
     int64_t share = h->cur_share;
     int ctl = c->controller == 3 ? 2 : c->controller; /* auto -> aimd  */
-    share = ctl == 1 ? ctl_delta(c, h, share, eff_target, cont)
-                     : ctl_aimd(c, h, share, eff_target, cont);
+    share = ctl == 1 ? ctl_delta(c, h, share, eff_target, obs)
+                     : ctl_aimd(c, h, share, eff_target, obs);
     int64_t smin = h->pool / 100000;
     if (smin < 1) smin = 1;
     int64_t smax = h->pool;
@@ -348,7 +354,13 @@ static void refill(int dev, int64_t grant) {
 static int g_shutdown;
 static int g_watcher_parked;
 
-__attribute__((destructor)) static void hook_fini(void) {
+static void hook_fini(void);
+/* atexit ordering: exit() runs atexit handlers BEFORE ELF destructors,
+ * and a preloaded library's own destructor runs after everyone else's
+ * — too late to stop the watcher from touching a finalized amd-smi.  */
+void vgpu_register_fini_atexit(void) { atexit(hook_fini); }
+
+static void hook_fini(void) {
     __atomic_store_n(&g_shutdown, 1, __ATOMIC_RELEASE);
     for (int i = 0; i < 50; i++) { /* <=500ms grace                    */
         if (__atomic_load_n(&g_watcher_parked, __ATOMIC_ACQUIRE)) break;
@@ -384,6 +396,7 @@ static void *watcher_main(void *arg) {
 }
 
 static void start_watcher(void) {
+    vgpu_register_fini_atexit();
     pthread_t t;
     pthread_attr_t a;
     pthread_attr_init(&a);
@@ -405,10 +418,15 @@ static void rate_limiter(int dev, int64_t grids) {
     int64_t *bucket = g_state.sm_node
                           ? &g_state.sm_node->devices[dev].tokens
                           : &h->tokens;
+    int waited = 0;
     for (;;) {
         if (__atomic_load_n(&g_shutdown, __ATOMIC_ACQUIRE)) break;
         int64_t cur = __atomic_load_n(bucket, __ATOMIC_RELAXED);
         if (cur <= 0) {
+            if (!waited) {
+                waited = 1;
+                __atomic_fetch_add(&h->waiting, 1, __ATOMIC_ACQ_REL);
+            }
             __atomic_store_n(&h->throttled, 1u, __ATOMIC_RELAXED);
             metrics_inc(MET_RATE_SLEEP);
             struct timespec ts = {0, TIME_TICK_MS * 1000000L};
@@ -419,6 +437,7 @@ static void rate_limiter(int dev, int64_t grids) {
                                         __ATOMIC_ACQ_REL, __ATOMIC_RELAXED))
             break;
     }
+    if (waited) __atomic_fetch_sub(&h->waiting, 1, __ATOMIC_ACQ_REL);
     __atomic_store_n(&h->throttled, 0u, __ATOMIC_RELAXED);
 }
 
