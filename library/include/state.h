@@ -88,6 +88,7 @@ typedef struct {
     hipError_t (*hipEventDestroy)(hipEvent_t);
     hipError_t (*hipStreamIsCapturing)(hipStream_t,
                                        hipStreamCaptureStatus *);
+    hipError_t (*hipGetLastError)(void);
     hipError_t (*hipGetProcAddress)(const char *, void **, int, uint64_t,
                                     hipDriverProcAddressQueryResult *);
 } hip_real_t;

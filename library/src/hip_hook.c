@@ -593,7 +593,10 @@ static hipError_t managed_spill(int dev, void **ptr, size_t size, int kind) {
         alloc_registry_add(*ptr, size, ALLOC_KIND_MANAGED, dev, idx, NULL);
         return hipSuccess;
     }
-    /* host-mapped fallback */
+    /* host-mapped fallback.  Clear HIP's sticky per-thread last-error
+     * first: the failed managed attempt above must not leak into the
+     * app's later hipGetLastError.                                    */
+    if (real_hip.hipGetLastError) real_hip.hipGetLastError();
     void *hptr = NULL;
     if (!real_hip.hipHostMalloc ||
         real_hip.hipHostMalloc(&hptr, size, hipHostMallocMapped) !=
@@ -602,6 +605,7 @@ static hipError_t managed_spill(int dev, void **ptr, size_t size, int kind) {
     void *dptr = hptr;
     if (real_hip.hipHostGetDevicePointer)
         real_hip.hipHostGetDevicePointer(&dptr, hptr, 0);
+    if (real_hip.hipGetLastError) real_hip.hipGetLastError();
     *ptr = dptr;
     int idx = vmem_ledger_add(dev, (uint64_t)(uintptr_t)dptr, size, kind);
     alloc_registry_add(dptr, size, ALLOC_KIND_HOSTSPILL, dev, idx, hptr);
@@ -628,6 +632,7 @@ EXPORT hipError_t hipMalloc(void **ptr, size_t size) {
           (vgpu_device_flags(dev) & DEV_FLAG_OVERSOLD)))) {
         /* real HBM exhausted: spill (reference driver-OOM fallback)   */
         malloc_done(lockfd);
+        if (real_hip.hipGetLastError) real_hip.hipGetLastError();
         return managed_spill(dev, ptr, size, VMEM_KIND_SYNC);
     }
     if (rc == hipSuccess) {

@@ -160,6 +160,7 @@ static int load_real_hip(void) {
     LOAD(hipEventElapsedTime);
     LOAD(hipEventDestroy);
     LOAD(hipStreamIsCapturing);
+    LOAD(hipGetLastError);
     LOAD(hipGetProcAddress);
 #undef LOAD
     return real_hip.hipMalloc && real_hip.hipLaunchKernel ? 0 : -1;
