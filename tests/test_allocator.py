@@ -1,0 +1,161 @@
+"""Allocator behavior tests on fake devices/links (reference
+pkg/device/allocator/*_test.go strategy: fake registry encoded exactly
+like real node annotations)."""
+import pytest
+
+from vgpu_manager_amd.device.allocator import (
+    AllocationError,
+    Allocator,
+    R_INSUFFICIENT_MEMORY,
+    R_INSUFFICIENT_SLOT,
+    R_TOPOLOGY_UNSATISFIED,
+    build_allocation_request,
+)
+from vgpu_manager_amd.device.types import (
+    decode_node_devices,
+    encode_node_devices,
+    fake_device,
+    fake_node,
+    marshal_pod_claim,
+    unmarshal_pod_claim,
+    NodeInfo,
+)
+from vgpu_manager_amd.util import consts
+
+
+def make_pod(number=1, cores=0, memory=0, ann=None, name="p"):
+    limits = {consts.vgpu_number_resource(): number}
+    if cores:
+        limits[consts.vgpu_core_resource()] = cores
+    if memory:
+        limits[consts.vgpu_memory_resource()] = memory
+    return {
+        "metadata": {"name": name, "namespace": "default",
+                     "uid": f"uid-{name}", "annotations": ann or {}},
+        "spec": {"containers": [{
+            "name": "main", "resources": {"limits": limits}}]},
+    }
+
+
+def test_claim_codec_roundtrip():
+    pod = make_pod(number=2, cores=100, memory=8192)
+    node = fake_node("n1")
+    req = build_allocation_request(pod)
+    cdcs = Allocator(node).allocate(req)
+    text = marshal_pod_claim(cdcs)
+    # byte-format: "main[0_GPU-fake-0000_50_4096,...]"
+    assert text.startswith("main[")
+    back = unmarshal_pod_claim(text)
+    assert back[0].name == "main"
+    assert len(back[0].claims) == 2
+    assert back[0].claims[0].cores == 50  # 100 cores over 2 devices
+
+
+def test_node_register_codec():
+    devs = [fake_device(i) for i in range(4)]
+    ann = encode_node_devices(devs)
+    back = decode_node_devices(ann)
+    assert [d.id for d in back] == [0, 1, 2, 3]
+    assert back[0].memory == 294912
+
+
+def test_oom_reason():
+    node = NodeInfo("n", [fake_device(0, memory=1000)])
+    req = build_allocation_request(make_pod(number=1, memory=2000))
+    with pytest.raises(AllocationError) as e:
+        Allocator(node).allocate(req)
+    assert e.value.reason == R_INSUFFICIENT_MEMORY
+
+
+def test_slot_exhaustion():
+    node = NodeInfo("n", [fake_device(0, number=1)])
+    a = Allocator(node)
+    a.allocate(build_allocation_request(make_pod(name="a")))
+    with pytest.raises(AllocationError) as e:
+        a.allocate(build_allocation_request(make_pod(name="b")))
+    assert e.value.reason == R_INSUFFICIENT_SLOT
+
+
+def test_binpack_vs_spread():
+    node = fake_node("n", 4)
+    a = Allocator(node)
+    first = a.allocate(build_allocation_request(make_pod(name="a")))
+    # binpack: second pod lands on the same device
+    second = a.allocate(build_allocation_request(make_pod(name="b")))
+    assert first[0].claims[0].id == second[0].claims[0].id
+
+    node2 = fake_node("n2", 4)
+    a2 = Allocator(node2)
+    ann = {consts.device_scheduler_policy_ann(): consts.POLICY_SPREAD}
+    f1 = a2.allocate(build_allocation_request(make_pod(name="a", ann=ann)))
+    f2 = a2.allocate(build_allocation_request(make_pod(name="b", ann=ann)))
+    assert f1[0].claims[0].id != f2[0].claims[0].id
+
+
+def test_numa_strict():
+    # 8 GPUs, 2 NUMA domains of 4; request 4 with numa-strict -> one numa
+    node = fake_node("n", 8, numa_split=2)
+    ann = {consts.topology_mode_ann(): consts.TOPO_NUMA_STRICT}
+    cdcs = Allocator(node).allocate(
+        build_allocation_request(make_pod(number=4, ann=ann)))
+    numas = {node.devices[c.id].info.numa for c in cdcs[0].claims}
+    assert len(numas) == 1
+    # request 5 with strict -> fails (no numa has 5)
+    node2 = fake_node("n", 8, numa_split=2)
+    with pytest.raises(AllocationError) as e:
+        Allocator(node2).allocate(
+            build_allocation_request(make_pod(number=5, ann=ann)))
+    assert e.value.reason == R_TOPOLOGY_UNSATISFIED
+
+
+def test_link_mode_prefers_xgmi():
+    # mixed topology: no xGMI (PCIe islands); link mode picks same-NUMA
+    node = fake_node("n", 8, numa_split=2, full_xgmi=False)
+    ann = {consts.topology_mode_ann(): consts.TOPO_LINK}
+    cdcs = Allocator(node).allocate(
+        build_allocation_request(make_pod(number=2, ann=ann)))
+    ids = [c.id for c in cdcs[0].claims]
+    numas = {node.devices[i].info.numa for i in ids}
+    assert len(numas) == 1, f"link mode crossed NUMA: {ids}"
+
+
+def test_link_strict_on_full_xgmi_ok():
+    node = fake_node("n", 8, full_xgmi=True)
+    ann = {consts.topology_mode_ann(): consts.TOPO_LINK_STRICT}
+    cdcs = Allocator(node).allocate(
+        build_allocation_request(make_pod(number=4, ann=ann)))
+    assert len(cdcs[0].claims) == 4
+
+
+def test_init_container_lifecycle():
+    # init container needs 2 GPUs, app needs 1: reservation = max = 2
+    # devices' slots, but only the app's usage persists plus init peak
+    node = fake_node("n", 2, numa_split=0)
+    pod = {
+        "metadata": {"name": "p", "annotations": {}},
+        "spec": {
+            "initContainers": [{
+                "name": "init",
+                "resources": {"limits": {
+                    consts.vgpu_number_resource(): 1,
+                    consts.vgpu_memory_resource(): 4096}}}],
+            "containers": [{
+                "name": "main",
+                "resources": {"limits": {
+                    consts.vgpu_number_resource(): 1,
+                    consts.vgpu_memory_resource(): 2048}}}],
+        },
+    }
+    cdcs = Allocator(node).allocate(build_allocation_request(pod))
+    assert {c.name for c in cdcs} == {"init", "main"}
+    # peak on the init device is max(4096, app usage on that device)
+    init_dev = cdcs[0].claims[0].id
+    assert node.devices[init_dev].used_memory >= 4096
+
+
+def test_uuid_include_exclude():
+    node = fake_node("n", 4)
+    ann = {consts.include_gpu_uuid_ann(): "GPU-fake-0002"}
+    cdcs = Allocator(node).allocate(
+        build_allocation_request(make_pod(ann=ann)))
+    assert cdcs[0].claims[0].id == 2

@@ -1,0 +1,320 @@
+"""Node-local allocation engine (scheduler + device plugin shared).
+
+Behavior parity with reference pkg/device/allocator (allocator.go,
+priority.go, tiered.go), MI355X-shaped:
+  * requests decode from pod resource limits
+    (amd.com/vgpu-number|vgpu-cores|vgpu-memory);
+  * init-container lifecycle: reservation = sidecars + max(app, maxInit)
+    (reference init_container_vgpu_support_design.md);
+  * device filtering with machine-readable reason codes;
+  * binpack/spread device priority;
+  * topology modes: numa / numa-strict / link / link-strict — the xGMI
+    scorer minimizes pairwise link cost (XGMI < same-NUMA-PCIe <
+    cross-NUMA); combination enumeration capped (reference tiered.go:42)
+"""
+from __future__ import annotations
+
+import itertools
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+from ..util import consts
+from .types import (
+    ContainerDeviceClaim,
+    DeviceClaim,
+    DeviceUsage,
+    LINK_PCIE_NUMA,
+    LINK_SYS,
+    LINK_XGMI,
+    NodeInfo,
+)
+
+MAX_COMBINATIONS = 50000   # enumeration cap (reference tiered.go:42)
+
+# link cost for the scorer: lower is better
+_LINK_COST = {LINK_XGMI: 0, LINK_PCIE_NUMA: 1, LINK_SYS: 4}
+
+
+class AllocationError(Exception):
+    def __init__(self, reason: str, msg: str = ""):
+        super().__init__(f"{reason}: {msg}" if msg else reason)
+        self.reason = reason
+
+
+# stable reason codes (reference pkg/scheduler/reason)
+R_INSUFFICIENT_SLOT = "InsufficientVGPUSlot"
+R_INSUFFICIENT_CORES = "InsufficientVGPUCores"
+R_INSUFFICIENT_MEMORY = "InsufficientVGPUMemory"
+R_NO_HEALTHY_DEVICE = "NoHealthyDevice"
+R_FILTERED_BY_UUID = "FilteredByUUID"
+R_FILTERED_BY_TYPE = "FilteredByType"
+R_TOPOLOGY_UNSATISFIED = "TopologyUnsatisfied"
+R_INVALID_REQUEST = "InvalidResourceRequest"
+
+
+@dataclass
+class ContainerRequest:
+    name: str
+    number: int = 0
+    cores: int = 0      # per device, 0-100
+    memory: int = 0     # per device, MiB; 0 = node default share
+    is_init: bool = False
+    is_sidecar: bool = False
+
+
+@dataclass
+class AllocationRequest:
+    containers: List[ContainerRequest] = field(default_factory=list)
+    topology_mode: str = consts.TOPO_NONE
+    device_policy: str = consts.POLICY_BINPACK
+    include_uuids: List[str] = field(default_factory=list)
+    exclude_uuids: List[str] = field(default_factory=list)
+    include_types: List[str] = field(default_factory=list)
+    exclude_types: List[str] = field(default_factory=list)
+
+
+def _parse_qty(v) -> int:
+    """k8s resource quantity -> int (plain integers only for vgpu-*)."""
+    if isinstance(v, int):
+        return v
+    s = str(v).strip()
+    mult = 1
+    for suffix, m in (("Ki", 1024), ("Mi", 1 << 20), ("Gi", 1 << 30),
+                      ("k", 1000), ("M", 10 ** 6), ("G", 10 ** 9)):
+        if s.endswith(suffix):
+            s = s[: -len(suffix)]
+            mult = m
+            break
+    return int(float(s) * mult)
+
+
+def build_allocation_request(pod: dict) -> AllocationRequest:
+    """Decode a pod manifest (dict form) into an AllocationRequest.
+    Reference: allocator/request.go BuildAllocationRequest.            """
+    meta = pod.get("metadata", {})
+    ann = meta.get("annotations", {}) or {}
+    spec = pod.get("spec", {})
+
+    req = AllocationRequest()
+    req.topology_mode = ann.get(consts.topology_mode_ann(), consts.TOPO_NONE)
+    req.device_policy = ann.get(consts.device_scheduler_policy_ann(),
+                                consts.POLICY_BINPACK)
+
+    def split_ann(key):
+        v = ann.get(key, "")
+        return [x for x in v.split(",") if x] if v else []
+
+    req.include_uuids = split_ann(consts.include_gpu_uuid_ann())
+    req.exclude_uuids = split_ann(consts.exclude_gpu_uuid_ann())
+    req.include_types = split_ann(consts.include_gpu_type_ann())
+    req.exclude_types = split_ann(consts.exclude_gpu_type_ann())
+
+    def decode_containers(containers, is_init):
+        for c in containers or []:
+            limits = (c.get("resources", {}) or {}).get("limits", {}) or {}
+            num = limits.get(consts.vgpu_number_resource())
+            if num is None:
+                continue
+            number = _parse_qty(num)
+            if number <= 0 or number > consts.MAX_DEVICE_COUNT:
+                raise AllocationError(R_INVALID_REQUEST,
+                                      f"vgpu-number {number}")
+            cores = _parse_qty(limits.get(consts.vgpu_core_resource(), 0))
+            if cores < 0 or cores > consts.CORES_PER_GPU * number:
+                raise AllocationError(R_INVALID_REQUEST, f"cores {cores}")
+            memory = _parse_qty(limits.get(consts.vgpu_memory_resource(), 0))
+            restart = c.get("restartPolicy")
+            req.containers.append(ContainerRequest(
+                name=c["name"], number=number,
+                cores=cores // number if number else cores,
+                memory=memory // number if number else memory,
+                is_init=is_init and restart != "Always",
+                is_sidecar=is_init and restart == "Always"))
+
+    decode_containers(spec.get("initContainers"), True)
+    decode_containers(spec.get("containers"), False)
+    return req
+
+
+class Allocator:
+    """Allocates devices on one NodeInfo (mutating its usage)."""
+
+    def __init__(self, node: NodeInfo):
+        self.node = node
+
+    # ---- filtering ----
+    def _filter(self, req: AllocationRequest, cr: ContainerRequest
+                ) -> List[DeviceUsage]:
+        devs = list(self.node.devices.values())
+        if not any(d.info.healthy for d in devs):
+            raise AllocationError(R_NO_HEALTHY_DEVICE, self.node.name)
+        out = []
+        last_reason = R_INSUFFICIENT_SLOT
+        for d in devs:
+            if not d.info.healthy:
+                continue
+            if req.include_uuids and d.info.uuid not in req.include_uuids:
+                last_reason = R_FILTERED_BY_UUID
+                continue
+            if d.info.uuid in req.exclude_uuids:
+                last_reason = R_FILTERED_BY_UUID
+                continue
+            if req.include_types and not any(
+                    t.lower() in d.info.type.lower()
+                    for t in req.include_types):
+                last_reason = R_FILTERED_BY_TYPE
+                continue
+            if any(t.lower() in d.info.type.lower()
+                   for t in req.exclude_types):
+                last_reason = R_FILTERED_BY_TYPE
+                continue
+            if d.free_number() < 1:
+                last_reason = R_INSUFFICIENT_SLOT
+                continue
+            if cr.cores and d.free_cores() < cr.cores:
+                last_reason = R_INSUFFICIENT_CORES
+                continue
+            mem = cr.memory or (d.info.memory // d.info.number)
+            if d.free_memory() < mem:
+                last_reason = R_INSUFFICIENT_MEMORY
+                continue
+            out.append(d)
+        if len(out) < cr.number:
+            raise AllocationError(last_reason,
+                                  f"{len(out)}/{cr.number} devices fit")
+        return out
+
+    # ---- priority sort ----
+    def _sort(self, devs: List[DeviceUsage], policy: str
+              ) -> List[DeviceUsage]:
+        def used_frac(d: DeviceUsage):
+            return (d.used_cores / max(d.info.core, 1) +
+                    d.used_memory / max(d.info.memory, 1) +
+                    d.used_number / max(d.info.number, 1))
+
+        # binpack: most-used first (fill devices up); spread: least-used
+        return sorted(devs, key=lambda d: (
+            -used_frac(d) if policy == consts.POLICY_BINPACK
+            else used_frac(d), d.info.id))
+
+    # ---- topology ----
+    def _link_cost(self, a: int, b: int) -> int:
+        topo = self.node.topology
+        if not topo:
+            return _LINK_COST[LINK_SYS]
+        for dt in topo.devices:
+            if dt.id == a:
+                link = dt.links.get(b)
+                if link is None:
+                    return _LINK_COST[LINK_SYS] * 2  # unconnected
+                return _LINK_COST.get(link.kind, _LINK_COST[LINK_SYS])
+        return _LINK_COST[LINK_SYS]
+
+    def _numa_of(self, dev_id: int) -> int:
+        u = self.node.devices.get(dev_id)
+        return u.info.numa if u else -1
+
+    def _pick_topology(self, devs: List[DeviceUsage], n: int, mode: str,
+                       policy: str) -> List[DeviceUsage]:
+        if n <= 1 or mode == consts.TOPO_NONE:
+            return devs[:n]
+        strict = mode in (consts.TOPO_NUMA_STRICT, consts.TOPO_LINK_STRICT)
+        numa_mode = mode in (consts.TOPO_NUMA, consts.TOPO_NUMA_STRICT)
+
+        if numa_mode:
+            # group by NUMA; pick the group that fits with best policy
+            groups: Dict[int, List[DeviceUsage]] = {}
+            for d in devs:
+                groups.setdefault(d.info.numa, []).append(d)
+            fitting = [g for g in groups.values() if len(g) >= n]
+            if fitting:
+                # binpack: smallest fitting group; spread: largest
+                fitting.sort(key=lambda g: (len(g) if policy ==
+                                            consts.POLICY_BINPACK
+                                            else -len(g)))
+                return self._sort(fitting[0], policy)[:n]
+            if strict:
+                raise AllocationError(R_TOPOLOGY_UNSATISFIED,
+                                      f"no NUMA group with {n} free")
+            return devs[:n]
+
+        # link mode: minimize total pairwise link cost over the subset.
+        ids = [d.info.id for d in devs]
+        best, best_cost = None, None
+        count = 0
+        for combo in itertools.combinations(range(len(ids)), n):
+            count += 1
+            if count > MAX_COMBINATIONS:
+                break
+            cost = 0
+            for i, j in itertools.combinations(combo, 2):
+                cost += self._link_cost(ids[i], ids[j])
+            # tiebreak toward the policy order (devs is policy-sorted)
+            if best_cost is None or cost < best_cost:
+                best, best_cost = combo, cost
+        if best is None:
+            return devs[:n]
+        if strict and best_cost is not None and best_cost > 0:
+            raise AllocationError(R_TOPOLOGY_UNSATISFIED,
+                                  f"best link cost {best_cost} > 0")
+        return [devs[i] for i in best]
+
+    # ---- allocation ----
+    def allocate_container(self, req: AllocationRequest,
+                           cr: ContainerRequest) -> ContainerDeviceClaim:
+        devs = self._filter(req, cr)
+        devs = self._sort(devs, req.device_policy)
+        chosen = self._pick_topology(devs, cr.number, req.topology_mode,
+                                     req.device_policy)
+        if len(chosen) < cr.number:
+            raise AllocationError(R_INSUFFICIENT_SLOT,
+                                  f"{len(chosen)}/{cr.number}")
+        cdc = ContainerDeviceClaim(name=cr.name)
+        for d in chosen:
+            mem = cr.memory or (d.info.memory // d.info.number)
+            claim = DeviceClaim(id=d.info.id, uuid=d.info.uuid,
+                                cores=cr.cores, memory=mem)
+            cdc.claims.append(claim)
+            d.used_number += 1
+            d.used_cores += cr.cores
+            d.used_memory += mem
+        return cdc
+
+    def allocate(self, req: AllocationRequest) -> List[ContainerDeviceClaim]:
+        """Allocate all containers.  Init containers (non-sidecar) are
+        allocated for sizing but their usage is released after the max
+        is accounted — reservation = sidecars + max(app, maxInit).     """
+        out: List[ContainerDeviceClaim] = []
+        init_peak: Dict[int, DeviceClaim] = {}
+
+        # first pass: plain init containers — track the per-device peak
+        for cr in req.containers:
+            if not cr.is_init or cr.is_sidecar:
+                continue
+            cdc = self.allocate_container(req, cr)
+            out.append(cdc)
+            for c in cdc.claims:
+                prev = init_peak.get(c.id)
+                if prev is None or (c.cores > prev.cores or
+                                    c.memory > prev.memory):
+                    init_peak[c.id] = c
+            # release (init containers run sequentially, then exit)
+            self.node.remove_pod_claims([cdc])
+
+        # second pass: sidecars + app containers accumulate
+        for cr in req.containers:
+            if cr.is_init and not cr.is_sidecar:
+                continue
+            out.append(self.allocate_container(req, cr))
+
+        # re-apply the init peak where it exceeds the app usage
+        for dev_id, peak in init_peak.items():
+            u = self.node.devices.get(dev_id)
+            if u is None:
+                continue
+            if u.used_cores < peak.cores:
+                u.used_cores = peak.cores
+            if u.used_memory < peak.memory:
+                u.used_memory = peak.memory
+
+        return out

@@ -1,0 +1,287 @@
+"""Device model and wire formats.
+
+Byte-compatible with the reference formats (pkg/device/types.go):
+  * node annotation `node-device-register`: JSON list of DeviceInfo
+  * claim text: ``"<id>_<uuid>_<cores>_<memory>"``
+  * container claim: ``"<name>[<claim>,<claim>]"``
+  * pod claim: container claims joined with ``";"``
+  * topology annotation: JSON NodeTopologyInfo (xGMI link model)
+
+The MI355X device model replaces NVLink counts with xGMI link
+weights/hops and MIG with SPX/CPX compute-partition modes.
+"""
+from __future__ import annotations
+
+import json
+import re
+import time
+from dataclasses import dataclass, field, asdict
+from typing import Dict, List, Optional
+
+
+@dataclass
+class DeviceInfo:
+    """One physical GPU as registered on the node annotation."""
+
+    id: int
+    type: str = "MI355X"
+    uuid: str = ""
+    core: int = 100            # allocatable cores (100 = whole GPU)
+    memory: int = 294912       # MiB; 288 GiB HBM3E
+    number: int = 10           # vGPU split count
+    numa: int = -1
+    cpx: bool = False          # CPX compute-partition mode (MIG analog)
+    busId: str = ""
+    capability: float = 9.5    # gfx950
+    healthy: bool = True
+
+    def to_json(self) -> dict:
+        return asdict(self)
+
+    @staticmethod
+    def from_json(d: dict) -> "DeviceInfo":
+        known = {f for f in DeviceInfo.__dataclass_fields__}
+        return DeviceInfo(**{k: v for k, v in d.items() if k in known})
+
+
+def encode_node_devices(devices: List[DeviceInfo]) -> str:
+    return json.dumps([d.to_json() for d in devices], separators=(",", ":"))
+
+
+def decode_node_devices(val: str) -> List[DeviceInfo]:
+    if not val or not val.strip():
+        raise ValueError("input value is empty")
+    out = [DeviceInfo.from_json(d) for d in json.loads(val)]
+    out.sort(key=lambda d: d.id)
+    return out
+
+
+# ---- claims ----
+
+@dataclass
+class DeviceClaim:
+    id: int
+    uuid: str
+    cores: int
+    memory: int  # MiB
+
+    def marshal(self) -> str:
+        return f"{self.id}_{self.uuid}_{self.cores}_{self.memory}"
+
+    @staticmethod
+    def unmarshal(text: str) -> "DeviceClaim":
+        text = text.replace(" ", "")
+        parts = text.split("_")
+        if len(parts) != 4:
+            raise ValueError(f"claim format error: {text!r}")
+        return DeviceClaim(id=int(parts[0]), uuid=parts[1],
+                           cores=int(parts[2]), memory=int(parts[3]))
+
+
+@dataclass
+class ContainerDeviceClaim:
+    name: str
+    claims: List[DeviceClaim] = field(default_factory=list)
+
+    def marshal(self) -> str:
+        inner = ",".join(c.marshal() for c in self.claims)
+        return f"{self.name}[{inner}]"
+
+    @staticmethod
+    def unmarshal(text: str) -> "ContainerDeviceClaim":
+        text = text.replace(" ", "")
+        m = re.fullmatch(r"([^\[\]]*)\[(.*)\]", text)
+        if not m:
+            raise ValueError(f"container claim format error: {text!r}")
+        name, inner = m.group(1), m.group(2)
+        claims = [DeviceClaim.unmarshal(t) for t in inner.split(",") if t]
+        return ContainerDeviceClaim(name=name, claims=claims)
+
+
+def marshal_pod_claim(cdcs: List[ContainerDeviceClaim]) -> str:
+    return ";".join(c.marshal() for c in cdcs)
+
+
+def unmarshal_pod_claim(text: str) -> List[ContainerDeviceClaim]:
+    text = text.replace(" ", "")
+    if not text:
+        raise ValueError("input text is empty")
+    return [ContainerDeviceClaim.unmarshal(t)
+            for t in text.split(";") if t]
+
+
+# ---- topology (xGMI) ----
+
+# P2P link classes on an AMD node, strongest first.  On a fully
+# connected 8x MI355X node every pair is XGMI; the ladder still matters
+# for PCIe-attached mixes and multi-hop fabrics (reference tier ladder
+# NV18..SYS collapses to this).
+LINK_XGMI = "XGMI"        # direct xGMI link
+LINK_PCIE_NUMA = "PIX"    # same NUMA / same PCIe root
+LINK_SYS = "SYS"          # cross-NUMA interconnect
+
+
+@dataclass
+class DeviceLink:
+    peer_id: int
+    kind: str = LINK_XGMI
+    weight: int = 1           # amdsmi topo weight (lower = closer)
+    hops: int = 1
+
+
+@dataclass
+class DeviceTopology:
+    id: int
+    uuid: str = ""
+    numa: int = -1
+    links: Dict[int, DeviceLink] = field(default_factory=dict)
+
+
+@dataclass
+class NodeTopologyInfo:
+    devices: List[DeviceTopology] = field(default_factory=list)
+
+    def encode(self) -> str:
+        return json.dumps([
+            dict(id=d.id, uuid=d.uuid, numa=d.numa,
+                 links={str(k): dict(peer_id=l.peer_id, kind=l.kind,
+                                     weight=l.weight, hops=l.hops)
+                        for k, l in d.links.items()})
+            for d in self.devices
+        ], separators=(",", ":"))
+
+    @staticmethod
+    def decode(val: str) -> "NodeTopologyInfo":
+        out = NodeTopologyInfo()
+        for d in json.loads(val):
+            dt = DeviceTopology(id=d["id"], uuid=d.get("uuid", ""),
+                                numa=d.get("numa", -1))
+            for k, l in d.get("links", {}).items():
+                dt.links[int(k)] = DeviceLink(**l)
+            out.devices.append(dt)
+        return out
+
+
+# ---- node config info (annotation) ----
+
+@dataclass
+class NodeConfigInfo:
+    deviceSplitCount: int = 10
+    deviceMemoryScaling: float = 1.0
+    deviceMemoryFactor: int = 1
+    deviceCoresScaling: float = 1.0
+    excludeDevices: List[int] = field(default_factory=list)
+    openVCore: bool = False
+    openVMemory: bool = False
+
+    def encode(self) -> str:
+        return json.dumps(asdict(self), separators=(",", ":"))
+
+    @staticmethod
+    def decode(val: str) -> "NodeConfigInfo":
+        d = json.loads(val)
+        known = {f for f in NodeConfigInfo.__dataclass_fields__}
+        return NodeConfigInfo(**{k: v for k, v in d.items() if k in known})
+
+
+# ---- node usage accounting (scheduler-side simulation) ----
+
+@dataclass
+class DeviceUsage:
+    """Accumulated usage of one device across assigned pods."""
+
+    info: DeviceInfo
+    used_number: int = 0
+    used_cores: int = 0
+    used_memory: int = 0      # MiB
+
+    def free_number(self) -> int:
+        return self.info.number - self.used_number
+
+    def free_cores(self) -> int:
+        return self.info.core - self.used_cores
+
+    def free_memory(self) -> int:
+        return self.info.memory - self.used_memory
+
+
+class NodeInfo:
+    """A node's devices + usage, built from the node annotations and the
+    assigned pods' real/pre-allocation claims (reference NodeInfo,
+    types.go:752-)."""
+
+    def __init__(self, name: str, devices: List[DeviceInfo],
+                 topology: Optional[NodeTopologyInfo] = None):
+        self.name = name
+        self.devices: Dict[int, DeviceUsage] = {
+            d.id: DeviceUsage(info=d) for d in devices
+        }
+        self.topology = topology
+
+    def add_pod_claims(self, cdcs: List[ContainerDeviceClaim]) -> None:
+        for cdc in cdcs:
+            for c in cdc.claims:
+                u = self.devices.get(c.id)
+                if u is None:
+                    continue
+                u.used_number += 1
+                u.used_cores += c.cores
+                u.used_memory += c.memory
+
+    def remove_pod_claims(self, cdcs: List[ContainerDeviceClaim]) -> None:
+        for cdc in cdcs:
+            for c in cdc.claims:
+                u = self.devices.get(c.id)
+                if u is None:
+                    continue
+                u.used_number -= 1
+                u.used_cores -= c.cores
+                u.used_memory -= c.memory
+
+    def clone(self) -> "NodeInfo":
+        n = NodeInfo(self.name, [], self.topology)
+        for did, u in self.devices.items():
+            n.devices[did] = DeviceUsage(info=u.info,
+                                         used_number=u.used_number,
+                                         used_cores=u.used_cores,
+                                         used_memory=u.used_memory)
+        return n
+
+
+# ---- fakes for tests (reference NewFakeDevice / NewFakeNodeInfo) ----
+
+def fake_device(idx: int, *, uuid: str = "", memory: int = 294912,
+                core: int = 100, number: int = 10, numa: int = -1,
+                healthy: bool = True, dtype: str = "MI355X") -> DeviceInfo:
+    return DeviceInfo(id=idx, uuid=uuid or f"GPU-fake-{idx:04d}",
+                      memory=memory, core=core, number=number, numa=numa,
+                      healthy=healthy, type=dtype)
+
+
+def fake_node(name: str, n_devices: int = 8, *, numa_split: int = 2,
+              full_xgmi: bool = True, **dev_kwargs) -> NodeInfo:
+    """An 8-GPU MI355X node: fully-connected xGMI mesh, 2 NUMA domains
+    of 4 GPUs (the standard OAM baseboard shape)."""
+    devs = []
+    topo = NodeTopologyInfo()
+    for i in range(n_devices):
+        numa = i // max(1, (n_devices // numa_split)) if numa_split else -1
+        devs.append(fake_device(i, numa=numa, **dev_kwargs))
+        dt = DeviceTopology(id=i, uuid=f"GPU-fake-{i:04d}", numa=numa)
+        for j in range(n_devices):
+            if i == j:
+                continue
+            if full_xgmi:
+                dt.links[j] = DeviceLink(peer_id=j, kind=LINK_XGMI,
+                                         weight=15, hops=1)
+            else:
+                same = (j // max(1, (n_devices // numa_split))) == numa
+                dt.links[j] = DeviceLink(
+                    peer_id=j, kind=LINK_PCIE_NUMA if same else LINK_SYS,
+                    weight=40 if same else 80, hops=1 if same else 2)
+        topo.devices.append(dt)
+    return NodeInfo(name, devs, topo)
+
+
+def now_rfc3339() -> str:
+    return time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
