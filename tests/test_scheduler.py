@@ -1,0 +1,206 @@
+"""Scheduler extender tests with the fake kube client (reference
+pkg/scheduler/filter tests use client-go's fake clientset the same
+way)."""
+import json
+
+import pytest
+
+from vgpu_manager_amd.client.kube import FakeKubeClient
+from vgpu_manager_amd.device.types import (
+    encode_node_devices,
+    fake_device,
+)
+from vgpu_manager_amd.scheduler.bind import NodeBinder
+from vgpu_manager_amd.scheduler.filter import GpuFilter, R_NODE_NOT_VGPU
+from vgpu_manager_amd.scheduler.preempt import VgpuPreempter
+from vgpu_manager_amd.util import consts
+
+from tests.test_allocator import make_pod
+
+
+def make_node(name, n_gpus=2, memory=294912):
+    return {
+        "metadata": {
+            "name": name,
+            "annotations": {
+                consts.node_register_ann(): encode_node_devices(
+                    [fake_device(i, memory=memory) for i in range(n_gpus)]),
+            },
+        },
+    }
+
+
+@pytest.fixture
+def client():
+    c = FakeKubeClient()
+    c.add_node(make_node("gpu-node-1"))
+    c.add_node(make_node("gpu-node-2"))
+    c.add_node({"metadata": {"name": "cpu-node", "annotations": {}}})
+    return c
+
+
+def filter_args(pod, nodes):
+    return {"Pod": pod, "NodeNames": nodes}
+
+
+def test_filter_selects_gpu_node_and_patches(client):
+    pod = make_pod(number=1, cores=50, memory=4096, name="w1")
+    client.add_pod(pod)
+    f = GpuFilter(client)
+    res = f.filter(filter_args(pod, ["cpu-node", "gpu-node-1",
+                                     "gpu-node-2"]))
+    assert res["Error"] == ""
+    assert len(res["NodeNames"]) == 1
+    assert res["NodeNames"][0].startswith("gpu-node")
+    assert res["FailedNodes"]["cpu-node"] == R_NODE_NOT_VGPU
+    patched = client.get_pod("default", "w1")
+    ann = patched["metadata"]["annotations"]
+    assert consts.pre_alloc_ann() in ann
+    assert ann[consts.predicate_node_ann()] == res["NodeNames"][0]
+    assert patched["metadata"]["labels"][consts.assigned_phase_label()] \
+        == consts.PHASE_ALLOCATING
+
+
+def test_filter_dryrun_does_not_patch(client):
+    pod = make_pod(number=1, name="w2")
+    client.add_pod(pod)
+    res = GpuFilter(client).filter(filter_args(pod, ["gpu-node-1"]),
+                                   dry_run=True)
+    assert res["NodeNames"] == ["gpu-node-1"]
+    ann = client.get_pod("default", "w2")["metadata"].get(
+        "annotations", {})
+    assert consts.pre_alloc_ann() not in ann
+
+
+def test_filter_non_vgpu_pod_passthrough(client):
+    pod = {"metadata": {"name": "plain", "annotations": {}},
+           "spec": {"containers": [{"name": "c", "resources": {}}]}}
+    res = GpuFilter(client).filter(
+        filter_args(pod, ["cpu-node", "gpu-node-1"]))
+    assert set(res["NodeNames"]) == {"cpu-node", "gpu-node-1"}
+
+
+def test_filter_capacity_rejection(client):
+    # node has 2 GPUs; ask for 4
+    pod = make_pod(number=4, name="big")
+    client.add_pod(pod)
+    res = GpuFilter(client).filter(
+        filter_args(pod, ["gpu-node-1", "gpu-node-2"]))
+    assert res["NodeNames"] == []
+    assert len(res["FailedNodes"]) == 2
+
+
+def test_filter_binpack_node_choice(client):
+    # put an existing claimed pod on gpu-node-1 -> binpack prefers it
+    existing = make_pod(number=1, cores=10, memory=1024, name="e")
+    existing["spec"]["nodeName"] = "gpu-node-1"
+    existing["metadata"]["annotations"][consts.real_alloc_ann()] = \
+        "e[0_GPU-fake-0000_10_1024]"
+    client.add_pod(existing)
+    pod = make_pod(number=1, name="w3")
+    client.add_pod(pod)
+    res = GpuFilter(client).filter(
+        filter_args(pod, ["gpu-node-1", "gpu-node-2"]))
+    assert res["NodeNames"] == ["gpu-node-1"]
+
+
+def test_bind_verifies_predicate_node(client):
+    pod = make_pod(number=1, name="w4")
+    pod["metadata"]["annotations"][consts.pre_alloc_ann()] = \
+        "main[0_GPU-fake-0000_0_1024]"
+    pod["metadata"]["annotations"][consts.predicate_node_ann()] = \
+        "gpu-node-1"
+    client.add_pod(pod)
+    b = NodeBinder(client)
+    res = b.bind({"PodName": "w4", "PodNamespace": "default",
+                  "Node": "gpu-node-2"})
+    assert "!=" in res["Error"]
+    res = b.bind({"PodName": "w4", "PodNamespace": "default",
+                  "Node": "gpu-node-1"})
+    assert res["Error"] == ""
+    assert client.bindings == [("default", "w4", "gpu-node-1")]
+
+
+def test_bind_rejects_without_preallocation(client):
+    pod = make_pod(number=1, name="w5")
+    client.add_pod(pod)
+    res = NodeBinder(client).bind({"PodName": "w5",
+                                   "PodNamespace": "default",
+                                   "Node": "gpu-node-1"})
+    assert "no pre-allocated" in res["Error"]
+
+
+def test_preempt_keeps_sufficient_victims(client):
+    # fill gpu-node-1 entirely with a victim pod
+    victim = make_pod(number=2, name="victim")
+    victim["spec"]["nodeName"] = "gpu-node-1"
+    victim["spec"]["priority"] = 0
+    victim["metadata"]["uid"] = "uid-victim"
+    victim["metadata"]["annotations"][consts.real_alloc_ann()] = \
+        "main[0_GPU-fake-0000_100_294912,1_GPU-fake-0001_100_294912]"
+    client.add_pod(victim)
+    pending = make_pod(number=1, name="pending")
+    pending["spec"] = {**pending["spec"], "priority": 100}
+    res = VgpuPreempter(client).preempt({
+        "Pod": pending,
+        "NodeNameToVictims": {
+            "gpu-node-1": {"Pods": [victim], "NumPDBViolations": 0}},
+    })
+    meta = res["NodeNameToMetaVictims"]
+    assert "gpu-node-1" in meta
+    assert meta["gpu-node-1"]["Pods"][0]["UID"] == "uid-victim"
+
+
+def test_preempt_drops_unfixable_node(client):
+    # pending wants 4 GPUs; node only has 2 even empty
+    pending = make_pod(number=4, name="pending4")
+    victim = make_pod(number=1, name="v2")
+    victim["spec"]["nodeName"] = "gpu-node-1"
+    victim["metadata"]["annotations"][consts.real_alloc_ann()] = \
+        "main[0_GPU-fake-0000_0_1024]"
+    client.add_pod(victim)
+    res = VgpuPreempter(client).preempt({
+        "Pod": pending,
+        "NodeNameToVictims": {
+            "gpu-node-1": {"Pods": [victim], "NumPDBViolations": 0}},
+    })
+    assert res["NodeNameToMetaVictims"] == {}
+
+
+def test_preempt_finds_additional_victims(client):
+    # two 1-GPU victims each holding a full GPU; proposed set has only
+    # one; pending needs both GPUs -> extra victim found
+    for i, name in enumerate(["v-a", "v-b"]):
+        v = make_pod(number=1, name=name)
+        v["spec"]["nodeName"] = "gpu-node-1"
+        v["spec"]["priority"] = 0
+        v["metadata"]["uid"] = f"uid-{name}"
+        v["metadata"]["annotations"][consts.real_alloc_ann()] = \
+            f"main[{i}_GPU-fake-{i:04d}_100_294912]"
+        client.add_pod(v)
+    pending = make_pod(number=2, name="pending2")
+    pending["spec"]["priority"] = 100
+    res = VgpuPreempter(client).preempt({
+        "Pod": pending,
+        "NodeNameToVictims": {
+            "gpu-node-1": {"Pods": [client.get_pod("default", "v-a")],
+                           "NumPDBViolations": 0}},
+    })
+    pods = res["NodeNameToMetaVictims"]["gpu-node-1"]["Pods"]
+    assert {p["UID"] for p in pods} == {"uid-v-a", "uid-v-b"}
+
+
+def test_http_routes(client):
+    from starlette.testclient import TestClient
+    from vgpu_manager_amd.scheduler.http import create_app
+
+    app = create_app(client)
+    tc = TestClient(app)
+    assert tc.get("/healthz").json()["status"] == "ok"
+    assert "version" in tc.get("/version").json()
+    pod = make_pod(number=1, name="hw")
+    client.add_pod(pod)
+    r = tc.post("/scheduler/filter",
+                json={"Pod": pod, "NodeNames": ["gpu-node-1"]})
+    assert r.status_code == 200
+    assert r.json()["NodeNames"] == ["gpu-node-1"]

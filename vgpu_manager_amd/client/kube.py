@@ -1,0 +1,274 @@
+"""Kubernetes API access.
+
+`KubeClient` is the minimal surface the control plane needs (reference
+pkg/client): node/pod reads, pod annotation patches, bindings,
+evictions.  `RestKubeClient` talks to a real apiserver with `requests`
+(in-cluster service account or kubeconfig-less URL+token);
+`FakeKubeClient` is the in-memory double every test uses (reference
+tests use client-go's fake clientset the same way).
+
+A mutation-aware pod view bridges informer lag (reference
+pod_lister.go:62): patches applied through this client are visible to
+subsequent reads immediately even when a stale list layer backs it.
+"""
+from __future__ import annotations
+
+import json
+import os
+import threading
+import time
+from typing import Callable, Dict, List, Optional
+
+
+class KubeError(Exception):
+    pass
+
+
+class KubeClient:
+    # nodes
+    def get_node(self, name: str) -> dict:
+        raise NotImplementedError
+
+    def list_nodes(self) -> List[dict]:
+        raise NotImplementedError
+
+    def patch_node_annotations(self, name: str, annotations: Dict[str, str]
+                               ) -> None:
+        raise NotImplementedError
+
+    # pods
+    def get_pod(self, namespace: str, name: str) -> dict:
+        raise NotImplementedError
+
+    def list_pods(self, node_name: Optional[str] = None,
+                  label_selector: Optional[Dict[str, str]] = None
+                  ) -> List[dict]:
+        raise NotImplementedError
+
+    def patch_pod_metadata(self, namespace: str, name: str,
+                           annotations: Optional[Dict[str, str]] = None,
+                           labels: Optional[Dict[str, str]] = None) -> None:
+        raise NotImplementedError
+
+    def create_binding(self, namespace: str, name: str, node: str) -> None:
+        raise NotImplementedError
+
+    def evict_pod(self, namespace: str, name: str) -> None:
+        raise NotImplementedError
+
+    def delete_pod(self, namespace: str, name: str) -> None:
+        raise NotImplementedError
+
+    def create_event(self, namespace: str, involved: dict, reason: str,
+                     message: str, etype: str = "Warning") -> None:
+        pass
+
+
+def _match_labels(meta: dict, selector: Optional[Dict[str, str]]) -> bool:
+    if not selector:
+        return True
+    labels = meta.get("labels", {}) or {}
+    return all(labels.get(k) == v for k, v in selector.items())
+
+
+class FakeKubeClient(KubeClient):
+    """In-memory apiserver double; thread-safe."""
+
+    def __init__(self):
+        self._mu = threading.RLock()
+        self.nodes: Dict[str, dict] = {}
+        self.pods: Dict[tuple, dict] = {}
+        self.bindings: List[tuple] = []
+        self.evictions: List[tuple] = []
+        self.events: List[dict] = []
+
+    # -- test helpers --
+    def add_node(self, node: dict) -> None:
+        with self._mu:
+            self.nodes[node["metadata"]["name"]] = node
+
+    def add_pod(self, pod: dict) -> None:
+        with self._mu:
+            meta = pod["metadata"]
+            self.pods[(meta.get("namespace", "default"),
+                       meta["name"])] = pod
+
+    # -- KubeClient --
+    def get_node(self, name):
+        with self._mu:
+            if name not in self.nodes:
+                raise KubeError(f"node {name} not found")
+            return json.loads(json.dumps(self.nodes[name]))
+
+    def list_nodes(self):
+        with self._mu:
+            return [json.loads(json.dumps(n)) for n in self.nodes.values()]
+
+    def patch_node_annotations(self, name, annotations):
+        with self._mu:
+            node = self.nodes.get(name)
+            if node is None:
+                raise KubeError(f"node {name} not found")
+            node.setdefault("metadata", {}).setdefault(
+                "annotations", {}).update(annotations)
+
+    def get_pod(self, namespace, name):
+        with self._mu:
+            pod = self.pods.get((namespace, name))
+            if pod is None:
+                raise KubeError(f"pod {namespace}/{name} not found")
+            return json.loads(json.dumps(pod))
+
+    def list_pods(self, node_name=None, label_selector=None):
+        with self._mu:
+            out = []
+            for pod in self.pods.values():
+                if node_name and pod.get("spec", {}).get("nodeName") != \
+                        node_name:
+                    continue
+                if not _match_labels(pod.get("metadata", {}),
+                                     label_selector):
+                    continue
+                out.append(json.loads(json.dumps(pod)))
+            return out
+
+    def patch_pod_metadata(self, namespace, name, annotations=None,
+                           labels=None):
+        with self._mu:
+            pod = self.pods.get((namespace, name))
+            if pod is None:
+                raise KubeError(f"pod {namespace}/{name} not found")
+            meta = pod.setdefault("metadata", {})
+            if annotations:
+                meta.setdefault("annotations", {}).update(annotations)
+            if labels:
+                meta.setdefault("labels", {}).update(labels)
+
+    def create_binding(self, namespace, name, node):
+        with self._mu:
+            pod = self.pods.get((namespace, name))
+            if pod is None:
+                raise KubeError(f"pod {namespace}/{name} not found")
+            pod.setdefault("spec", {})["nodeName"] = node
+            self.bindings.append((namespace, name, node))
+
+    def evict_pod(self, namespace, name):
+        with self._mu:
+            self.evictions.append((namespace, name))
+            self.pods.pop((namespace, name), None)
+
+    def delete_pod(self, namespace, name):
+        with self._mu:
+            self.pods.pop((namespace, name), None)
+
+    def create_event(self, namespace, involved, reason, message,
+                     etype="Warning"):
+        with self._mu:
+            self.events.append(dict(namespace=namespace, reason=reason,
+                                    message=message, type=etype))
+
+
+class RestKubeClient(KubeClient):
+    """Thin apiserver REST client (in-cluster or explicit URL/token)."""
+
+    def __init__(self, base_url: Optional[str] = None,
+                 token: Optional[str] = None, verify=None):
+        import requests
+        self._requests = requests
+        sa = "/var/run/secrets/kubernetes.io/serviceaccount"
+        if base_url is None:
+            host = os.environ.get("KUBERNETES_SERVICE_HOST", "kubernetes")
+            port = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
+            base_url = f"https://{host}:{port}"
+        if token is None and os.path.exists(f"{sa}/token"):
+            token = open(f"{sa}/token").read().strip()
+        if verify is None:
+            verify = f"{sa}/ca.crt" if os.path.exists(f"{sa}/ca.crt") \
+                else False
+        self.base = base_url.rstrip("/")
+        self.session = requests.Session()
+        if token:
+            self.session.headers["Authorization"] = f"Bearer {token}"
+        self.verify = verify
+
+    def _req(self, method, path, body=None, content_type=None):
+        headers = {}
+        if content_type:
+            headers["Content-Type"] = content_type
+        r = self.session.request(method, f"{self.base}{path}",
+                                 json=body, headers=headers,
+                                 verify=self.verify, timeout=30)
+        if r.status_code >= 300:
+            raise KubeError(f"{method} {path}: {r.status_code} {r.text[:200]}")
+        return r.json() if r.text else {}
+
+    def get_node(self, name):
+        return self._req("GET", f"/api/v1/nodes/{name}")
+
+    def list_nodes(self):
+        return self._req("GET", "/api/v1/nodes").get("items", [])
+
+    def patch_node_annotations(self, name, annotations):
+        body = {"metadata": {"annotations": annotations}}
+        self._req("PATCH", f"/api/v1/nodes/{name}", body,
+                  "application/strategic-merge-patch+json")
+
+    def get_pod(self, namespace, name):
+        return self._req("GET",
+                         f"/api/v1/namespaces/{namespace}/pods/{name}")
+
+    def list_pods(self, node_name=None, label_selector=None):
+        q = []
+        if node_name:
+            q.append(f"fieldSelector=spec.nodeName%3D{node_name}")
+        if label_selector:
+            sel = ",".join(f"{k}%3D{v}" for k, v in label_selector.items())
+            q.append(f"labelSelector={sel}")
+        qs = ("?" + "&".join(q)) if q else ""
+        return self._req("GET", f"/api/v1/pods{qs}").get("items", [])
+
+    def patch_pod_metadata(self, namespace, name, annotations=None,
+                           labels=None):
+        meta = {}
+        if annotations:
+            meta["annotations"] = annotations
+        if labels:
+            meta["labels"] = labels
+        self._req("PATCH", f"/api/v1/namespaces/{namespace}/pods/{name}",
+                  {"metadata": meta},
+                  "application/strategic-merge-patch+json")
+
+    def create_binding(self, namespace, name, node):
+        body = {"apiVersion": "v1", "kind": "Binding",
+                "metadata": {"name": name},
+                "target": {"apiVersion": "v1", "kind": "Node",
+                           "name": node}}
+        self._req("POST",
+                  f"/api/v1/namespaces/{namespace}/pods/{name}/binding",
+                  body)
+
+    def evict_pod(self, namespace, name):
+        body = {"apiVersion": "policy/v1", "kind": "Eviction",
+                "metadata": {"name": name, "namespace": namespace}}
+        self._req("POST",
+                  f"/api/v1/namespaces/{namespace}/pods/{name}/eviction",
+                  body)
+
+    def delete_pod(self, namespace, name):
+        self._req("DELETE", f"/api/v1/namespaces/{namespace}/pods/{name}")
+
+    def create_event(self, namespace, involved, reason, message,
+                     etype="Warning"):
+        now = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+        body = {
+            "metadata": {"generateName": "vgpu-manager-"},
+            "involvedObject": involved, "reason": reason,
+            "message": message[:1024], "type": etype,
+            "firstTimestamp": now, "lastTimestamp": now,
+            "source": {"component": "vgpu-scheduler"},
+        }
+        try:
+            self._req("POST", f"/api/v1/namespaces/{namespace}/events",
+                      body)
+        except KubeError:
+            pass

@@ -1,0 +1,146 @@
+"""The scheduler-extender Filter verb (reference
+pkg/scheduler/filter/filter_predicate.go, re-designed in Python).
+
+Flow: decode the pod's device request -> gate nodes (vgpu-enabled,
+memory policy) -> build per-node usage snapshots -> sort nodes by
+policy fitness -> simulate the allocator per node in order -> patch the
+winner pod with pre-allocated + predicate-node + assigned-phase.
+
+Dry-run mode (`/scheduler/filter-dryrun`, reference
+ca_extender_dryrun_filter_design.md) runs the same simulation without
+patching — used by Cluster Autoscaler scale-up simulation.
+"""
+from __future__ import annotations
+
+import logging
+import time
+from typing import Dict, List, Optional, Tuple
+
+from ..client.kube import KubeClient, KubeError
+from ..device.allocator import (
+    AllocationError,
+    Allocator,
+    build_allocation_request,
+)
+from ..device.types import marshal_pod_claim
+from ..util import consts
+from . import metrics
+from .serial import KeyedLocker
+from .snapshot import build_node_info
+
+log = logging.getLogger("vgpu.scheduler.filter")
+
+R_NODE_NOT_VGPU = "NodeNotVGPUEnabled"
+R_INTERNAL = "InternalError"
+
+
+class GpuFilter:
+    def __init__(self, client: KubeClient, serialize: bool = True):
+        self.client = client
+        self.locker = KeyedLocker() if serialize else None
+
+    # ---- node fitness sort (reference sortNodeInfos/priority.go) ----
+    def _node_score(self, info, policy: str) -> float:
+        total_free = sum(d.free_cores() + d.free_memory() // 1024
+                         for d in info.devices.values())
+        # binpack: prefer fuller nodes (lower free); spread: emptier
+        return total_free if policy == consts.POLICY_BINPACK \
+            else -total_free
+
+    def filter(self, args: dict, dry_run: bool = False) -> dict:
+        t0 = time.monotonic()
+        pod = args.get("Pod") or args.get("pod") or {}
+        meta = pod.get("metadata", {})
+        pod_name = meta.get("name", "?")
+        namespace = meta.get("namespace", "default")
+
+        node_names = self._candidate_nodes(args)
+        failed: Dict[str, str] = {}
+
+        try:
+            request = build_allocation_request(pod)
+        except AllocationError as e:
+            return self._result([], {n: e.reason for n in node_names},
+                                error=str(e))
+        if not request.containers:
+            # not a vGPU pod: pass everything through
+            return self._result(node_names, {})
+
+        policy_ann = (meta.get("annotations", {}) or {}).get(
+            consts.node_scheduler_policy_ann(), consts.POLICY_BINPACK)
+
+        lock_key = "global-filter"
+        if self.locker and not dry_run:
+            self.locker.acquire(lock_key)
+        try:
+            candidates: List[Tuple[float, str, list]] = []
+            for name in node_names:
+                try:
+                    node = self.client.get_node(name)
+                except KubeError:
+                    failed[name] = R_INTERNAL
+                    continue
+                pods = self.client.list_pods(node_name=name)
+                info = build_node_info(node, pods)
+                if info is None:
+                    failed[name] = R_NODE_NOT_VGPU
+                    continue
+                candidates.append((self._node_score(info, policy_ann),
+                                   name, [node, pods]))
+
+            candidates.sort(key=lambda c: (c[0], c[1]))
+            chosen: Optional[str] = None
+            claims = None
+            for _, name, (node, pods) in candidates:
+                info = build_node_info(node, pods)
+                try:
+                    claims = Allocator(info).allocate(request)
+                    chosen = name
+                    break
+                except AllocationError as e:
+                    failed[name] = e.reason
+
+            if chosen is None:
+                metrics.observe("filter", time.monotonic() - t0, False)
+                return self._result([], failed)
+
+            if not dry_run:
+                text = marshal_pod_claim(claims)
+                try:
+                    self.client.patch_pod_metadata(
+                        namespace, pod_name,
+                        annotations={
+                            consts.pre_alloc_ann(): text,
+                            consts.predicate_node_ann(): chosen,
+                            consts.predicate_time_ann():
+                                str(int(time.time())),
+                        },
+                        labels={consts.assigned_phase_label():
+                                consts.PHASE_ALLOCATING})
+                except KubeError as e:
+                    metrics.observe("filter", time.monotonic() - t0, False)
+                    return self._result([], failed, error=str(e))
+
+            metrics.observe("filter", time.monotonic() - t0, True)
+            return self._result([chosen], failed)
+        finally:
+            if self.locker and not dry_run:
+                self.locker.release(lock_key)
+
+    @staticmethod
+    def _candidate_nodes(args: dict) -> List[str]:
+        names = args.get("NodeNames") or args.get("nodenames")
+        if names:
+            return list(names)
+        nodes = (args.get("Nodes") or {}).get("Items") or []
+        return [n["metadata"]["name"] for n in nodes]
+
+    @staticmethod
+    def _result(node_names: List[str], failed: Dict[str, str],
+                error: str = "") -> dict:
+        return {
+            "Nodes": None,
+            "NodeNames": node_names,
+            "FailedNodes": failed,
+            "Error": error,
+        }

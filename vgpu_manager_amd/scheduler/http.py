@@ -1,0 +1,78 @@
+"""HTTP wiring of the scheduler extender (reference pkg/route):
+POST /scheduler/filter, /scheduler/filter-dryrun, /scheduler/bind,
+/scheduler/preempt; plus healthz/version/metrics.
+"""
+from __future__ import annotations
+
+import json
+
+from fastapi import FastAPI, Request, Response
+
+from ..client.kube import KubeClient
+from ..version import VERSION
+from .bind import NodeBinder
+from .filter import GpuFilter
+from .preempt import VgpuPreempter
+
+
+def create_app(client: KubeClient) -> FastAPI:
+    app = FastAPI(title="vgpu-scheduler-extender", version=VERSION)
+    gpu_filter = GpuFilter(client)
+    binder = NodeBinder(client)
+    preempter = VgpuPreempter(client)
+
+    @app.post("/scheduler/filter")
+    async def filter_(request: Request):
+        args = await request.json()
+        return gpu_filter.filter(args, dry_run=False)
+
+    @app.post("/scheduler/filter-dryrun")
+    async def filter_dryrun(request: Request):
+        args = await request.json()
+        return gpu_filter.filter(args, dry_run=True)
+
+    @app.post("/scheduler/bind")
+    async def bind(request: Request):
+        args = await request.json()
+        return binder.bind(args)
+
+    @app.post("/scheduler/preempt")
+    async def preempt(request: Request):
+        args = await request.json()
+        return preempter.preempt(args)
+
+    @app.get("/healthz")
+    async def healthz():
+        return {"status": "ok"}
+
+    @app.get("/version")
+    async def version():
+        return {"version": VERSION}
+
+    @app.get("/metrics")
+    async def metrics_endpoint():
+        try:
+            from prometheus_client import generate_latest
+            return Response(generate_latest(), media_type="text/plain")
+        except Exception:
+            return Response("", media_type="text/plain")
+
+    return app
+
+
+def main():  # pragma: no cover
+    import argparse
+    import uvicorn
+    from ..client.kube import RestKubeClient
+
+    ap = argparse.ArgumentParser("vgpu-device-scheduler")
+    ap.add_argument("--bind", default="0.0.0.0:3456")
+    ap.add_argument("--apiserver", default=None)
+    args = ap.parse_args()
+    host, port = args.bind.rsplit(":", 1)
+    app = create_app(RestKubeClient(base_url=args.apiserver))
+    uvicorn.run(app, host=host, port=int(port))
+
+
+if __name__ == "__main__":  # pragma: no cover
+    main()

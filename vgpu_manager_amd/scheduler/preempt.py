@@ -1,0 +1,147 @@
+"""Preempt verb (reference pkg/scheduler/preempt/preempt_predicate.go).
+
+For every candidate node with a proposed victim set: simulate the node
+without the victims; keep the set if the pending pod then fits; search
+additional lower-priority victims when it does not; drop the node when
+even that fails.  Any internal error degrades to passthrough — the
+extender must never veto in-tree preemption by crashing (reference
+:168 passthrough contract).
+"""
+from __future__ import annotations
+
+import logging
+import time
+from typing import Dict, List, Optional
+
+from ..client.kube import KubeClient, KubeError
+from ..device.allocator import (
+    AllocationError,
+    Allocator,
+    build_allocation_request,
+)
+from ..util import consts
+from . import metrics
+from .snapshot import build_node_info, pod_claim_annotation
+
+log = logging.getLogger("vgpu.scheduler.preempt")
+
+
+def _pod_priority(pod: dict) -> int:
+    return pod.get("spec", {}).get("priority", 0) or 0
+
+
+class VgpuPreempter:
+    def __init__(self, client: KubeClient):
+        self.client = client
+
+    def preempt(self, args: dict) -> dict:
+        t0 = time.monotonic()
+        try:
+            result = self._preempt_inner(args)
+            metrics.observe("preempt", time.monotonic() - t0, True)
+            return result
+        except Exception as e:  # passthrough on any internal error
+            log.warning("preempt passthrough on error: %s", e)
+            metrics.observe("preempt", time.monotonic() - t0, False)
+            victims = args.get("NodeNameToVictims") or {}
+            meta = args.get("NodeNameToMetaVictims") or {}
+            return {"NodeNameToMetaVictims": meta or
+                    self._to_meta(victims)}
+
+    # ---- inner ----
+    def _preempt_inner(self, args: dict) -> dict:
+        pod = args.get("Pod") or {}
+        request = build_allocation_request(pod)
+        victims_by_node: Dict[str, dict] = dict(
+            args.get("NodeNameToVictims") or {})
+        if not victims_by_node:
+            victims_by_node = self._from_meta(
+                args.get("NodeNameToMetaVictims") or {})
+        if not request.containers:
+            return {"NodeNameToMetaVictims":
+                    self._to_meta(victims_by_node)}
+
+        out: Dict[str, dict] = {}
+        for node_name, victims in victims_by_node.items():
+            refined = self._refine_for_node(node_name, pod, request,
+                                            victims)
+            if refined is not None:
+                out[node_name] = refined
+        return {"NodeNameToMetaVictims": self._to_meta(out)}
+
+    def _refine_for_node(self, node_name: str, pending: dict, request,
+                         victims: dict) -> Optional[dict]:
+        try:
+            node = self.client.get_node(node_name)
+        except KubeError:
+            return victims  # passthrough for unknown nodes
+        pods = self.client.list_pods(node_name=node_name)
+
+        victim_pods = list(victims.get("Pods") or [])
+        victim_keys = {(p.get("metadata", {}).get("namespace", "default"),
+                        p.get("metadata", {}).get("name"))
+                       for p in victim_pods}
+
+        def can_allocate(extra_removed: set) -> bool:
+            removed = victim_keys | extra_removed
+            kept = [p for p in pods
+                    if (p.get("metadata", {}).get("namespace", "default"),
+                        p.get("metadata", {}).get("name")) not in removed]
+            info = build_node_info(node, kept)
+            if info is None:
+                return False
+            try:
+                Allocator(info).allocate(request)
+                return True
+            except AllocationError:
+                return False
+
+        if can_allocate(set()):
+            return victims
+
+        # search additional lower-priority vGPU victims
+        pending_prio = _pod_priority(pending)
+        extra_candidates = sorted(
+            (p for p in pods
+             if pod_claim_annotation(p)
+             and _pod_priority(p) < pending_prio
+             and (p.get("metadata", {}).get("namespace", "default"),
+                  p.get("metadata", {}).get("name")) not in victim_keys),
+            key=_pod_priority)
+        extra: List[dict] = []
+        extra_keys: set = set()
+        for p in extra_candidates:
+            extra.append(p)
+            extra_keys.add((p.get("metadata", {}).get("namespace",
+                                                      "default"),
+                            p.get("metadata", {}).get("name")))
+            if can_allocate(extra_keys):
+                return {"Pods": victim_pods + extra,
+                        "NumPDBViolations":
+                            victims.get("NumPDBViolations", 0)}
+        return None  # node cannot host the pod even with extra victims
+
+    # ---- meta codecs (extender v1 MetaVictims carry only UIDs) ----
+    @staticmethod
+    def _to_meta(by_node: Dict[str, dict]) -> Dict[str, dict]:
+        out = {}
+        for node, victims in by_node.items():
+            out[node] = {
+                "Pods": [{"UID": p.get("metadata", {}).get("uid", "")}
+                         for p in victims.get("Pods") or []],
+                "NumPDBViolations": victims.get("NumPDBViolations", 0),
+            }
+        return out
+
+    def _from_meta(self, meta: Dict[str, dict]) -> Dict[str, dict]:
+        out = {}
+        for node, victims in meta.items():
+            pods = []
+            uids = {p.get("UID") for p in victims.get("Pods") or []}
+            for p in self.client.list_pods(node_name=node):
+                if p.get("metadata", {}).get("uid") in uids:
+                    pods.append(p)
+            out[node] = {"Pods": pods,
+                         "NumPDBViolations":
+                             victims.get("NumPDBViolations", 0)}
+        return out

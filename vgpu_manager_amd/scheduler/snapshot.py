@@ -1,0 +1,63 @@
+"""Builds scheduler-side NodeInfo snapshots from node annotations and
+the pods assigned to each node (reference preFilterNodeInfos,
+filter_predicate.go:690)."""
+from __future__ import annotations
+
+from typing import List, Optional
+
+from ..device.types import (
+    NodeInfo,
+    NodeTopologyInfo,
+    decode_node_devices,
+    unmarshal_pod_claim,
+)
+from ..util import consts
+
+
+def node_device_annotation(node: dict) -> Optional[str]:
+    return (node.get("metadata", {}).get("annotations", {}) or {}).get(
+        consts.node_register_ann())
+
+
+def build_node_info(node: dict, pods: List[dict]) -> Optional[NodeInfo]:
+    """Returns None if the node is not vgpu-enabled (no register ann)."""
+    ann = node.get("metadata", {}).get("annotations", {}) or {}
+    reg = ann.get(consts.node_register_ann())
+    if not reg:
+        return None
+    try:
+        devices = decode_node_devices(reg)
+    except (ValueError, KeyError):
+        return None
+    topo = None
+    topo_ann = ann.get(consts.node_topology_ann())
+    if topo_ann:
+        try:
+            topo = NodeTopologyInfo.decode(topo_ann)
+        except (ValueError, KeyError):
+            topo = None
+    info = NodeInfo(node["metadata"]["name"], devices, topo)
+    for pod in pods:
+        add_pod_usage(info, pod)
+    return info
+
+
+def pod_claim_annotation(pod: dict) -> Optional[str]:
+    """Real allocation wins over pre-allocation (reference types.go)."""
+    ann = pod.get("metadata", {}).get("annotations", {}) or {}
+    return ann.get(consts.real_alloc_ann()) or ann.get(
+        consts.pre_alloc_ann())
+
+
+def add_pod_usage(info: NodeInfo, pod: dict) -> None:
+    phase = pod.get("status", {}).get("phase", "Running")
+    if phase in ("Succeeded", "Failed"):
+        return
+    text = pod_claim_annotation(pod)
+    if not text:
+        return
+    try:
+        cdcs = unmarshal_pod_claim(text)
+    except ValueError:
+        return
+    info.add_pod_claims(cdcs)
