@@ -1,0 +1,216 @@
+"""Device-plugin tests: pbwire codec, vnum Allocate/Preferred flows over
+a REAL grpc unix socket (fake manager + fake kube client), config
+region output, PreStartContainer cleanup."""
+import json
+import os
+import threading
+import time
+from concurrent import futures
+
+import grpc
+import pytest
+
+from vgpu_manager_amd.client.kube import FakeKubeClient
+from vgpu_manager_amd.config.regions import VgpuConfigReader
+from vgpu_manager_amd.device.manager import FakeDeviceManager
+from vgpu_manager_amd.device.types import marshal_pod_claim
+from vgpu_manager_amd.deviceplugin import api
+from vgpu_manager_amd.deviceplugin.server import (
+    PluginServer,
+    QuantityPlugin,
+)
+from vgpu_manager_amd.deviceplugin.vnum_plugin import (
+    VnumPlugin,
+    fake_id,
+    parse_fake_id,
+)
+from vgpu_manager_amd.scheduler.filter import GpuFilter
+from vgpu_manager_amd.util import consts
+
+from tests.test_allocator import make_pod
+from tests.test_scheduler import make_node
+
+
+def test_pbwire_roundtrip():
+    req = api.AllocateRequest(container_requests=[
+        api.ContainerAllocateRequest(devices_ids=["a", "b"]),
+    ])
+    back = api.AllocateRequest.decode(req.encode())
+    assert back.container_requests[0].devices_ids == ["a", "b"]
+
+    resp = api.ContainerAllocateResponse(
+        envs={"K": "V", "X": "Y"},
+        mounts=[api.Mount(container_path="/c", host_path="/h",
+                          read_only=True)],
+        devices=[api.DeviceSpec(container_path="/dev/kfd",
+                                host_path="/dev/kfd",
+                                permissions="rw")])
+    back = api.ContainerAllocateResponse.decode(resp.encode())
+    assert back.envs == {"K": "V", "X": "Y"}
+    assert back.mounts[0].read_only is True
+    assert back.devices[0].permissions == "rw"
+
+    d = api.Device(ID="GPU-x::3", health=api.HEALTHY,
+                   topology=api.TopologyInfo(nodes=[api.NUMANode(ID=1)]))
+    back = api.Device.decode(d.encode())
+    assert back.topology.nodes[0].ID == 1
+
+
+@pytest.fixture
+def plugin_env(tmp_path):
+    client = FakeKubeClient()
+    client.add_node(make_node("node-a", n_gpus=2))
+    mgr = FakeDeviceManager("node-a", n_devices=2, numa_split=2)
+    plugin = VnumPlugin(mgr, client, base_dir=str(tmp_path / "etc"),
+                        driver_lib=str(tmp_path / "libvgpu-control.so"))
+    return client, mgr, plugin
+
+
+def schedule_pod(client, pod, node="node-a"):
+    """Run the real filter to produce the pre-allocation annotation."""
+    client.add_pod(pod)
+    res = GpuFilter(client).filter({"Pod": pod, "NodeNames": [node]})
+    assert res["NodeNames"] == [node], res
+    return client.get_pod(pod["metadata"].get("namespace", "default"),
+                          pod["metadata"]["name"])
+
+
+def test_fake_device_inventory(plugin_env):
+    _, mgr, plugin = plugin_env
+    devs = plugin.fake_devices()
+    # 2 GPUs x split 10
+    assert len(devs) == 20
+    assert devs[0].health == api.HEALTHY
+    uuid, k = parse_fake_id(devs[0].ID)
+    assert k == 0 and uuid.startswith("GPU-fake")
+
+
+def test_preferred_allocation_honors_preallocation(plugin_env):
+    client, mgr, plugin = plugin_env
+    pod = make_pod(number=1, cores=30, memory=2048, name="w")
+    pod["metadata"]["annotations"][consts.node_scheduler_policy_ann()] = \
+        consts.POLICY_BINPACK
+    scheduled = schedule_pod(client, pod)
+    claimed_uuid = scheduled["metadata"]["annotations"][
+        consts.pre_alloc_ann()].split("_")[1]
+
+    avail = [fake_id(d.uuid, k) for d in mgr.devices for k in range(3)]
+    req = api.PreferredAllocationRequest(container_requests=[
+        api.ContainerPreferredAllocationRequest(
+            available_device_ids=avail, allocation_size=1)])
+    resp = plugin.GetPreferredAllocation(req, None)
+    got = resp.container_responses[0].device_ids
+    assert len(got) == 1
+    assert parse_fake_id(got[0])[0] == claimed_uuid
+
+
+def test_allocate_writes_config_and_patches_pod(plugin_env, tmp_path):
+    client, mgr, plugin = plugin_env
+    pod = make_pod(number=1, cores=25, memory=4096, name="w2")
+    scheduled = schedule_pod(client, pod)
+    uuid = scheduled["metadata"]["annotations"][
+        consts.pre_alloc_ann()].split("_")[1]
+
+    req = api.AllocateRequest(container_requests=[
+        api.ContainerAllocateRequest(devices_ids=[fake_id(uuid, 0)])])
+    resp = plugin.Allocate(req, None)
+    cr = resp.container_responses[0]
+
+    # envs
+    assert cr.envs[consts.ENV_MEM_LIMIT.format(0)] == str(4096 << 20)
+    assert cr.envs[consts.ENV_CORE_LIMIT.format(0)] == "25"
+    assert cr.envs[consts.ENV_POD_UID] == "uid-w2"
+    # mounts include the shim + preload + config
+    cpaths = {m.container_path for m in cr.mounts}
+    assert "/etc/ld.so.preload" in cpaths
+    assert f"{consts.MANAGER_DIR}/config" in cpaths
+    assert "/tmp/.sm_node" in cpaths
+    # device nodes: kfd + the claimed GPU's renderD
+    dpaths = {d.container_path for d in cr.devices}
+    assert "/dev/kfd" in dpaths
+    assert any(p.startswith("/dev/dri/renderD") for p in dpaths)
+
+    # config region written and readable
+    cdir = plugin._container_dir("uid-w2", "main")
+    snap = VgpuConfigReader(
+        os.path.join(cdir, "config", "vgpu.config")).snapshot()
+    assert snap["pod_uid"] == "uid-w2"
+    assert snap["devices"][0]["total_memory"] == 4096 << 20
+    assert snap["devices"][0]["core_limit"] == 25
+    assert json.load(open(os.path.join(cdir, "devices.json")))[
+        "host_indices"] == [snap["devices"][0]["host_index"]]
+
+    # pod patched to success
+    patched = client.get_pod("default", "w2")
+    assert patched["metadata"]["labels"][consts.assigned_phase_label()] \
+        == consts.PHASE_SUCCESS
+    assert consts.real_alloc_ann() in patched["metadata"]["annotations"]
+
+
+def test_prestart_verifies_and_cleans(plugin_env):
+    client, mgr, plugin = plugin_env
+    pod = make_pod(number=1, memory=1024, name="w3")
+    scheduled = schedule_pod(client, pod)
+    uuid = scheduled["metadata"]["annotations"][
+        consts.pre_alloc_ann()].split("_")[1]
+    plugin.Allocate(api.AllocateRequest(container_requests=[
+        api.ContainerAllocateRequest(devices_ids=[fake_id(uuid, 0)])]),
+        None)
+    cdir = plugin._container_dir("uid-w3", "main")
+    stale = os.path.join(cdir, "vmem_node", "vmem_node.config")
+    open(stale, "w").write("stale")
+    plugin.PreStartContainer(api.PreStartContainerRequest(
+        devices_ids=[fake_id(uuid, 0)]), None)
+    assert not os.path.exists(stale)
+
+
+def test_allocate_without_pending_pod_fails_loudly(plugin_env):
+    _, mgr, plugin = plugin_env
+    with pytest.raises(RuntimeError):
+        plugin.Allocate(api.AllocateRequest(container_requests=[
+            api.ContainerAllocateRequest(
+                devices_ids=[fake_id(mgr.devices[0].uuid, 0)])]), None)
+
+
+def test_grpc_end_to_end(plugin_env, tmp_path):
+    """Serve the plugin on a real unix socket and call it through grpc
+    with the hand-rolled codec."""
+    client, mgr, plugin = plugin_env
+    server = PluginServer(plugin, "test-vgpu.sock",
+                          plugins_dir=str(tmp_path / "plugins"))
+    server.start()
+    try:
+        ch = grpc.insecure_channel(f"unix://{server.socket_path}")
+        opts = ch.unary_unary(
+            "/v1beta1.DevicePlugin/GetDevicePluginOptions",
+            request_serializer=lambda m: m.encode(),
+            response_deserializer=api.DevicePluginOptions.decode)(
+                api.Empty(), timeout=5)
+        assert opts.pre_start_required is True
+        assert opts.get_preferred_allocation_available is True
+
+        lw = ch.unary_stream(
+            "/v1beta1.DevicePlugin/ListAndWatch",
+            request_serializer=lambda m: m.encode(),
+            response_deserializer=api.ListAndWatchResponse.decode)(
+                api.Empty(), timeout=5)
+        first = next(iter(lw))
+        assert len(first.devices) == 20
+        ch.close()
+    finally:
+        server.stop()
+
+
+def test_health_change_propagates(plugin_env):
+    _, mgr, plugin = plugin_env
+    mgr.set_health(0, False)
+    devs = plugin.fake_devices()
+    unhealthy = [d for d in devs if d.health == api.UNHEALTHY]
+    assert len(unhealthy) == 10  # all fake ids of GPU 0
+
+
+def test_quantity_plugin():
+    qp = QuantityPlugin(consts.vgpu_core_resource(), 200)
+    resp = qp.Allocate(api.AllocateRequest(container_requests=[
+        api.ContainerAllocateRequest(devices_ids=["x"])]), None)
+    assert len(resp.container_responses) == 1

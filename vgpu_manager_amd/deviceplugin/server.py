@@ -1,0 +1,163 @@
+"""Device-plugin gRPC server scaffold (reference pkg/deviceplugin/base):
+serves a plugin on a unix socket under the kubelet plugins dir and
+registers with kubelet; restarts on kubelet socket recreation.
+"""
+from __future__ import annotations
+
+import logging
+import os
+import threading
+import time
+from concurrent import futures
+from typing import List, Optional
+
+import grpc
+
+from ..util import consts
+from . import api
+
+log = logging.getLogger("vgpu.deviceplugin.server")
+
+
+class QuantityPlugin:
+    """Quantity-only plugin (vgpu-cores / vgpu-memory): exposes capacity
+    so pods can request the resource; no allocation logic (reference
+    vcore_plugin.go / vmem_plugin.go)."""
+
+    def __init__(self, resource_name: str, count: int):
+        self.resource_name = resource_name
+        self.count = count
+        self._stopped = False
+
+    def GetDevicePluginOptions(self, request, context):
+        return api.DevicePluginOptions()
+
+    def ListAndWatch(self, request, context):
+        devices = [api.Device(ID=f"{self.resource_name}-{i}",
+                              health=api.HEALTHY)
+                   for i in range(self.count)]
+        while not self._stopped:
+            yield api.ListAndWatchResponse(devices=devices)
+            time.sleep(60)
+
+    def GetPreferredAllocation(self, request, context):
+        return api.PreferredAllocationResponse()
+
+    def Allocate(self, request, context):
+        resp = api.AllocateResponse()
+        for _ in request.container_requests:
+            resp.container_responses.append(
+                api.ContainerAllocateResponse())
+        return resp
+
+    def PreStartContainer(self, request, context):
+        return api.PreStartContainerResponse()
+
+    def stop(self):
+        self._stopped = True
+
+
+class PluginServer:
+    def __init__(self, plugin, endpoint_name: str,
+                 plugins_dir: str = api.PLUGINS_DIR):
+        self.plugin = plugin
+        self.endpoint_name = endpoint_name
+        self.plugins_dir = plugins_dir
+        self.socket_path = os.path.join(plugins_dir, endpoint_name)
+        self.server: Optional[grpc.Server] = None
+
+    def start(self) -> None:
+        os.makedirs(self.plugins_dir, exist_ok=True)
+        if os.path.exists(self.socket_path):
+            os.unlink(self.socket_path)
+        self.server = grpc.server(
+            futures.ThreadPoolExecutor(max_workers=8))
+        self.server.add_generic_rpc_handlers(
+            (api.device_plugin_handler(self.plugin),))
+        self.server.add_insecure_port(f"unix://{self.socket_path}")
+        self.server.start()
+        log.info("plugin %s serving on %s",
+                 self.plugin.resource_name, self.socket_path)
+
+    def register(self, kubelet_socket: str = api.KUBELET_SOCKET) -> None:
+        opts = self.plugin.GetDevicePluginOptions(api.Empty(), None)
+        api.register_with_kubelet(kubelet_socket, self.endpoint_name,
+                                  self.plugin.resource_name, opts)
+        log.info("registered %s with kubelet",
+                 self.plugin.resource_name)
+
+    def stop(self) -> None:
+        if hasattr(self.plugin, "stop"):
+            self.plugin.stop()
+        if self.server:
+            self.server.stop(grace=1.0)
+        if os.path.exists(self.socket_path):
+            try:
+                os.unlink(self.socket_path)
+            except OSError:
+                pass
+
+
+class PluginSet:
+    """Builds + runs the plugin set (reference factory.go): always
+    vgpu-number, optional vgpu-cores / vgpu-memory reporting plugins."""
+
+    def __init__(self, manager, client, *, open_vcore=False,
+                 open_vmemory=False,
+                 plugins_dir: str = api.PLUGINS_DIR, **vnum_kwargs):
+        from .vnum_plugin import VnumPlugin
+        self.servers: List[PluginServer] = []
+        vnum = VnumPlugin(manager, client, **vnum_kwargs)
+        self.servers.append(PluginServer(vnum, "amd-vgpu-number.sock",
+                                         plugins_dir))
+        if open_vcore:
+            total = consts.CORES_PER_GPU * len(manager.devices)
+            self.servers.append(PluginServer(
+                QuantityPlugin(consts.vgpu_core_resource(), total),
+                "amd-vgpu-cores.sock", plugins_dir))
+        if open_vmemory:
+            total_mib = sum(d.memory for d in manager.devices)
+            # one unit per 1 GiB to keep kubelet's device list bounded
+            self.servers.append(PluginServer(
+                QuantityPlugin(consts.vgpu_memory_resource(),
+                               max(total_mib // 1024, 1)),
+                "amd-vgpu-memory.sock", plugins_dir))
+
+    def start_all(self, kubelet_socket: str = api.KUBELET_SOCKET,
+                  register: bool = True) -> None:
+        for s in self.servers:
+            s.start()
+            if register:
+                s.register(kubelet_socket)
+
+    def stop_all(self) -> None:
+        for s in self.servers:
+            s.stop()
+
+
+def watch_kubelet_restart(plugin_set: PluginSet,
+                          kubelet_socket: str = api.KUBELET_SOCKET,
+                          poll_s: float = 5.0) -> threading.Thread:
+    """Re-register when kubelet's socket inode changes (the reference
+    uses fsnotify; polling the inode is equivalent and dependency-free).
+    """
+    def run():
+        last_ino = None
+        while True:
+            try:
+                ino = os.stat(kubelet_socket).st_ino
+            except OSError:
+                ino = None
+            if last_ino is not None and ino is not None and \
+                    ino != last_ino:
+                log.warning("kubelet restart detected; re-registering")
+                try:
+                    plugin_set.start_all(kubelet_socket)
+                except Exception as e:
+                    log.error("re-register failed: %s", e)
+            last_ino = ino
+            time.sleep(poll_s)
+
+    t = threading.Thread(target=run, daemon=True, name="kubelet-watch")
+    t.start()
+    return t

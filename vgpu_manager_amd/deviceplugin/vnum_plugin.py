@@ -1,0 +1,372 @@
+"""The vgpu-number device plugin — THE allocation plugin.
+
+Reference pkg/deviceplugin/vgpu/vnum_plugin.go re-designed for MI355X:
+  * kubelet sees split-count fake device IDs per physical GPU
+    ("<uuid>::<k>");
+  * GetPreferredAllocation honours the scheduler's pre-allocation;
+  * Allocate resolves the pending pod's container claim, writes the
+    container's vgpu.config region + devices.json, builds the env /
+    mount / device-node response (plain /dev/kfd + /dev/dri/renderD*
+    injection — no vendor container toolkit), and patches the pod's
+    real-allocated annotation;
+  * PreStartContainer re-verifies and cleans stale runtime regions.
+"""
+from __future__ import annotations
+
+import json
+import logging
+import os
+import threading
+import time
+from typing import Dict, List, Optional, Tuple
+
+from ..client.kube import KubeClient, KubeError
+from ..config.regions import DeviceLimit, VgpuConfigWriter
+from ..device.manager import DeviceManager
+from ..device.types import (
+    ContainerDeviceClaim,
+    marshal_pod_claim,
+    unmarshal_pod_claim,
+)
+from ..util import consts
+from . import api
+
+log = logging.getLogger("vgpu.deviceplugin.vnum")
+
+FAKE_ID_SEP = "::"
+
+
+def fake_id(uuid: str, k: int) -> str:
+    return f"{uuid}{FAKE_ID_SEP}{k}"
+
+
+def parse_fake_id(fid: str) -> Tuple[str, int]:
+    uuid, _, k = fid.rpartition(FAKE_ID_SEP)
+    return uuid, int(k)
+
+
+class VnumPlugin:
+    def __init__(self, manager: DeviceManager, client: KubeClient,
+                 base_dir: str = consts.MANAGER_DIR,
+                 driver_lib: str = "/usr/local/vgpu-manager/"
+                                   + consts.DRIVER_LIB_NAME,
+                 shared_watcher: bool = False):
+        self.manager = manager
+        self.client = client
+        self.base_dir = base_dir
+        self.driver_lib = driver_lib
+        self.shared_watcher = shared_watcher
+        self.resource_name = consts.vgpu_number_resource()
+        self._lw_cond = threading.Condition()
+        self._stopped = False
+        manager.on_health_change(lambda _d: self.trigger_listandwatch())
+
+    # ---------------- device inventory ----------------
+    def fake_devices(self) -> List[api.Device]:
+        out = []
+        for d in self.manager.devices:
+            topo = None
+            if d.numa >= 0:
+                topo = api.TopologyInfo(
+                    nodes=[api.NUMANode(ID=d.numa)])
+            for k in range(d.number):
+                out.append(api.Device(
+                    ID=fake_id(d.uuid, k),
+                    health=api.HEALTHY if d.healthy else api.UNHEALTHY,
+                    topology=topo))
+        return out
+
+    def trigger_listandwatch(self) -> None:
+        with self._lw_cond:
+            self._lw_cond.notify_all()
+
+    # ---------------- gRPC servicer ----------------
+    def GetDevicePluginOptions(self, request, context):
+        return api.DevicePluginOptions(
+            pre_start_required=True,
+            get_preferred_allocation_available=True)
+
+    def ListAndWatch(self, request, context):
+        while True:
+            yield api.ListAndWatchResponse(devices=self.fake_devices())
+            with self._lw_cond:
+                if self._stopped:
+                    return
+                self._lw_cond.wait(timeout=60.0)
+                if self._stopped:
+                    return
+
+    # ---- pending-pod resolution ----
+    def _pending_pods(self) -> List[dict]:
+        pods = self.client.list_pods(
+            label_selector={consts.assigned_phase_label():
+                            consts.PHASE_ALLOCATING})
+        mine = []
+        for p in pods:
+            ann = p.get("metadata", {}).get("annotations", {}) or {}
+            if ann.get(consts.predicate_node_ann()) != \
+                    self.manager.node_name:
+                continue
+            if not ann.get(consts.pre_alloc_ann()):
+                continue
+            mine.append(p)
+        mine.sort(key=lambda p: int(
+            (p["metadata"].get("annotations", {}) or {}).get(
+                consts.predicate_time_ann(), "0") or "0"))
+        return mine
+
+    def _claim_cursor(self, pod: dict, size: int
+                      ) -> Optional[ContainerDeviceClaim]:
+        """Next not-yet-really-allocated container claim of `size`
+        devices (reference GetCurrentPreAllocateContainerDevice)."""
+        ann = pod.get("metadata", {}).get("annotations", {}) or {}
+        try:
+            pre = unmarshal_pod_claim(ann.get(consts.pre_alloc_ann(), ""))
+        except ValueError:
+            return None
+        done = set()
+        real_txt = ann.get(consts.real_alloc_ann(), "")
+        if real_txt:
+            try:
+                done = {c.name for c in unmarshal_pod_claim(real_txt)}
+            except ValueError:
+                pass
+        for cdc in pre:
+            if cdc.name in done:
+                continue
+            if len(cdc.claims) == size:
+                return cdc
+        return None
+
+    def GetPreferredAllocation(self, request, context):
+        resp = api.PreferredAllocationResponse()
+        for creq in request.container_requests:
+            size = creq.allocation_size
+            chosen: List[str] = list(creq.must_include_device_ids)[:size]
+            claim = None
+            for pod in self._pending_pods():
+                claim = self._claim_cursor(pod, size)
+                if claim:
+                    break
+            if claim:
+                by_uuid: Dict[str, List[str]] = {}
+                for fid in creq.available_device_ids:
+                    uuid, _ = parse_fake_id(fid)
+                    by_uuid.setdefault(uuid, []).append(fid)
+                chosen = []
+                for c in claim.claims:
+                    pool = by_uuid.get(c.uuid) or []
+                    if pool:
+                        chosen.append(pool.pop(0))
+            while len(chosen) < size and creq.available_device_ids:
+                for fid in creq.available_device_ids:
+                    if fid not in chosen:
+                        chosen.append(fid)
+                        break
+                else:
+                    break
+            resp.container_responses.append(
+                api.ContainerPreferredAllocationResponse(
+                    device_ids=chosen[:size]))
+        return resp
+
+    # ---- Allocate ----
+    def _container_dir(self, pod_uid: str, container: str) -> str:
+        return os.path.join(self.base_dir, f"{pod_uid}_{container}")
+
+    def Allocate(self, request, context):
+        resp = api.AllocateResponse()
+        pods = self._pending_pods()
+        for creq in request.container_requests:
+            size = len(creq.devices_ids)
+            pod, claim = None, None
+            for p in pods:
+                claim = self._claim_cursor(p, size)
+                if claim:
+                    pod = p
+                    break
+            if pod is None or claim is None:
+                self._fail_pending(pods, f"no pending claim of size "
+                                         f"{size}")
+                raise RuntimeError(
+                    f"Allocate: no pending pod claim matches size {size}")
+            try:
+                resp.container_responses.append(
+                    self._allocate_container(pod, claim))
+                self._mark_real_allocation(pod, claim)
+            except Exception as e:
+                self._fail_pod(pod, str(e))
+                raise
+        return resp
+
+    def _allocate_container(self, pod: dict, claim: ContainerDeviceClaim
+                            ) -> api.ContainerAllocateResponse:
+        meta = pod["metadata"]
+        pod_uid = meta.get("uid", "")
+        cdir = self._container_dir(pod_uid, claim.name)
+        cfg_dir = os.path.join(cdir, "config")
+        for sub in ("config", "vgpu_lock", "vmem_node", "sm_node"):
+            os.makedirs(os.path.join(cdir, sub), exist_ok=True)
+
+        dev_by_uuid = {d.uuid: d for d in self.manager.devices}
+        limits, envs = [], {}
+        policy = (meta.get("annotations", {}) or {}).get(
+            consts.compute_policy_ann(), consts.COMPUTE_FIXED)
+        host_indices = []
+        for k, c in enumerate(claim.claims):
+            dev = dev_by_uuid.get(c.uuid)
+            if dev is None:
+                raise RuntimeError(f"claimed uuid {c.uuid} not on node")
+            host_indices.append(dev.id)
+            mem_bytes = c.memory << 20
+            limits.append(DeviceLimit(
+                uuid=c.uuid, host_index=dev.id, memory_bytes=mem_bytes,
+                core_limit=0 if policy == consts.COMPUTE_NONE
+                else int(c.cores),
+                soft_core_limit=100 if policy == consts.COMPUTE_BALANCE
+                else 0))
+            envs[consts.ENV_MEM_LIMIT.format(k)] = str(mem_bytes)
+            if policy != consts.COMPUTE_NONE and c.cores:
+                envs[consts.ENV_CORE_LIMIT.format(k)] = str(c.cores)
+            if policy == consts.COMPUTE_BALANCE:
+                envs[consts.ENV_CORE_SOFT_LIMIT.format(k)] = "100"
+
+        envs[consts.ENV_POD_NAME] = meta.get("name", "")
+        envs[consts.ENV_POD_NAMESPACE] = meta.get("namespace", "")
+        envs[consts.ENV_POD_UID] = pod_uid
+        envs[consts.ENV_CONTAINER_NAME] = claim.name
+        envs[consts.ENV_COMPUTE_POLICY] = policy
+        envs[consts.ENV_MANAGER_VISIBLE_DEVICES] = ",".join(
+            c.uuid for c in claim.claims)
+
+        # vgpu.config region + devices.json
+        w = VgpuConfigWriter(os.path.join(cfg_dir, "vgpu.config"))
+        w.write(pod_uid=pod_uid, pod_name=meta.get("name", ""),
+                pod_namespace=meta.get("namespace", ""),
+                container_name=claim.name, limits=limits,
+                compute_policy=policy)
+        w.close()
+        with open(os.path.join(cdir, "devices.json"), "w") as f:
+            json.dump({"claims": [c.marshal() for c in claim.claims],
+                       "host_indices": host_indices,
+                       "allocated_at": int(time.time())}, f)
+
+        # ld.so.preload file for the bind mount
+        preload_path = os.path.join(cdir, "ld.so.preload")
+        with open(preload_path, "w") as f:
+            f.write(f"{consts.MANAGER_DIR}/driver/"
+                    f"{consts.DRIVER_LIB_NAME}\n")
+
+        resp = api.ContainerAllocateResponse(envs=envs)
+        ro, rw = True, False
+        mounts = [
+            (f"{consts.MANAGER_DIR}/driver/{consts.DRIVER_LIB_NAME}",
+             self.driver_lib, ro),
+            ("/etc/ld.so.preload", preload_path, ro),
+            (f"{consts.MANAGER_DIR}/config", cfg_dir, ro),
+            ("/tmp/.vgpu_lock", os.path.join(cdir, "vgpu_lock"), rw),
+            ("/tmp/.vmem_node", os.path.join(cdir, "vmem_node"), rw),
+            ("/tmp/.sm_node", os.path.join(cdir, "sm_node"), rw),
+        ]
+        if self.shared_watcher:
+            mounts.append((f"{consts.MANAGER_DIR}/watcher",
+                           os.path.join(self.base_dir, "watcher"), ro))
+        for cpath, hpath, read_only in mounts:
+            resp.mounts.append(api.Mount(container_path=cpath,
+                                         host_path=hpath,
+                                         read_only=read_only))
+        resp.devices.append(api.DeviceSpec(
+            container_path="/dev/kfd", host_path="/dev/kfd",
+            permissions="rw"))
+        for idx in host_indices:
+            node = f"/dev/dri/renderD{128 + idx}"
+            resp.devices.append(api.DeviceSpec(
+                container_path=node, host_path=node, permissions="rw"))
+        return resp
+
+    def _mark_real_allocation(self, pod: dict,
+                              claim: ContainerDeviceClaim) -> None:
+        meta = pod["metadata"]
+        ann = meta.get("annotations", {}) or {}
+        try:
+            real = unmarshal_pod_claim(
+                ann.get(consts.real_alloc_ann(), "")) \
+                if ann.get(consts.real_alloc_ann()) else []
+        except ValueError:
+            real = []
+        real.append(claim)
+        pre = unmarshal_pod_claim(ann[consts.pre_alloc_ann()])
+        phase = consts.PHASE_SUCCESS if len(real) >= len(pre) \
+            else consts.PHASE_ALLOCATING
+        try:
+            self.client.patch_pod_metadata(
+                meta.get("namespace", "default"), meta.get("name", ""),
+                annotations={consts.real_alloc_ann():
+                             marshal_pod_claim(real)},
+                labels={consts.assigned_phase_label(): phase})
+            # keep the local copy coherent for multi-container pods
+            ann[consts.real_alloc_ann()] = marshal_pod_claim(real)
+            meta["annotations"] = ann
+        except KubeError as e:
+            log.error("real-allocation patch failed: %s", e)
+            raise
+
+    def _fail_pod(self, pod: dict, msg: str) -> None:
+        meta = pod["metadata"]
+        try:
+            self.client.patch_pod_metadata(
+                meta.get("namespace", "default"), meta.get("name", ""),
+                labels={consts.assigned_phase_label():
+                        consts.PHASE_FAILED})
+            self.client.create_event(
+                meta.get("namespace", "default"),
+                {"kind": "Pod", "name": meta.get("name", ""),
+                 "namespace": meta.get("namespace", "default"),
+                 "uid": meta.get("uid", "")},
+                "VGPUAllocateFailed", msg)
+        except KubeError:
+            pass
+
+    def _fail_pending(self, pods: List[dict], msg: str) -> None:
+        for p in pods[:1]:
+            self._fail_pod(p, msg)
+
+    # ---- PreStartContainer ----
+    def PreStartContainer(self, request, context):
+        # find the container dir whose devices.json covers the ids
+        uuids = {parse_fake_id(fid)[0] for fid in request.devices_ids}
+        try:
+            entries = os.listdir(self.base_dir)
+        except FileNotFoundError:
+            entries = []
+        for ent in entries:
+            cdir = os.path.join(self.base_dir, ent)
+            dj = os.path.join(cdir, "devices.json")
+            if not os.path.exists(dj):
+                continue
+            try:
+                data = json.load(open(dj))
+                claimed = {c.split("_")[1] for c in data.get("claims", [])}
+            except (ValueError, IndexError):
+                continue
+            if not uuids.issubset(claimed):
+                continue
+            cfg = os.path.join(cdir, "config", "vgpu.config")
+            if not os.path.exists(cfg) or \
+                    os.path.getsize(cfg) != 512 + 16 * 128:
+                raise RuntimeError(f"vgpu.config missing/stale in {cdir}")
+            # clean stale runtime regions from any previous container run
+            for stale in ("config/pids.config",
+                          "vmem_node/vmem_node.config",
+                          "sm_node/sm_node.config",
+                          "sm_node/sm_node.lock"):
+                path = os.path.join(cdir, stale)
+                if os.path.exists(path):
+                    os.unlink(path)
+            return api.PreStartContainerResponse()
+        raise RuntimeError("PreStartContainer: no allocation found for "
+                           f"{sorted(uuids)}")
+
+    def stop(self) -> None:
+        self._stopped = True
+        self.trigger_listandwatch()

@@ -1,0 +1,226 @@
+"""Minimal protobuf wire-format codec.
+
+grpcio is available in this stack but protoc/grpcio-tools are not, so
+the kubelet device-plugin API and the registry API are implemented
+with this declarative codec: each message is a class with a FIELDS
+spec; encode/decode handle varint, length-delimited, nested messages,
+repeated fields and maps — everything the kubelet v1beta1 surface
+uses.  Field numbers mirror the upstream api.proto exactly.
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, List, Tuple
+
+WIRE_VARINT = 0
+WIRE_I64 = 1
+WIRE_LEN = 2
+WIRE_I32 = 5
+
+
+def _enc_varint(v: int) -> bytes:
+    if v < 0:
+        v += 1 << 64
+    out = bytearray()
+    while True:
+        b = v & 0x7F
+        v >>= 7
+        if v:
+            out.append(b | 0x80)
+        else:
+            out.append(b)
+            return bytes(out)
+
+
+def _dec_varint(buf: bytes, pos: int) -> Tuple[int, int]:
+    result = 0
+    shift = 0
+    while True:
+        b = buf[pos]
+        pos += 1
+        result |= (b & 0x7F) << shift
+        if not b & 0x80:
+            return result, pos
+        shift += 7
+        if shift > 70:
+            raise ValueError("varint too long")
+
+
+def _tag(field_no: int, wire: int) -> bytes:
+    return _enc_varint((field_no << 3) | wire)
+
+
+# field kinds
+K_INT = "int"          # varint int32/int64/uint
+K_BOOL = "bool"
+K_STR = "string"
+K_BYTES = "bytes"
+K_MSG = "message"
+K_MAP_SS = "map_ss"    # map<string,string>
+
+
+class Message:
+    """Subclass with FIELDS = {field_no: (name, kind, repeated, cls)}"""
+
+    FIELDS: Dict[int, tuple] = {}
+
+    def __init__(self, **kwargs):
+        for no, (name, kind, repeated, _cls) in self.FIELDS.items():
+            if repeated:
+                default: Any = []
+            elif kind == K_MAP_SS:
+                default = {}
+            elif kind == K_INT:
+                default = 0
+            elif kind == K_BOOL:
+                default = False
+            elif kind in (K_STR,):
+                default = ""
+            elif kind == K_BYTES:
+                default = b""
+            else:
+                default = None
+            setattr(self, name, kwargs.get(name, default))
+
+    # ---- encoding ----
+    def encode(self) -> bytes:
+        out = bytearray()
+        for no, (name, kind, repeated, cls) in sorted(self.FIELDS.items()):
+            val = getattr(self, name)
+            if kind == K_MAP_SS:
+                for k, v in (val or {}).items():
+                    entry = (_tag(1, WIRE_LEN) +
+                             _enc_varint(len(k.encode())) + k.encode() +
+                             _tag(2, WIRE_LEN) +
+                             _enc_varint(len(v.encode())) + v.encode())
+                    out += _tag(no, WIRE_LEN) + _enc_varint(len(entry)) \
+                        + entry
+                continue
+            vals = val if repeated else ([val] if self._present(val, kind)
+                                         else [])
+            for v in vals:
+                if kind == K_INT:
+                    out += _tag(no, WIRE_VARINT) + _enc_varint(int(v))
+                elif kind == K_BOOL:
+                    out += _tag(no, WIRE_VARINT) + _enc_varint(
+                        1 if v else 0)
+                elif kind == K_STR:
+                    b = v.encode()
+                    out += _tag(no, WIRE_LEN) + _enc_varint(len(b)) + b
+                elif kind == K_BYTES:
+                    out += _tag(no, WIRE_LEN) + _enc_varint(len(v)) + v
+                elif kind == K_MSG:
+                    b = v.encode()
+                    out += _tag(no, WIRE_LEN) + _enc_varint(len(b)) + b
+        return bytes(out)
+
+    @staticmethod
+    def _present(val, kind) -> bool:
+        if val is None:
+            return False
+        if kind == K_INT:
+            return val != 0
+        if kind == K_BOOL:
+            return bool(val)
+        if kind in (K_STR,):
+            return val != ""
+        if kind == K_BYTES:
+            return val != b""
+        return True
+
+    # ---- decoding ----
+    @classmethod
+    def decode(cls, buf: bytes) -> "Message":
+        msg = cls()
+        pos = 0
+        n = len(buf)
+        while pos < n:
+            tag, pos = _dec_varint(buf, pos)
+            field_no, wire = tag >> 3, tag & 7
+            spec = cls.FIELDS.get(field_no)
+            if spec is None:
+                pos = cls._skip(buf, pos, wire)
+                continue
+            name, kind, repeated, sub = spec
+            if wire == WIRE_VARINT:
+                v, pos = _dec_varint(buf, pos)
+                val: Any = bool(v) if kind == K_BOOL else v
+            elif wire == WIRE_LEN:
+                ln, pos = _dec_varint(buf, pos)
+                raw = buf[pos:pos + ln]
+                pos += ln
+                if kind == K_STR:
+                    val = raw.decode()
+                elif kind == K_BYTES:
+                    val = raw
+                elif kind == K_MSG:
+                    val = sub.decode(raw)
+                elif kind == K_MAP_SS:
+                    k, v = cls._decode_map_entry(raw)
+                    getattr(msg, name)[k] = v
+                    continue
+                elif kind == K_INT and repeated:
+                    # packed repeated varints
+                    p2 = 0
+                    while p2 < len(raw):
+                        v, p2 = _dec_varint(raw, p2)
+                        getattr(msg, name).append(v)
+                    continue
+                else:
+                    val = raw
+            elif wire == WIRE_I64:
+                val = int.from_bytes(buf[pos:pos + 8], "little")
+                pos += 8
+            elif wire == WIRE_I32:
+                val = int.from_bytes(buf[pos:pos + 4], "little")
+                pos += 4
+            else:
+                raise ValueError(f"bad wire type {wire}")
+            if repeated:
+                getattr(msg, name).append(val)
+            else:
+                setattr(msg, name, val)
+        return msg
+
+    @staticmethod
+    def _decode_map_entry(raw: bytes) -> Tuple[str, str]:
+        k = v = ""
+        pos = 0
+        while pos < len(raw):
+            tag, pos = _dec_varint(raw, pos)
+            no, wire = tag >> 3, tag & 7
+            if wire != WIRE_LEN:
+                raise ValueError("map entry wire")
+            ln, pos = _dec_varint(raw, pos)
+            s = raw[pos:pos + ln].decode()
+            pos += ln
+            if no == 1:
+                k = s
+            elif no == 2:
+                v = s
+        return k, v
+
+    @staticmethod
+    def _skip(buf: bytes, pos: int, wire: int) -> int:
+        if wire == WIRE_VARINT:
+            _, pos = _dec_varint(buf, pos)
+        elif wire == WIRE_LEN:
+            ln, pos = _dec_varint(buf, pos)
+            pos += ln
+        elif wire == WIRE_I64:
+            pos += 8
+        elif wire == WIRE_I32:
+            pos += 4
+        else:
+            raise ValueError(f"cannot skip wire {wire}")
+        return pos
+
+    def __repr__(self):
+        fields = ", ".join(
+            f"{name}={getattr(self, name)!r}"
+            for _, (name, *_rest) in sorted(self.FIELDS.items()))
+        return f"{type(self).__name__}({fields})"
+
+    def __eq__(self, other):
+        return type(self) is type(other) and all(
+            getattr(self, name) == getattr(other, name)
+            for _, (name, *_r) in self.FIELDS.items())
