@@ -1,0 +1,103 @@
+"""Registry server tests: real unix sockets in tmpdir, peercred-based
+spoofing rejection, concurrent registration storms (reference
+pkg/device/registry security/stress tests)."""
+import json
+import os
+import subprocess
+import threading
+
+import pytest
+
+from vgpu_manager_amd.config.regions import _MappedRegion
+from vgpu_manager_amd.config.abi import PidsDataT, VGPU_PIDS_MAGIC
+from vgpu_manager_amd.registry.server import (
+    RegistryServer,
+    RegistryState,
+    register_via_socket,
+)
+
+from tests.conftest import LIB_DIR
+
+
+@pytest.fixture
+def registry(tmp_path):
+    base = str(tmp_path / "etc")
+    cdir = os.path.join(base, "uid-1_main", "config")
+    os.makedirs(cdir)
+    state = RegistryState(base_dir=base)
+    sock = str(tmp_path / "registry.sock")
+    server = RegistryServer(sock, state)
+    server.start_background()
+    yield sock, base, server
+    server.stop()
+
+
+def read_pids(base):
+    path = os.path.join(base, "uid-1_main", "config", "pids.config")
+    region = _MappedRegion(path, PidsDataT, VGPU_PIDS_MAGIC, create=False)
+    pids = list(region.data.pids[:region.data.pid_count])
+    region.close()
+    return pids
+
+
+def test_register_self(registry):
+    sock, base, _ = registry
+    out = register_via_socket(sock, "uid-1", "main")
+    assert out["ok"], out
+    assert os.getpid() in read_pids(base)
+
+
+def test_register_unknown_allocation_rejected(registry):
+    sock, _, _ = registry
+    out = register_via_socket(sock, "uid-nope", "main")
+    assert not out["ok"]
+    assert "unknown allocation" in out["error"]
+
+
+def test_spoofed_pid_rejected(registry):
+    sock, base, server = registry
+    # claim pid 1 (different cgroup from the test process in most
+    # environments; if not, force the verifier)
+    server.verify = lambda peer, claimed: peer == claimed
+    out = register_via_socket(sock, "uid-1", "main", pids=[1])
+    assert not out["ok"]
+    assert "not in caller" in out["error"]
+
+
+def test_path_traversal_rejected(registry):
+    sock, _, _ = registry
+    out = register_via_socket(sock, "../../etc", "main")
+    assert not out["ok"]
+
+
+def test_concurrent_registration_storm(registry):
+    sock, base, _ = registry
+    errs = []
+
+    def worker():
+        try:
+            out = register_via_socket(sock, "uid-1", "main")
+            assert out["ok"], out
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    threads = [threading.Thread(target=worker) for _ in range(32)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errs
+    assert os.getpid() in read_pids(base)
+
+
+def test_c_device_client(registry, built_core):
+    """The C device-client binary registers through the same socket."""
+    sock, base, _ = registry
+    binary = os.path.join(LIB_DIR, "build", "device-client")
+    subprocess.run(["make", "-s", "tools"], cwd=LIB_DIR, check=True)
+    r = subprocess.run([binary, "--socket", sock, "--pod-uid", "uid-1",
+                        "--container", "main"],
+                       capture_output=True, text=True, timeout=30)
+    assert r.returncode == 0, r.stdout + r.stderr
+    pids = read_pids(base)
+    assert len(pids) >= 1

@@ -1,0 +1,188 @@
+"""Client-registration registry (reference pkg/device/registry).
+
+When containers cannot see host /proc (DevicePluginClientMode), each
+in-container process registers its PID here; the shim's memory/util
+accounting then filters amd-smi process lists to these pids.
+
+MI355X-native re-design: instead of gRPC (whose Python server hides
+SO_PEERCRED), the registry speaks a length-delimited JSON protocol on a
+unix SOCK_STREAM socket, so the server authenticates the CALLER from
+the kernel's peer credentials and verifies the claimed pids belong to
+the caller's cgroup — same security property as the reference's
+peercred + cgroup walk (server.go:207-759), simpler transport.
+
+Protocol: client sends one JSON line
+    {"pod_uid": ..., "container_name": ..., "pids": [int, ...]}
+server replies {"ok": true} or {"ok": false, "error": ...}.
+"""
+from __future__ import annotations
+
+import json
+import logging
+import os
+import socket
+import socketserver
+import struct
+import threading
+from typing import Callable, Dict, List, Optional, Set
+
+from ..config.regions import PidsWriter
+from ..util import consts
+
+log = logging.getLogger("vgpu.registry")
+
+SO_PEERCRED = 17  # linux
+
+
+def peer_credentials(conn: socket.socket):
+    data = conn.getsockopt(socket.SOL_SOCKET, SO_PEERCRED,
+                           struct.calcsize("3i"))
+    pid, uid, gid = struct.unpack("3i", data)
+    return pid, uid, gid
+
+
+def cgroup_of(pid: int) -> str:
+    try:
+        return open(f"/proc/{pid}/cgroup").read()
+    except OSError:
+        return ""
+
+
+def same_container(peer_pid: int, claimed_pid: int) -> bool:
+    """The claimed pid must share the peer's cgroup (v2 single line) —
+    a process can only register pids of its own container."""
+    if peer_pid == claimed_pid:
+        return True
+    a, b = cgroup_of(peer_pid), cgroup_of(claimed_pid)
+    return bool(a) and a == b
+
+
+class RegistryState:
+    """pid sets per (pod_uid, container), persisted to pids.config."""
+
+    def __init__(self, base_dir: str = consts.MANAGER_DIR):
+        self.base_dir = base_dir
+        self._mu = threading.Lock()
+        self._pids: Dict[tuple, Set[int]] = {}
+
+    def container_dir(self, pod_uid: str, container: str) -> str:
+        return os.path.join(self.base_dir, f"{pod_uid}_{container}")
+
+    def register(self, pod_uid: str, container: str,
+                 pids: List[int]) -> int:
+        cdir = self.container_dir(pod_uid, container)
+        if not os.path.isdir(cdir):
+            raise ValueError(f"unknown allocation {pod_uid}_{container}")
+        with self._mu:
+            key = (pod_uid, container)
+            s = self._pids.setdefault(key, set())
+            s.update(int(p) for p in pids)
+            # prune dead pids
+            s.intersection_update(
+                {p for p in s if os.path.exists(f"/proc/{p}")})
+            alive = sorted(s)
+            w = PidsWriter(os.path.join(cdir, "config", "pids.config"))
+            w.write(alive)
+            w.close()
+            return len(alive)
+
+
+class _Handler(socketserver.StreamRequestHandler):
+    def handle(self):
+        server: RegistryServer = self.server  # type: ignore
+        try:
+            peer_pid, peer_uid, _ = peer_credentials(self.connection)
+            line = self.rfile.readline(65536)
+            req = json.loads(line)
+            pod_uid = str(req.get("pod_uid", ""))
+            container = str(req.get("container_name", ""))
+            pids = [int(p) for p in req.get("pids", [])] or [peer_pid]
+            if not pod_uid or not container:
+                raise ValueError("pod_uid and container_name required")
+            if "/" in pod_uid or "/" in container or \
+                    ".." in pod_uid or ".." in container:
+                raise ValueError("invalid identifier")
+            for p in pids:
+                if not server.verify(peer_pid, p):
+                    raise PermissionError(
+                        f"pid {p} not in caller's container "
+                        f"(peer {peer_pid})")
+            # always include the actual caller
+            if peer_pid not in pids:
+                pids.append(peer_pid)
+            n = server.state.register(pod_uid, container, pids)
+            self.wfile.write(json.dumps(
+                {"ok": True, "count": n}).encode() + b"\n")
+        except Exception as e:
+            try:
+                self.wfile.write(json.dumps(
+                    {"ok": False, "error": str(e)}).encode() + b"\n")
+            except OSError:
+                pass
+
+
+class RegistryServer(socketserver.ThreadingUnixStreamServer):
+    daemon_threads = True
+
+    def __init__(self, socket_path: str, state: RegistryState,
+                 verify: Optional[Callable[[int, int], bool]] = None):
+        sdir = os.path.dirname(socket_path)
+        os.makedirs(sdir, exist_ok=True)
+        if os.path.exists(socket_path):
+            os.unlink(socket_path)
+        super().__init__(socket_path, _Handler)
+        os.chmod(socket_path, 0o666)
+        self.state = state
+        self.verify = verify or same_container
+        self._thread: Optional[threading.Thread] = None
+
+    def start_background(self) -> None:
+        self._thread = threading.Thread(target=self.serve_forever,
+                                        daemon=True, name="vgpu-registry")
+        self._thread.start()
+
+    def stop(self) -> None:
+        self.shutdown()
+        self.server_close()
+
+
+def register_via_socket(socket_path: str, pod_uid: str, container: str,
+                        pids: Optional[List[int]] = None,
+                        timeout: float = 5.0) -> dict:
+    """Client side (used by the python device-client and tests; the C
+    library ships its own tiny client, library/tools/device_client.c)."""
+    s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    s.settimeout(timeout)
+    try:
+        s.connect(socket_path)
+        req = {"pod_uid": pod_uid, "container_name": container,
+               "pids": pids or [os.getpid()]}
+        s.sendall(json.dumps(req).encode() + b"\n")
+        resp = b""
+        while not resp.endswith(b"\n"):
+            chunk = s.recv(4096)
+            if not chunk:
+                break
+            resp += chunk
+        return json.loads(resp or b"{}")
+    finally:
+        s.close()
+
+
+def main():  # pragma: no cover — the mounted device-client entry
+    import argparse
+    ap = argparse.ArgumentParser("device-client")
+    ap.add_argument("--socket", default=consts.MANAGER_DIR +
+                    "/registry/socket.sock")
+    ap.add_argument("--pod-uid", required=True)
+    ap.add_argument("--container", required=True)
+    ap.add_argument("--pid", type=int, action="append", default=[])
+    args = ap.parse_args()
+    out = register_via_socket(args.socket, args.pod_uid, args.container,
+                              args.pid or None)
+    print(json.dumps(out))
+    raise SystemExit(0 if out.get("ok") else 1)
+
+
+if __name__ == "__main__":  # pragma: no cover
+    main()

@@ -1,0 +1,230 @@
+"""Admission webhook: pod mutate + validate (reference pkg/webhook).
+
+Mutate: pods requesting vgpu resources get the vgpu scheduler name and
+defaulted policy annotations; optional DRA conversion rewrites
+vgpu-number/cores/memory container limits into a generated
+ResourceClaim (stashing originals in the original-resources
+annotation, reference pod_mutate.go:249-300).
+
+Validate: resource sanity (number bounds, cores vs CORES_PER_GPU,
+memory sanity), annotation value checks (reference pod_validate.go).
+
+Both speak AdmissionReview v1 with a base64 JSONPatch response.
+"""
+from __future__ import annotations
+
+import base64
+import copy
+import json
+from typing import List, Optional, Tuple
+
+from ..util import consts
+
+VGPU_SCHEDULER_NAME = "vgpu-scheduler"
+
+VALID_TOPOLOGY_MODES = {consts.TOPO_NONE, consts.TOPO_NUMA,
+                        consts.TOPO_NUMA_STRICT, consts.TOPO_LINK,
+                        consts.TOPO_LINK_STRICT}
+VALID_POLICIES = {consts.POLICY_BINPACK, consts.POLICY_SPREAD}
+VALID_COMPUTE = {consts.COMPUTE_FIXED, consts.COMPUTE_BALANCE,
+                 consts.COMPUTE_NONE}
+
+
+def _is_vgpu_pod(pod: dict) -> bool:
+    for c in (pod.get("spec", {}).get("containers") or []) + \
+             (pod.get("spec", {}).get("initContainers") or []):
+        limits = (c.get("resources", {}) or {}).get("limits", {}) or {}
+        if consts.vgpu_number_resource() in limits:
+            return True
+    return False
+
+
+def mutate_pod(pod: dict, *, default_scheduler: str = VGPU_SCHEDULER_NAME,
+               dra_mode: bool = False) -> List[dict]:
+    """Returns a JSONPatch list."""
+    patches: List[dict] = []
+    ann = pod.get("metadata", {}).get("annotations", {}) or {}
+    if ann.get(consts.IGNORE_WEBHOOK_ANN) in ("true", "1"):
+        return patches
+    if not _is_vgpu_pod(pod):
+        return patches
+
+    spec = pod.get("spec", {})
+    if spec.get("schedulerName") in (None, "", "default-scheduler"):
+        patches.append({"op": "add", "path": "/spec/schedulerName",
+                        "value": default_scheduler})
+
+    if not pod.get("metadata", {}).get("annotations"):
+        patches.append({"op": "add", "path": "/metadata/annotations",
+                        "value": {}})
+
+    def default_ann(key, value):
+        if key not in ann:
+            patches.append({"op": "add",
+                            "path": "/metadata/annotations/" +
+                                    key.replace("/", "~1"),
+                            "value": value})
+
+    default_ann(consts.device_scheduler_policy_ann(),
+                consts.POLICY_BINPACK)
+    default_ann(consts.node_scheduler_policy_ann(),
+                consts.POLICY_BINPACK)
+    default_ann(consts.compute_policy_ann(), consts.COMPUTE_FIXED)
+
+    if dra_mode:
+        patches.extend(_dra_conversion_patches(pod))
+    return patches
+
+
+def _dra_conversion_patches(pod: dict) -> List[dict]:
+    """Rewrite vgpu-* limits into a generated ResourceClaim reference.
+    The claim template itself is created by the controller from the
+    stashed originals."""
+    patches: List[dict] = []
+    originals = {}
+    res_names = {consts.vgpu_number_resource(),
+                 consts.vgpu_core_resource(),
+                 consts.vgpu_memory_resource()}
+    for ci, c in enumerate(pod.get("spec", {}).get("containers") or []):
+        limits = (c.get("resources", {}) or {}).get("limits", {}) or {}
+        mine = {k: str(v) for k, v in limits.items() if k in res_names}
+        if not mine:
+            continue
+        originals[c["name"]] = mine
+        for k in mine:
+            patches.append({
+                "op": "remove",
+                "path": f"/spec/containers/{ci}/resources/limits/" +
+                        k.replace("/", "~1")})
+        patches.append({
+            "op": "add",
+            "path": f"/spec/containers/{ci}/resources/claims",
+            "value": [{"name": "vgpu-claim"}]})
+    if originals:
+        patches.append({
+            "op": "add",
+            "path": "/metadata/annotations/" +
+                    consts.DRA_ORIGINAL_RESOURCES_ANN.replace("/", "~1"),
+            "value": json.dumps(originals, separators=(",", ":"))})
+        patches.append({
+            "op": "add",
+            "path": "/spec/resourceClaims",
+            "value": [{"name": "vgpu-claim",
+                       "resourceClaimTemplateName":
+                           f"vgpu-{pod.get('metadata', {}).get('name', 'pod')}"}]})
+    return patches
+
+
+def validate_pod(pod: dict) -> Tuple[bool, str]:
+    ann = pod.get("metadata", {}).get("annotations", {}) or {}
+    if ann.get(consts.IGNORE_WEBHOOK_ANN) in ("true", "1"):
+        return True, ""
+    containers = (pod.get("spec", {}).get("containers") or []) + \
+                 (pod.get("spec", {}).get("initContainers") or [])
+    for c in containers:
+        limits = (c.get("resources", {}) or {}).get("limits", {}) or {}
+        num = limits.get(consts.vgpu_number_resource())
+        cores = limits.get(consts.vgpu_core_resource())
+        mem = limits.get(consts.vgpu_memory_resource())
+        if num is None:
+            if cores is not None or mem is not None:
+                return False, (f"container {c.get('name')}: vgpu-cores/"
+                               "memory require vgpu-number")
+            continue
+        try:
+            n = int(num)
+        except (TypeError, ValueError):
+            return False, f"vgpu-number not an integer: {num!r}"
+        if not 1 <= n <= consts.MAX_DEVICE_COUNT:
+            return False, (f"vgpu-number {n} out of range 1.."
+                           f"{consts.MAX_DEVICE_COUNT}")
+        if cores is not None:
+            cr = int(cores)
+            if not 0 <= cr <= consts.CORES_PER_GPU * n:
+                return False, f"vgpu-cores {cr} out of range"
+        if mem is not None and int(mem) < 0:
+            return False, "vgpu-memory negative"
+
+    checks = [
+        (consts.topology_mode_ann(), VALID_TOPOLOGY_MODES),
+        (consts.device_scheduler_policy_ann(), VALID_POLICIES),
+        (consts.node_scheduler_policy_ann(), VALID_POLICIES),
+        (consts.compute_policy_ann(), VALID_COMPUTE),
+    ]
+    for key, valid in checks:
+        v = ann.get(key)
+        if v is not None and v not in valid:
+            return False, f"annotation {key}={v!r} invalid " \
+                          f"(expect one of {sorted(valid)})"
+    return True, ""
+
+
+# ---- AdmissionReview plumbing ----
+
+def handle_admission_review(body: dict, *, mutating: bool,
+                            dra_mode: bool = False) -> dict:
+    req = body.get("request", {}) or {}
+    uid = req.get("uid", "")
+    obj = req.get("object", {}) or {}
+    response = {"uid": uid, "allowed": True}
+    if mutating:
+        patches = mutate_pod(obj, dra_mode=dra_mode)
+        if patches:
+            response["patchType"] = "JSONPatch"
+            response["patch"] = base64.b64encode(
+                json.dumps(patches).encode()).decode()
+    else:
+        ok, msg = validate_pod(obj)
+        response["allowed"] = ok
+        if not ok:
+            response["status"] = {"message": msg, "code": 400}
+    return {"apiVersion": "admission.k8s.io/v1",
+            "kind": "AdmissionReview", "response": response}
+
+
+def apply_json_patch(obj: dict, patches: List[dict]) -> dict:
+    """Minimal JSONPatch applier (add/remove/replace) for tests."""
+    obj = copy.deepcopy(obj)
+    for p in patches:
+        parts = [x.replace("~1", "/").replace("~0", "~")
+                 for x in p["path"].lstrip("/").split("/")]
+        tgt = obj
+        for part in parts[:-1]:
+            if isinstance(tgt, list):
+                tgt = tgt[int(part)]
+            else:
+                tgt = tgt.setdefault(part, {})
+        last = parts[-1]
+        if p["op"] in ("add", "replace"):
+            if isinstance(tgt, list):
+                tgt.insert(int(last), p["value"])
+            else:
+                tgt[last] = p["value"]
+        elif p["op"] == "remove":
+            if isinstance(tgt, list):
+                tgt.pop(int(last))
+            else:
+                tgt.pop(last, None)
+    return obj
+
+
+def create_app(dra_mode: bool = False):
+    from fastapi import FastAPI, Request
+
+    app = FastAPI(title="vgpu-webhook")
+
+    @app.post("/webhook/mutate-pod")
+    async def mutate(request: Request):
+        return handle_admission_review(await request.json(),
+                                       mutating=True, dra_mode=dra_mode)
+
+    @app.post("/webhook/validate-pod")
+    async def validate(request: Request):
+        return handle_admission_review(await request.json(),
+                                       mutating=False)
+
+    @app.get("/healthz")
+    async def healthz():
+        return {"status": "ok"}
+
+    return app
