@@ -123,6 +123,7 @@ class _Handler(socketserver.StreamRequestHandler):
 
 class RegistryServer(socketserver.ThreadingUnixStreamServer):
     daemon_threads = True
+    request_queue_size = 128  # registration storms must not EAGAIN
 
     def __init__(self, socket_path: str, state: RegistryState,
                  verify: Optional[Callable[[int, int], bool]] = None):
@@ -154,7 +155,15 @@ def register_via_socket(socket_path: str, pod_uid: str, container: str,
     s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
     s.settimeout(timeout)
     try:
-        s.connect(socket_path)
+        for attempt in range(20):
+            try:
+                s.connect(socket_path)
+                break
+            except (BlockingIOError, ConnectionRefusedError):
+                if attempt == 19:
+                    raise
+                import time as _t
+                _t.sleep(0.05)
         req = {"pod_uid": pod_uid, "container_name": container,
                "pids": pids or [os.getpid()]}
         s.sendall(json.dumps(req).encode() + b"\n")
