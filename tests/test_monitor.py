@@ -1,0 +1,108 @@
+"""Monitor tests: lister inode-cache behavior, collectors over fake
+regions, the sampler publishing into the sm_util region with a fake
+source."""
+import os
+import time
+
+from prometheus_client import CollectorRegistry, generate_latest
+
+from vgpu_manager_amd.config.regions import (
+    DeviceLimit,
+    VgpuConfigWriter,
+)
+from vgpu_manager_amd.config.abi import (
+    UtilRegionT,
+    VGPU_UTIL_MAGIC,
+)
+from vgpu_manager_amd.config.regions import _MappedRegion
+from vgpu_manager_amd.device.manager import FakeDeviceManager
+from vgpu_manager_amd.monitor.collector import NodeVgpuCollector
+from vgpu_manager_amd.monitor.lister import ContainerLister
+from vgpu_manager_amd.monitor.sampler import SampleSource, UtilSampler
+
+
+def write_container(base, pod_uid, cont, mem=1 << 30, cores=50):
+    cdir = os.path.join(base, f"{pod_uid}_{cont}")
+    os.makedirs(os.path.join(cdir, "vmem_node"), exist_ok=True)
+    w = VgpuConfigWriter(os.path.join(cdir, "config", "vgpu.config"))
+    w.write(pod_uid=pod_uid, pod_name="p", pod_namespace="ns",
+            container_name=cont,
+            limits=[DeviceLimit(uuid="GPU-x", host_index=0,
+                                memory_bytes=mem, core_limit=cores)])
+    w.close()
+    return cdir
+
+
+def test_lister_discovers_and_drops(tmp_path):
+    base = str(tmp_path)
+    write_container(base, "uid-a", "main")
+    lister = ContainerLister(base_dir=base)
+    entries = lister.scan()
+    assert len(entries) == 1
+    assert entries[0].pod_uid == "uid-a"
+    snap = entries[0].cfg.snapshot()
+    assert snap["devices"][0]["core_limit"] == 50
+
+    # container goes away -> entry dropped
+    import shutil
+    shutil.rmtree(os.path.join(base, "uid-a_main"))
+    assert lister.scan() == []
+
+
+def test_lister_reloads_on_inode_change(tmp_path):
+    base = str(tmp_path)
+    cdir = write_container(base, "uid-a", "main", cores=10)
+    lister = ContainerLister(base_dir=base)
+    e = lister.scan()[0]
+    ino1 = e.cfg_ino
+    # rewrite the file (new inode)
+    os.unlink(os.path.join(cdir, "config", "vgpu.config"))
+    write_container(base, "uid-a", "main", cores=77)
+    e = lister.scan()[0]
+    assert e.cfg_ino != ino1
+    assert e.cfg.snapshot()["devices"][0]["core_limit"] == 77
+
+
+def test_node_collector_metrics(tmp_path):
+    base = str(tmp_path)
+    write_container(base, "uid-a", "main", mem=2 << 30)
+    mgr = FakeDeviceManager("node-x", n_devices=2)
+    mgr.set_health(1, False)
+    registry = CollectorRegistry()
+    registry.register(NodeVgpuCollector(
+        mgr, ContainerLister(base_dir=base)))
+    text = generate_latest(registry).decode()
+    assert 'node_vgpu_device_healthy{device="1",node="node-x"' in text
+    assert "container_vgpu_device_memory_limit_bytes" in text
+    assert str(2 << 30) + ".0" in text
+
+
+class FakeSource(SampleSource):
+    def __init__(self, n=2):
+        self.n = n
+
+    def device_count(self):
+        return self.n
+
+    def sample(self, dev):
+        return dict(dev_busy_permille=500 + dev,
+                    vram_used_bytes=1234,
+                    procs=[dict(pid=100 + dev, gfx_busy_permille=250,
+                                vram_bytes=55, cu_occupancy=64)])
+
+
+def test_sampler_publishes_region(tmp_path):
+    path = str(tmp_path / "sm_util.config")
+    sampler = UtilSampler(FakeSource(2), path)
+    sampler.run_once()
+    sampler.run_once()
+    region = _MappedRegion(path, UtilRegionT, VGPU_UTIL_MAGIC,
+                           create=False)
+    d0 = region.data.devices[0]
+    assert d0.seq % 2 == 0 and d0.seq >= 4
+    assert d0.dev_busy_permille == 500
+    assert d0.procs[0].pid == 100
+    assert region.data.devices[1].dev_busy_permille == 501
+    assert region.data.heartbeat_ns > 0
+    region.close()
+    sampler.stop()

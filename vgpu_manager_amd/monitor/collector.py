@@ -1,0 +1,139 @@
+"""Prometheus collectors (reference pkg/metrics/collector):
+physical_gpu_device_* (amd-smi), node_vgpu_device_* (registration
+view), container_vgpu_device_* (per-container regions via the lister).
+"""
+from __future__ import annotations
+
+import logging
+from typing import Optional
+
+from prometheus_client.core import GaugeMetricFamily
+from prometheus_client.registry import Collector
+
+from ..config.abi import DEV_FLAG_MEM_LIMIT
+from ..device.manager import DeviceManager
+from .lister import ContainerLister
+
+log = logging.getLogger("vgpu.monitor.collector")
+
+
+class NodeVgpuCollector(Collector):
+    """node_vgpu_device_*: what the node registered vs assigned."""
+
+    def __init__(self, manager: DeviceManager,
+                 lister: Optional[ContainerLister] = None):
+        self.manager = manager
+        self.lister = lister
+
+    def collect(self):
+        node = self.manager.node_name
+        total = GaugeMetricFamily(
+            "node_vgpu_device_memory_total_mib",
+            "Registered vGPU memory per device",
+            labels=["node", "device", "uuid"])
+        cores = GaugeMetricFamily(
+            "node_vgpu_device_cores_total",
+            "Registered vGPU cores per device",
+            labels=["node", "device", "uuid"])
+        number = GaugeMetricFamily(
+            "node_vgpu_device_number_total",
+            "vGPU split count per device",
+            labels=["node", "device", "uuid"])
+        healthy = GaugeMetricFamily(
+            "node_vgpu_device_healthy",
+            "Device health (1 healthy)",
+            labels=["node", "device", "uuid"])
+        for d in self.manager.devices:
+            lbl = [node, str(d.id), d.uuid]
+            total.add_metric(lbl, d.memory)
+            cores.add_metric(lbl, d.core)
+            number.add_metric(lbl, d.number)
+            healthy.add_metric(lbl, 1 if d.healthy else 0)
+        yield from (total, cores, number, healthy)
+
+        if self.lister is not None:
+            yield from self._container_metrics(node)
+
+    def _container_metrics(self, node):
+        mem_quota = GaugeMetricFamily(
+            "container_vgpu_device_memory_limit_bytes",
+            "Container per-device HBM quota",
+            labels=["node", "pod_uid", "container", "vdev", "uuid"])
+        mem_used = GaugeMetricFamily(
+            "container_vgpu_device_memory_used_bytes",
+            "Container per-device hooked usage",
+            labels=["node", "pod_uid", "container", "vdev", "uuid"])
+        vmem_used = GaugeMetricFamily(
+            "container_vgpu_device_vmemory_used_bytes",
+            "Container per-device managed-spill usage",
+            labels=["node", "pod_uid", "container", "vdev", "uuid"])
+        core_limit = GaugeMetricFamily(
+            "container_vgpu_device_core_limit",
+            "Container per-device CU limit %",
+            labels=["node", "pod_uid", "container", "vdev", "uuid"])
+        for e in self.lister.scan():
+            try:
+                snap = e.cfg.snapshot()
+            except Exception:
+                continue
+            usage = e.vmem.device_usage() if e.vmem else None
+            for i, d in enumerate(snap["devices"]):
+                uuid = d["uuid"].rstrip("\x00")
+                lbl = [node, e.pod_uid, e.container, str(i), uuid]
+                if d["flags"] & DEV_FLAG_MEM_LIMIT:
+                    mem_quota.add_metric(lbl, d["total_memory"])
+                core_limit.add_metric(lbl, d["core_limit"])
+                if usage and i < len(usage):
+                    mem_used.add_metric(lbl,
+                                        usage[i]["dev_hooked_used"])
+                    vmem_used.add_metric(lbl, usage[i]["vmem_used"])
+        yield from (mem_quota, mem_used, vmem_used, core_limit)
+
+
+class PhysicalGpuCollector(Collector):
+    """physical_gpu_device_*: raw amd-smi truth (host view)."""
+
+    def __init__(self, amdsmi_module=None):
+        self.amdsmi = amdsmi_module
+        self._handles = None
+
+    def _ensure(self):
+        if self.amdsmi is None:
+            import amdsmi
+            self.amdsmi = amdsmi
+        if self._handles is None:
+            self.amdsmi.amdsmi_init()
+            self._handles = self.amdsmi.amdsmi_get_processor_handles()
+        return self._handles
+
+    def collect(self):
+        util = GaugeMetricFamily("physical_gpu_device_gfx_busy_percent",
+                                 "GPU gfx activity", labels=["device"])
+        vram = GaugeMetricFamily("physical_gpu_device_vram_used_bytes",
+                                 "GPU VRAM used", labels=["device"])
+        vram_total = GaugeMetricFamily(
+            "physical_gpu_device_vram_total_bytes",
+            "GPU VRAM total", labels=["device"])
+        try:
+            handles = self._ensure()
+        except Exception:
+            return
+        a = self.amdsmi
+        for i, h in enumerate(handles):
+            try:
+                act = a.amdsmi_get_gpu_activity(h)
+                util.add_metric([str(i)],
+                                act.get("gfx_activity", 0)
+                                if isinstance(act, dict) else 0)
+            except Exception:
+                pass
+            try:
+                u = a.amdsmi_get_gpu_memory_usage(
+                    h, a.AmdSmiMemoryType.VRAM)
+                t = a.amdsmi_get_gpu_memory_total(
+                    h, a.AmdSmiMemoryType.VRAM)
+                vram.add_metric([str(i)], int(u))
+                vram_total.add_metric([str(i)], int(t))
+            except Exception:
+                pass
+        yield from (util, vram, vram_total)

@@ -1,0 +1,94 @@
+"""Container lister: discovers allocated containers by walking the
+manager dir, mmaps each container's vgpu.config + vmem regions, and
+caches the mappings keyed by (pod_uid, container), reloading when the
+backing inode changes (reference pkg/metrics/lister/container_lister.go
+:148-263)."""
+from __future__ import annotations
+
+import logging
+import os
+from typing import Dict, List, Optional, Tuple
+
+from ..config.regions import VgpuConfigReader, VmemRegionReader
+from ..util import consts
+
+log = logging.getLogger("vgpu.monitor.lister")
+
+
+class ContainerEntry:
+    def __init__(self, pod_uid: str, container: str, cdir: str):
+        self.pod_uid = pod_uid
+        self.container = container
+        self.cdir = cdir
+        self.cfg: Optional[VgpuConfigReader] = None
+        self.vmem: Optional[VmemRegionReader] = None
+        self.cfg_ino = 0
+        self.vmem_ino = 0
+
+    def refresh(self) -> None:
+        cfg_path = os.path.join(self.cdir, "config", "vgpu.config")
+        vmem_path = os.path.join(self.cdir, "vmem_node",
+                                 "vmem_node.config")
+        self.cfg = self._remap(self.cfg, cfg_path, "cfg_ino",
+                               VgpuConfigReader)
+        self.vmem = self._remap(self.vmem, vmem_path, "vmem_ino",
+                                VmemRegionReader)
+
+    def _remap(self, current, path, ino_attr, cls):
+        try:
+            ino = os.stat(path).st_ino
+        except OSError:
+            if current:
+                current.close()
+            setattr(self, ino_attr, 0)
+            return None
+        if current is not None and getattr(self, ino_attr) == ino:
+            return current
+        if current:
+            current.close()
+        try:
+            region = cls(path)
+        except (ValueError, OSError) as e:
+            log.debug("remap %s failed: %s", path, e)
+            setattr(self, ino_attr, 0)
+            return None
+        setattr(self, ino_attr, ino)
+        return region
+
+    def close(self) -> None:
+        if self.cfg:
+            self.cfg.close()
+        if self.vmem:
+            self.vmem.close()
+
+
+class ContainerLister:
+    def __init__(self, base_dir: str = consts.MANAGER_DIR):
+        self.base_dir = base_dir
+        self.entries: Dict[Tuple[str, str], ContainerEntry] = {}
+
+    def scan(self) -> List[ContainerEntry]:
+        seen = set()
+        try:
+            names = os.listdir(self.base_dir)
+        except OSError:
+            names = []
+        for name in names:
+            if "_" not in name:
+                continue
+            cdir = os.path.join(self.base_dir, name)
+            if not os.path.isdir(cdir):
+                continue
+            pod_uid, _, container = name.partition("_")
+            key = (pod_uid, container)
+            seen.add(key)
+            entry = self.entries.get(key)
+            if entry is None:
+                entry = ContainerEntry(pod_uid, container, cdir)
+                self.entries[key] = entry
+            entry.refresh()
+        # drop gone containers
+        for key in list(self.entries):
+            if key not in seen:
+                self.entries.pop(key).close()
+        return [e for e in self.entries.values() if e.cfg is not None]

@@ -1,0 +1,141 @@
+"""Host-side shared utilization sampler (SharedSMUtilizationWatcher).
+
+One privileged host process samples amd-smi per device on an absolute
+80ms/batch cadence (10ms overrun floor, <=4 devices per batch thread —
+reference pkg/device/manager/watcher.go:65-195) and publishes device
+busy + per-process gfx/VRAM/CU-occupancy into the sm_util region that
+every container's shim maps read-only.  This centralizes the sampling
+cost: N containers x M processes would otherwise each pay their own
+amd-smi queries (reference design rationale, hook.h:561-573).
+"""
+from __future__ import annotations
+
+import logging
+import threading
+import time
+from typing import List, Optional
+
+from ..config.regions import UtilRegionWriter
+
+log = logging.getLogger("vgpu.monitor.sampler")
+
+BATCH_CYCLE_MS = 80
+OVERRUN_FLOOR_MS = 10
+MAX_BATCH_DEVICES = 4
+
+
+class SampleSource:
+    """Abstract per-device sampling (amd-smi backed in production)."""
+
+    def device_count(self) -> int:
+        raise NotImplementedError
+
+    def sample(self, dev: int) -> dict:
+        """Returns dict(dev_busy_permille=..., vram_used_bytes=...,
+        procs=[{pid, gfx_busy_permille, vram_bytes, cu_occupancy}])"""
+        raise NotImplementedError
+
+
+class AmdSmiSource(SampleSource):
+    def __init__(self):
+        import amdsmi
+        self.a = amdsmi
+        amdsmi.amdsmi_init()
+        self.handles = amdsmi.amdsmi_get_processor_handles()
+        self._prev_gfx = {}  # (dev, pid) -> (engine_ns, mono_ns)
+        self._cu_total = 256
+
+    def device_count(self) -> int:
+        return len(self.handles)
+
+    def sample(self, dev: int) -> dict:
+        a = self.a
+        h = self.handles[dev]
+        busy = 0
+        vram_used = 0
+        procs = []
+        try:
+            act = a.amdsmi_get_gpu_activity(h)
+            g = act.get("gfx_activity", 0) if isinstance(act, dict) else 0
+            if isinstance(g, int) and 0 <= g <= 100:
+                busy = g * 10
+        except Exception:
+            pass
+        try:
+            vram_used = int(a.amdsmi_get_gpu_memory_usage(
+                h, a.AmdSmiMemoryType.VRAM))
+        except Exception:
+            pass
+        now = time.monotonic_ns()
+        try:
+            plist = a.amdsmi_get_gpu_process_list(h)
+            for p in plist:
+                info = p if isinstance(p, dict) else {}
+                pid = int(info.get("pid", 0))
+                mem = int(info.get("memory_usage", {}).get("vram_mem", 0)
+                          if isinstance(info.get("memory_usage"), dict)
+                          else info.get("mem", 0))
+                eng = info.get("engine_usage", {})
+                gfx_ns = int(eng.get("gfx", 0)) if isinstance(eng, dict) \
+                    else 0
+                cu = int(info.get("cu_occupancy", 0) or 0)
+                permille = 0
+                key = (dev, pid)
+                prev = self._prev_gfx.get(key)
+                if prev and gfx_ns >= prev[0] and now > prev[1]:
+                    permille = int((gfx_ns - prev[0]) * 1000 /
+                                   (now - prev[1]))
+                self._prev_gfx[key] = (gfx_ns, now)
+                if permille == 0 and cu > 0:
+                    permille = cu * 1000 // self._cu_total
+                procs.append(dict(pid=pid,
+                                  gfx_busy_permille=min(permille, 1000),
+                                  vram_bytes=mem, cu_occupancy=cu))
+        except Exception:
+            pass
+        return dict(dev_busy_permille=busy, vram_used_bytes=vram_used,
+                    procs=procs)
+
+
+class UtilSampler:
+    def __init__(self, source: SampleSource, region_path: str):
+        self.source = source
+        n = source.device_count()
+        self.writer = UtilRegionWriter(region_path, device_count=n)
+        self._stop = threading.Event()
+        self.cycles = 0
+
+    def run_once(self) -> None:
+        for dev in range(self.source.device_count()):
+            try:
+                s = self.source.sample(dev)
+            except Exception as e:
+                log.debug("sample dev %d failed: %s", dev, e)
+                continue
+            self.writer.publish(dev, **s)
+        self.cycles += 1
+
+    def run_forever(self) -> None:
+        """Absolute-time cadence divided by batches (reference
+        doWatcher): cycle = BATCH_CYCLE_MS per <=4-device batch."""
+        n = self.source.device_count()
+        batches = max(1, (n + MAX_BATCH_DEVICES - 1) // MAX_BATCH_DEVICES)
+        cycle_ns = BATCH_CYCLE_MS * 1_000_000 * batches
+        next_t = time.monotonic_ns()
+        while not self._stop.is_set():
+            self.run_once()
+            next_t += cycle_ns
+            now = time.monotonic_ns()
+            if next_t <= now + OVERRUN_FLOOR_MS * 1_000_000:
+                next_t = now + OVERRUN_FLOOR_MS * 1_000_000
+            self._stop.wait((next_t - now) / 1e9)
+
+    def start_background(self) -> threading.Thread:
+        t = threading.Thread(target=self.run_forever, daemon=True,
+                             name="vgpu-util-sampler")
+        t.start()
+        return t
+
+    def stop(self) -> None:
+        self._stop.set()
+        self.writer.close()
