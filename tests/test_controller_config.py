@@ -1,0 +1,110 @@
+"""Reschedule controller + node config + feature gates tests."""
+import os
+
+import pytest
+
+from vgpu_manager_amd.client.kube import FakeKubeClient
+from vgpu_manager_amd.controller.reschedule import (
+    CONFIG_REGION_SIZE,
+    RescheduleController,
+)
+from vgpu_manager_amd.util import consts
+from vgpu_manager_amd.util.nodeconfig import (
+    CORE_GATES,
+    DRA_GATES,
+    FakeIdStore,
+    FeatureGates,
+    load_node_config,
+)
+
+from tests.test_allocator import make_pod
+
+
+def test_reschedule_evicts_failed(tmp_path):
+    client = FakeKubeClient()
+    pod = make_pod(number=1, name="bad")
+    pod["metadata"]["labels"] = {
+        consts.assigned_phase_label(): consts.PHASE_FAILED}
+    pod["metadata"]["annotations"][consts.predicate_node_ann()] = "n1"
+    client.add_pod(pod)
+    ctl = RescheduleController(client, "n1",
+                               str(tmp_path / "ckpt.json"),
+                               base_dir=str(tmp_path / "etc"))
+    n = ctl.reconcile_once()
+    assert n == 1
+    assert client.evictions == [("default", "bad")]
+    # second pass: pod gone, nothing to do
+    assert ctl.reconcile_once() == 0
+
+
+def test_reschedule_stale_config(tmp_path):
+    client = FakeKubeClient()
+    pod = make_pod(number=1, name="stale")
+    pod["spec"]["nodeName"] = "n1"
+    client.add_pod(pod)
+    cdir = tmp_path / "etc" / "uid-stale_main" / "config"
+    os.makedirs(cdir)
+    (cdir / "vgpu.config").write_bytes(b"\x00" * 100)  # wrong size
+    ctl = RescheduleController(client, "n1",
+                               str(tmp_path / "ckpt.json"),
+                               base_dir=str(tmp_path / "etc"))
+    assert ctl.reconcile_once() == 1
+    assert client.evictions == [("default", "stale")]
+
+
+def test_checkpoint_prevents_double_evict(tmp_path):
+    client = FakeKubeClient()
+    ctl = RescheduleController(client, "n1",
+                               str(tmp_path / "ckpt.json"),
+                               base_dir=str(tmp_path / "etc"))
+    assert ctl.checkpoint.mark("uid-x", "test")
+    assert not ctl.checkpoint.mark("uid-x", "test")
+    ctl.checkpoint.done("uid-x")
+    assert ctl.checkpoint.mark("uid-x", "again")
+
+
+def test_feature_gates():
+    fg = FeatureGates(CORE_GATES)
+    fg.parse("SharedSMUtilizationWatcher=true,SerializedNodeBind=false")
+    assert fg.enabled("SharedSMUtilizationWatcher")
+    assert not fg.enabled("SerializedNodeBind")
+    with pytest.raises(ValueError):
+        fg.parse("NoSuchGate=true")
+
+    dra = FeatureGates(DRA_GATES)
+    dra.parse("ConsumableShares=true")
+    with pytest.raises(ValueError, match="requires DRADriver"):
+        dra.validate(dra.as_dict())
+    dra.parse("DRADriver=true")
+    dra.validate(dra.as_dict())
+
+
+def test_node_config_overrides(tmp_path):
+    p = tmp_path / "config.yaml"
+    p.write_text("""
+deviceSplitCount: 10
+deviceMemoryScaling: 1.5
+nodes:
+  node-b:
+    deviceSplitCount: 4
+    excludeDevices: [7]
+""")
+    cfg_a = load_node_config(str(p), "node-a")
+    assert cfg_a.deviceSplitCount == 10
+    assert cfg_a.deviceMemoryScaling == 1.5
+    cfg_b = load_node_config(str(p), "node-b")
+    assert cfg_b.deviceSplitCount == 4
+    assert cfg_b.excludeDevices == [7]
+    assert load_node_config(None, "x").deviceSplitCount == 10
+
+
+def test_fake_id_store(tmp_path):
+    store = FakeIdStore(str(tmp_path / "ids.json"))
+    order1 = store.stable_order(["GPU-b", "GPU-a"])
+    assert order1 == ["GPU-a", "GPU-b"]
+    # a new device appends, existing order preserved
+    store2 = FakeIdStore(str(tmp_path / "ids.json"))
+    order2 = store2.stable_order(["GPU-c", "GPU-a", "GPU-b"])
+    assert order2 == ["GPU-a", "GPU-b", "GPU-c"]
+    # removed device drops without disturbing order
+    assert store2.stable_order(["GPU-c", "GPU-a"]) == ["GPU-a", "GPU-c"]

@@ -1,0 +1,166 @@
+"""AllocationFailureReschedule controller (reference
+pkg/controller/reschedule): watches pods on this node whose
+assigned-phase is failed (or whose config region is layout-stale),
+evicts/deletes them so the scheduler retries elsewhere, and checkpoints
+in-flight recoveries so a controller restart never double-evicts.
+"""
+from __future__ import annotations
+
+import json
+import logging
+import os
+import threading
+import time
+from typing import Dict, List, Optional
+
+from ..client.kube import KubeClient, KubeError
+from ..util import consts
+
+log = logging.getLogger("vgpu.controller.reschedule")
+
+CONFIG_REGION_SIZE = 512 + 16 * 128
+
+
+class RecoveryCheckpoint:
+    """On-disk record of pods being recovered (reference
+    checkpoint.go) so restarts never double-evict."""
+
+    def __init__(self, path: str):
+        self.path = path
+        self._mu = threading.Lock()
+        self.entries: Dict[str, dict] = {}
+        self._load()
+
+    def _load(self) -> None:
+        try:
+            data = json.load(open(self.path))
+            self.entries = data.get("entries", {})
+        except (OSError, ValueError):
+            self.entries = {}
+
+    def _save(self) -> None:
+        tmp = self.path + ".tmp"
+        os.makedirs(os.path.dirname(self.path) or ".", exist_ok=True)
+        with open(tmp, "w") as f:
+            json.dump({"entries": self.entries}, f)
+        os.replace(tmp, self.path)
+
+    def mark(self, pod_uid: str, reason: str) -> bool:
+        """Returns False if already in-flight (skip double-evict)."""
+        with self._mu:
+            if pod_uid in self.entries:
+                return False
+            self.entries[pod_uid] = {"reason": reason,
+                                     "at": int(time.time())}
+            self._save()
+            return True
+
+    def done(self, pod_uid: str) -> None:
+        with self._mu:
+            if self.entries.pop(pod_uid, None) is not None:
+                self._save()
+
+    def prune(self, max_age_s: int = 3600) -> None:
+        with self._mu:
+            cutoff = time.time() - max_age_s
+            stale = [k for k, v in self.entries.items()
+                     if v.get("at", 0) < cutoff]
+            for k in stale:
+                self.entries.pop(k)
+            if stale:
+                self._save()
+
+
+class RescheduleController:
+    def __init__(self, client: KubeClient, node_name: str,
+                 checkpoint_path: str,
+                 base_dir: str = consts.MANAGER_DIR,
+                 use_eviction: bool = True):
+        self.client = client
+        self.node_name = node_name
+        self.base_dir = base_dir
+        self.use_eviction = use_eviction
+        self.checkpoint = RecoveryCheckpoint(checkpoint_path)
+        self._stop = threading.Event()
+
+    # ---- detection ----
+    def _failed_pods(self) -> List[dict]:
+        try:
+            pods = self.client.list_pods(
+                label_selector={consts.assigned_phase_label():
+                                consts.PHASE_FAILED})
+        except KubeError:
+            return []
+        return [p for p in pods
+                if (p.get("metadata", {}).get("annotations", {}) or {})
+                .get(consts.predicate_node_ann()) == self.node_name
+                or p.get("spec", {}).get("nodeName") == self.node_name]
+
+    def _stale_config_pods(self) -> List[dict]:
+        """Pods whose written config region has a stale layout size —
+        a plugin upgrade changed the ABI under a running pod."""
+        out = []
+        try:
+            pods = self.client.list_pods(node_name=self.node_name)
+        except KubeError:
+            return []
+        for p in pods:
+            uid = p.get("metadata", {}).get("uid", "")
+            if not uid:
+                continue
+            for c in p.get("spec", {}).get("containers", []):
+                cfg = os.path.join(self.base_dir,
+                                   f"{uid}_{c.get('name')}",
+                                   "config", "vgpu.config")
+                try:
+                    size = os.path.getsize(cfg)
+                except OSError:
+                    continue
+                if size != CONFIG_REGION_SIZE:
+                    out.append(p)
+                    break
+        return out
+
+    # ---- recovery ----
+    def _recover(self, pod: dict, reason: str) -> None:
+        meta = pod.get("metadata", {})
+        uid = meta.get("uid", "")
+        ns = meta.get("namespace", "default")
+        name = meta.get("name", "")
+        if not self.checkpoint.mark(uid, reason):
+            return
+        log.warning("recovering pod %s/%s: %s", ns, name, reason)
+        try:
+            if self.use_eviction:
+                self.client.evict_pod(ns, name)
+            else:
+                self.client.delete_pod(ns, name)
+            self.client.create_event(
+                ns, {"kind": "Pod", "name": name, "namespace": ns,
+                     "uid": uid},
+                "VGPUReschedule", f"evicted for rescheduling: {reason}")
+        except KubeError as e:
+            log.error("recovery of %s/%s failed: %s", ns, name, e)
+        finally:
+            self.checkpoint.done(uid)
+
+    def reconcile_once(self) -> int:
+        n = 0
+        for pod in self._failed_pods():
+            self._recover(pod, "allocation failed")
+            n += 1
+        for pod in self._stale_config_pods():
+            self._recover(pod, "stale vgpu.config layout")
+            n += 1
+        self.checkpoint.prune()
+        return n
+
+    def run_forever(self, interval_s: float = 10.0) -> None:
+        while not self._stop.wait(interval_s):
+            try:
+                self.reconcile_once()
+            except Exception as e:
+                log.error("reconcile failed: %s", e)
+
+    def stop(self) -> None:
+        self._stop.set()
