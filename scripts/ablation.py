@@ -1,0 +1,118 @@
+#!/usr/bin/env python3
+"""Controller ablation harness — measures CU-throttle accuracy (MAE).
+
+The reference's headline number: steady-state MAE of
+observed GPU utilization vs the configured hard core limit — stock
+delta controller 17.5-20.7%, AIMD 2.2-2.8% (RTX 4080).  This harness
+reproduces the measurement on MI355X: run a saturating launch storm
+under the shim with VGPU_CORE_LIMIT_0=<target>, sample amd-smi
+gfx_activity on the host at 10 Hz, compute MAE over the steady-state
+window.
+
+Usage (on a GPU box):
+    python scripts/ablation.py --targets 20,30,50 \
+        --controllers delta,aimd --duration 20 \
+        --out gpurun_out/ablation.json
+"""
+import argparse
+import ctypes
+import json
+import os
+import statistics
+import subprocess
+import sys
+import threading
+import time
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+BUILD = os.path.join(REPO, "library", "build")
+SHIM = os.path.join(BUILD, "libvgpu-control.so")
+WORKLOAD = os.path.join(BUILD, "libworkload.so")
+
+WORKER_CODE = r"""
+import ctypes, sys, time
+wk = ctypes.CDLL(%r)
+wk.wk_launch_busy.argtypes = [ctypes.c_int]*3 + [ctypes.c_longlong]
+assert wk.wk_init(0) == 0
+deadline = time.time() + %f
+while time.time() < deadline:
+    wk.wk_launch_busy(40, 2048, 256, 60000)
+    wk.wk_sync()
+print("worker done")
+"""
+
+
+def sample_busy(samples, stop, period=0.1):
+    import amdsmi
+    amdsmi.amdsmi_init()
+    h = amdsmi.amdsmi_get_processor_handles()[0]
+    while not stop.is_set():
+        try:
+            act = amdsmi.amdsmi_get_gpu_activity(h)
+            samples.append((time.time(),
+                            act.get("gfx_activity", 0)))
+        except Exception:
+            pass
+        stop.wait(period)
+
+
+def run_case(controller, target, duration, warm_frac=0.35):
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = SHIM
+    env["VGPU_CORE_LIMIT_0"] = str(target)
+    env["VGPU_CU_CONTROLLER"] = controller
+    samples = []
+    stop = threading.Event()
+    t = threading.Thread(target=sample_busy, args=(samples, stop),
+                         daemon=True)
+    t.start()
+    code = WORKER_CODE % (WORKLOAD, float(duration))
+    r = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True,
+                       timeout=duration * 8 + 120)
+    stop.set()
+    t.join(timeout=2)
+    if r.returncode != 0:
+        return dict(controller=controller, target=target,
+                    error=r.stderr[-400:])
+    if len(samples) < 10:
+        return dict(controller=controller, target=target,
+                    error="no samples")
+    t0 = samples[0][0]
+    t_end = samples[-1][0]
+    window = [b for (ts, b) in samples
+              if ts > t0 + (t_end - t0) * warm_frac]
+    mae = statistics.mean(abs(b - target) for b in window)
+    mean_busy = statistics.mean(window)
+    return dict(controller=controller, target=target,
+                mae=round(mae, 2), mean_busy=round(mean_busy, 1),
+                n_samples=len(window),
+                wall_s=round(t_end - t0, 1))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--targets", default="20,30,50")
+    ap.add_argument("--controllers", default="delta,aimd")
+    ap.add_argument("--duration", type=float, default=20.0)
+    ap.add_argument("--out", default="gpurun_out/ablation.json")
+    args = ap.parse_args()
+
+    results = []
+    for ctl in args.controllers.split(","):
+        for tgt in args.targets.split(","):
+            res = run_case(ctl.strip(), int(tgt), args.duration)
+            print(json.dumps(res), flush=True)
+            results.append(res)
+    os.makedirs(os.path.dirname(args.out) or ".", exist_ok=True)
+    with open(args.out, "w") as f:
+        json.dump({"results": results,
+                   "reference_baseline":
+                       {"delta_mae": "17.5-20.7 (RTX4080)",
+                        "aimd_mae": "2.2-2.8 (RTX4080)"}},
+                  f, indent=2)
+    print("wrote", args.out)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,52 @@
+"""device-monitor binary (reference cmd/device-monitor): Prometheus
+exporter + optional shared utilization sampler."""
+from __future__ import annotations
+
+import argparse
+import logging
+import os
+import sys
+import threading
+
+from prometheus_client import start_http_server, REGISTRY
+
+from ..monitor.collector import NodeVgpuCollector, PhysicalGpuCollector
+from ..monitor.lister import ContainerLister
+from ..util import consts
+
+
+def main(argv=None):
+    ap = argparse.ArgumentParser("vgpu-device-monitor")
+    ap.add_argument("--node-name",
+                    default=os.environ.get("NODE_NAME", "node"))
+    ap.add_argument("--port", type=int, default=9394)
+    ap.add_argument("--base-dir", default=consts.MANAGER_DIR)
+    ap.add_argument("--shared-watcher", action="store_true",
+                    help="run the shared utilization sampler too")
+    args = ap.parse_args(argv)
+    logging.basicConfig(level=logging.INFO)
+
+    lister = ContainerLister(base_dir=args.base_dir)
+    try:
+        from ..device.manager import AmdDeviceManager
+        manager = AmdDeviceManager(args.node_name)
+    except Exception:
+        from ..device.manager import FakeDeviceManager
+        manager = FakeDeviceManager(args.node_name, n_devices=0)
+    REGISTRY.register(NodeVgpuCollector(manager, lister))
+    REGISTRY.register(PhysicalGpuCollector())
+
+    if args.shared_watcher:
+        from ..monitor.sampler import AmdSmiSource, UtilSampler
+        sampler = UtilSampler(
+            AmdSmiSource(),
+            os.path.join(args.base_dir, "watcher", "sm_util.config"))
+        sampler.start_background()
+
+    start_http_server(args.port)
+    threading.Event().wait()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

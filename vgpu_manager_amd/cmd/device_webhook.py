@@ -1,0 +1,30 @@
+"""device-webhook binary (reference cmd/device-webhook)."""
+from __future__ import annotations
+
+import argparse
+import sys
+
+
+def main(argv=None):
+    ap = argparse.ArgumentParser("vgpu-device-webhook")
+    ap.add_argument("--bind", default="0.0.0.0:8443")
+    ap.add_argument("--tls-cert", default=None)
+    ap.add_argument("--tls-key", default=None)
+    ap.add_argument("--dra-mode", action="store_true")
+    args = ap.parse_args(argv)
+
+    import uvicorn
+    from ..webhook.admission import create_app
+
+    host, port = args.bind.rsplit(":", 1)
+    kwargs = {}
+    if args.tls_cert and args.tls_key:
+        kwargs = dict(ssl_certfile=args.tls_cert,
+                      ssl_keyfile=args.tls_key)
+    uvicorn.run(create_app(dra_mode=args.dra_mode), host=host,
+                port=int(port), **kwargs)
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
