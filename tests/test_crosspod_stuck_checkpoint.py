@@ -1,0 +1,128 @@
+"""Cross-pod gang alignment, stuck detection, kubelet checkpoint parse."""
+import json
+import time
+
+from vgpu_manager_amd.client.kube import FakeKubeClient
+from vgpu_manager_amd.deviceplugin.checkpoint import (
+    parse_kubelet_checkpoint,
+    pod_for_device,
+)
+from vgpu_manager_amd.monitor.stuck import (
+    find_stuck_pods,
+    recover_stuck_pods,
+)
+from vgpu_manager_amd.scheduler.crosspod import sibling_placement
+from vgpu_manager_amd.scheduler.filter import GpuFilter
+from vgpu_manager_amd.util import consts
+
+from tests.test_allocator import make_pod
+from tests.test_scheduler import make_node
+
+
+GANG_ANN = "scheduling.k8s.io/group-name"
+
+
+def gang_pod(name, gang="g1", node=None, claim=None, cross=True):
+    pod = make_pod(number=1, name=name)
+    ann = pod["metadata"]["annotations"]
+    ann[GANG_ANN] = gang
+    if cross:
+        ann[consts.cross_pod_topology_ann()] = "true"
+    if node:
+        pod["spec"]["nodeName"] = node
+    if claim:
+        ann[consts.real_alloc_ann()] = claim
+    return pod
+
+
+def test_sibling_placement_votes_numa():
+    client = FakeKubeClient()
+    # node with NUMA info: dev 0/1 numa 0, dev 2/3 numa 1
+    from vgpu_manager_amd.device.types import (
+        encode_node_devices, fake_device)
+    devs = [fake_device(i, numa=i // 2) for i in range(4)]
+    client.add_node({"metadata": {"name": "n1", "annotations": {
+        consts.node_register_ann(): encode_node_devices(devs)}}})
+    client.add_pod(gang_pod("sib1", node="n1",
+                            claim="main[2_GPU-fake-0002_10_1024]"))
+    me = gang_pod("me")
+    client.add_pod(me)
+    nodes, numa = sibling_placement(client, me)
+    assert nodes == {"n1": 1}
+    assert numa == 1  # device 2 is in numa 1
+
+
+def test_filter_prefers_gang_node():
+    client = FakeKubeClient()
+    client.add_node(make_node("gpu-node-1"))
+    client.add_node(make_node("gpu-node-2"))
+    # sibling on gpu-node-2
+    client.add_pod(gang_pod("sib", node="gpu-node-2",
+                            claim="main[0_GPU-fake-0000_10_1024]"))
+    me = gang_pod("me2")
+    client.add_pod(me)
+    res = GpuFilter(client).filter(
+        {"Pod": me, "NodeNames": ["gpu-node-1", "gpu-node-2"]})
+    assert res["NodeNames"] == ["gpu-node-2"]
+
+
+def test_stuck_detection_and_recovery():
+    client = FakeKubeClient()
+    pod = make_pod(number=1, name="stuck")
+    ann = pod["metadata"]["annotations"]
+    ann[consts.pre_alloc_ann()] = "main[0_GPU-x_0_1024]"
+    ann[consts.predicate_time_ann()] = str(int(time.time()) - 300)
+    pod["metadata"]["labels"] = {
+        consts.assigned_phase_label(): consts.PHASE_ALLOCATING}
+    client.add_pod(pod)
+
+    fresh = make_pod(number=1, name="fresh")
+    fresh["metadata"]["annotations"][consts.predicate_time_ann()] = \
+        str(int(time.time()))
+    fresh["metadata"]["labels"] = {
+        consts.assigned_phase_label(): consts.PHASE_ALLOCATING}
+    client.add_pod(fresh)
+
+    stuck = find_stuck_pods(client)
+    assert [p["metadata"]["name"] for p in stuck] == ["stuck"]
+    assert recover_stuck_pods(client) == 1
+    assert any(e["reason"] == "VGPUSchedulingStuck"
+               for e in client.events)
+
+
+def test_stuck_respects_custom_grace():
+    client = FakeKubeClient()
+    pod = make_pod(number=1, name="patient")
+    ann = pod["metadata"]["annotations"]
+    ann[consts.pre_alloc_ann()] = "x[0_GPU-x_0_1]"
+    ann[consts.predicate_time_ann()] = str(int(time.time()) - 300)
+    ann[consts.stuck_grace_period_ann()] = "3600"
+    pod["metadata"]["labels"] = {
+        consts.assigned_phase_label(): consts.PHASE_ALLOCATING}
+    client.add_pod(pod)
+    assert find_stuck_pods(client) == []
+
+
+def test_kubelet_checkpoint_both_formats(tmp_path):
+    plain = {
+        "Data": {"PodDeviceEntries": [
+            {"PodUID": "uid-1", "ContainerName": "c1",
+             "ResourceName": "amd.com/vgpu-number",
+             "DeviceIDs": ["GPU-a::0", "GPU-a::1"]},
+            {"PodUID": "uid-2", "ContainerName": "c2",
+             "ResourceName": "amd.com/vgpu-number",
+             "DeviceIDs": {"0": ["GPU-b::0"], "1": ["GPU-c::0"]}},
+        ]},
+        "Checksum": 12345,
+    }
+    p = tmp_path / "kubelet_internal_checkpoint"
+    p.write_text(json.dumps(plain))
+    ckpt = parse_kubelet_checkpoint(str(p))
+    assert ckpt["uid-1"]["c1"]["amd.com/vgpu-number"] == \
+        ["GPU-a::0", "GPU-a::1"]
+    assert sorted(ckpt["uid-2"]["c2"]["amd.com/vgpu-number"]) == \
+        ["GPU-b::0", "GPU-c::0"]
+    assert pod_for_device(ckpt, "amd.com/vgpu-number", "GPU-c::0") == \
+        ("uid-2", "c2")
+    assert pod_for_device(ckpt, "amd.com/vgpu-number", "nope") is None
+    assert parse_kubelet_checkpoint(str(tmp_path / "missing")) == {}

@@ -71,6 +71,7 @@ class AllocationRequest:
     exclude_uuids: List[str] = field(default_factory=list)
     include_types: List[str] = field(default_factory=list)
     exclude_types: List[str] = field(default_factory=list)
+    preferred_numa: Optional[int] = None  # cross-pod gang alignment
 
 
 def _parse_qty(v) -> int:
@@ -185,15 +186,18 @@ class Allocator:
         return out
 
     # ---- priority sort ----
-    def _sort(self, devs: List[DeviceUsage], policy: str
+    def _sort(self, devs: List[DeviceUsage], policy: str,
+              preferred_numa: Optional[int] = None
               ) -> List[DeviceUsage]:
         def used_frac(d: DeviceUsage):
             return (d.used_cores / max(d.info.core, 1) +
                     d.used_memory / max(d.info.memory, 1) +
                     d.used_number / max(d.info.number, 1))
 
-        # binpack: most-used first (fill devices up); spread: least-used
+        # gang alignment first, then binpack (most-used) / spread
         return sorted(devs, key=lambda d: (
+            0 if preferred_numa is not None and
+            d.info.numa == preferred_numa else 1,
             -used_frac(d) if policy == consts.POLICY_BINPACK
             else used_frac(d), d.info.id))
 
@@ -263,7 +267,7 @@ class Allocator:
     def allocate_container(self, req: AllocationRequest,
                            cr: ContainerRequest) -> ContainerDeviceClaim:
         devs = self._filter(req, cr)
-        devs = self._sort(devs, req.device_policy)
+        devs = self._sort(devs, req.device_policy, req.preferred_numa)
         chosen = self._pick_topology(devs, cr.number, req.topology_mode,
                                      req.device_policy)
         if len(chosen) < cr.number:

@@ -1,0 +1,101 @@
+"""Cross-pod topology alignment (reference FindGangSiblingDomain,
+filter_predicate.go:616-689 + cross_pod_nvlink_topology_design.md).
+
+Gang members (volcano / coscheduling / PodGroup annotations) with
+`cross-pod-topology: "true"` should land next to their siblings: on the
+node already hosting siblings, and on devices in the same NUMA domain
+as the siblings' devices (on a fully-connected xGMI node the NUMA
+domain is the remaining locality axis — SURVEY §5.8)."""
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+from ..client.kube import KubeClient, KubeError
+from ..device.types import decode_node_devices, unmarshal_pod_claim
+from ..util import consts
+
+GANG_KEYS = [
+    ("annotations", "scheduling.k8s.io/group-name"),
+    ("annotations", "scheduling.volcano.sh/group-name"),
+    ("annotations", "gang.scheduling.koordinator.sh/name"),
+    ("labels", "scheduling.x-k8s.io/pod-group"),
+    ("labels", "pod-group.scheduling.sigs.k8s.io/name"),
+]
+
+
+def gang_key(pod: dict) -> Optional[Tuple[str, str]]:
+    meta = pod.get("metadata", {})
+    for kind, key in GANG_KEYS:
+        val = (meta.get(kind, {}) or {}).get(key)
+        if val:
+            return key, val
+    return None
+
+
+def wants_cross_pod(pod: dict) -> bool:
+    ann = pod.get("metadata", {}).get("annotations", {}) or {}
+    return ann.get(consts.cross_pod_topology_ann(), "") in ("true", "1")
+
+
+def sibling_placement(client: KubeClient, pod: dict
+                      ) -> Tuple[Dict[str, int], Optional[int]]:
+    """Returns ({node_name: sibling_count}, preferred_numa).
+
+    preferred_numa is the NUMA domain most of the siblings' devices sit
+    in (on their node) — the allocator biases device choice toward it.
+    """
+    key = gang_key(pod)
+    if key is None or not wants_cross_pod(pod):
+        return {}, None
+    kind_key, val = key
+    my_name = pod.get("metadata", {}).get("name")
+    node_counts: Dict[str, int] = {}
+    numa_votes: Dict[int, int] = {}
+    try:
+        pods = client.list_pods()
+    except KubeError:
+        return {}, None
+    node_cache: Dict[str, dict] = {}
+    for p in pods:
+        if p.get("metadata", {}).get("name") == my_name:
+            continue
+        if gang_key(p) != key:
+            continue
+        node_name = p.get("spec", {}).get("nodeName") or \
+            (p.get("metadata", {}).get("annotations", {}) or {}).get(
+                consts.predicate_node_ann())
+        if not node_name:
+            continue
+        node_counts[node_name] = node_counts.get(node_name, 0) + 1
+        ann = p.get("metadata", {}).get("annotations", {}) or {}
+        claim_txt = ann.get(consts.real_alloc_ann()) or \
+            ann.get(consts.pre_alloc_ann())
+        if not claim_txt:
+            continue
+        try:
+            cdcs = unmarshal_pod_claim(claim_txt)
+        except ValueError:
+            continue
+        node = node_cache.get(node_name)
+        if node is None:
+            try:
+                node = client.get_node(node_name)
+            except KubeError:
+                continue
+            node_cache[node_name] = node
+        reg = (node.get("metadata", {}).get("annotations", {}) or {}
+               ).get(consts.node_register_ann())
+        if not reg:
+            continue
+        try:
+            numa_by_id = {d.id: d.numa for d in decode_node_devices(reg)}
+        except ValueError:
+            continue
+        for cdc in cdcs:
+            for c in cdc.claims:
+                numa = numa_by_id.get(c.id, -1)
+                if numa >= 0:
+                    numa_votes[numa] = numa_votes.get(numa, 0) + 1
+    preferred = max(numa_votes, key=numa_votes.get) if numa_votes \
+        else None
+    return node_counts, preferred

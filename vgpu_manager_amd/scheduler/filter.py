@@ -69,6 +69,14 @@ class GpuFilter:
         policy_ann = (meta.get("annotations", {}) or {}).get(
             consts.node_scheduler_policy_ann(), consts.POLICY_BINPACK)
 
+        # cross-pod gang alignment: bias toward nodes hosting siblings
+        # and toward the siblings' NUMA domain (crosspod.py)
+        from .crosspod import sibling_placement
+        sibling_nodes, preferred_numa = sibling_placement(self.client,
+                                                          pod)
+        if preferred_numa is not None:
+            request.preferred_numa = preferred_numa
+
         lock_key = "global-filter"
         if self.locker and not dry_run:
             self.locker.acquire(lock_key)
@@ -85,8 +93,10 @@ class GpuFilter:
                 if info is None:
                     failed[name] = R_NODE_NOT_VGPU
                     continue
-                candidates.append((self._node_score(info, policy_ann),
-                                   name, [node, pods]))
+                score = self._node_score(info, policy_ann)
+                # gang bonus dominates the fitness score
+                score -= sibling_nodes.get(name, 0) * 10 ** 9
+                candidates.append((score, name, [node, pods]))
 
             candidates.sort(key=lambda c: (c[0], c[1]))
             chosen: Optional[str] = None
