@@ -175,6 +175,9 @@ uint64_t dev_hooked_used(int dev);
 /* container used-bytes accounting (ledger/smi/max; hip_hook.c) */
 uint64_t vgpu_account_used(int dev);
 
+/* atfork child handler hook-side reset (hip_hook.c) */
+void vgpu_hook_fork_child(void);
+
 /* amd-smi sampling (watcher side; dlopens libamd_smi lazily) */
 bool smi_available(void);
 /* whole-device busy (permille) + container gfx engine ns + vram bytes */

@@ -43,7 +43,10 @@ static inline int cfg_dev(int dev) {
 /* utilization sampling + controllers + refill (watcher thread)        */
 /* ------------------------------------------------------------------ */
 
-static pthread_once_t g_watcher_once = PTHREAD_ONCE_INIT;
+/* CAS flag, not pthread_once: a fork() child inherits a consumed once
+ * but NOT the watcher thread — the flag is re-armed by the atfork
+ * child handler so the child's first throttled launch restarts it.    */
+static int g_watcher_state; /* 0 = not running, 1 = started            */
 
 static void dev_hot_init(int dev) {
     dev_hot_t *h = &g_state.dev[dev];
@@ -362,6 +365,8 @@ void vgpu_register_fini_atexit(void) { atexit(hook_fini); }
 
 static void hook_fini(void) {
     __atomic_store_n(&g_shutdown, 1, __ATOMIC_RELEASE);
+    if (!__atomic_load_n(&g_watcher_state, __ATOMIC_ACQUIRE))
+        return; /* no watcher in THIS process (e.g. fork child idle)   */
     for (int i = 0; i < 50; i++) { /* <=500ms grace                    */
         if (__atomic_load_n(&g_watcher_parked, __ATOMIC_ACQUIRE)) break;
         struct timespec ts = {0, 10000000L};
@@ -409,9 +414,28 @@ static void start_watcher(void) {
 /* ------------------------------------------------------------------ */
 /* rate limiter (launch path)                                          */
 /* ------------------------------------------------------------------ */
+/* fork-safety: called from the atfork child handler (loader.c). The
+ * watcher pthread does not survive fork; re-arm its start flag and
+ * zero the per-device pools so dev_hot_init reseeds initial shares —
+ * otherwise the child's first throttled launch parks forever on an
+ * empty bucket nobody refills (reference cuda_hook.c:260-315 analog). */
+void vgpu_hook_fork_child(void) {
+    __atomic_store_n(&g_watcher_state, 0, __ATOMIC_RELEASE);
+    __atomic_store_n(&g_watcher_parked, 0, __ATOMIC_RELEASE);
+    __atomic_store_n(&g_shutdown, 0, __ATOMIC_RELEASE);
+    for (int i = 0; i < MAX_DEVICE_COUNT; i++) {
+        g_state.dev[i].pool = 0;
+        g_state.dev[i].waiting = 0;
+        g_state.dev[i].throttled = 0;
+    }
+}
+
 static void rate_limiter(int dev, int64_t grids) {
     dev_hot_t *h = &g_state.dev[dev];
-    pthread_once(&g_watcher_once, start_watcher);
+    int expect = 0;
+    if (__atomic_compare_exchange_n(&g_watcher_state, &expect, 1, false,
+                                    __ATOMIC_ACQ_REL, __ATOMIC_RELAXED))
+        start_watcher();
     dev_hot_init(dev);
     __atomic_fetch_add(&h->launch_count, 1, __ATOMIC_RELAXED);
     if (grids > h->pool) grids = h->pool;

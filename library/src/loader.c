@@ -264,6 +264,7 @@ static void fork_child_handler(void) {
         g_state.dev[i].gap_stop = NULL;
         pthread_mutex_init(&g_state.dev[i].gap_mu, NULL);
     }
+    vgpu_hook_fork_child(); /* re-arm watcher start + reseed pools     */
     vgpu_load_pid_set(&g_state.pids);
 }
 

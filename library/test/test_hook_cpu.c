@@ -16,6 +16,8 @@
 #include <stdlib.h>
 #include <string.h>
 #include <time.h>
+#include <unistd.h>
+#include <sys/wait.h>
 
 extern uint64_t stub_count_malloc(void);
 extern uint64_t stub_count_managed(void);
@@ -129,6 +131,36 @@ static int scenario_throttle(void) {
     return 0;
 }
 
+static int scenario_fork(void) {
+    /* env: VGPU_CORE_LIMIT_0=50.  The watcher pthread does not survive
+     * fork(); without the atfork re-arm (vgpu_hook_fork_child) the
+     * child's first bucket depletion parks forever with no refiller.
+     * The alarm turns that hang into a crisp failure.                 */
+    dim3 grid = {16384, 1, 1}, block = {256, 1, 1};
+    for (int i = 0; i < 20; i++) /* parent: start watcher, drain bucket */
+        CHECK(hipLaunchKernel((void *)scenario_fork, grid, block, NULL, 0,
+                              NULL) == hipSuccess);
+    pid_t pid = fork();
+    CHECK(pid >= 0);
+    if (pid == 0) {
+        alarm(30);
+        for (int i = 0; i < 100; i++)
+            if (hipLaunchKernel((void *)scenario_fork, grid, block, NULL, 0,
+                                NULL) != hipSuccess)
+                _exit(1);
+        _exit(0);
+    }
+    int st = 0;
+    CHECK(waitpid(pid, &st, 0) == pid);
+    CHECK(WIFEXITED(st) && WEXITSTATUS(st) == 0);
+    /* parent keeps throttling normally after the fork                  */
+    for (int i = 0; i < 20; i++)
+        CHECK(hipLaunchKernel((void *)scenario_fork, grid, block, NULL, 0,
+                              NULL) == hipSuccess);
+    printf("PASS fork\n");
+    return 0;
+}
+
 int main(int argc, char **argv) {
     if (argc < 2) {
         fprintf(stderr, "usage: %s quota|oversold|launch|nolimit\n", argv[0]);
@@ -139,5 +171,6 @@ int main(int argc, char **argv) {
     if (strcmp(argv[1], "launch") == 0) return scenario_launch();
     if (strcmp(argv[1], "throttle") == 0) return scenario_throttle();
     if (strcmp(argv[1], "nolimit") == 0) return scenario_nolimit();
+    if (strcmp(argv[1], "fork") == 0) return scenario_fork();
     return 2;
 }

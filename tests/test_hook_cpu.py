@@ -34,6 +34,8 @@ def run_scenario(scenario, env_extra):
     ("oversold", {"VGPU_MEM_LIMIT_0": "1m", "VGPU_MEM_OVERSOLD": "1"}),
     ("launch", {"VGPU_CORE_LIMIT_0": "50"}),
     ("throttle", {"VGPU_CORE_LIMIT_0": "50"}),
+    # fork: child must restart the watcher (atfork re-arm) or it hangs
+    ("fork", {"VGPU_CORE_LIMIT_0": "50"}),
 ])
 def test_hook_scenario(built_library, scenario, env):
     run_scenario(scenario, env)
