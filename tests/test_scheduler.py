@@ -190,6 +190,55 @@ def test_preempt_finds_additional_victims(client):
     assert {p["UID"] for p in pods} == {"uid-v-a", "uid-v-b"}
 
 
+def test_preempt_pdb_blocks_extra_victims(client):
+    # same shape as additional-victims, but the would-be extra victim
+    # is protected by a PDB with 0 disruptions allowed -> node dropped
+    for i, name in enumerate(["v-a", "v-b"]):
+        v = make_pod(number=1, name=name)
+        v["spec"]["nodeName"] = "gpu-node-1"
+        v["spec"]["priority"] = 0
+        v["metadata"]["uid"] = f"uid-{name}"
+        v["metadata"]["labels"] = {"app": name}
+        v["metadata"]["annotations"][consts.real_alloc_ann()] = \
+            f"main[{i}_GPU-fake-{i:04d}_100_294912]"
+        client.add_pod(v)
+    client.add_pdb({
+        "metadata": {"name": "pdb-b", "namespace": "default"},
+        "spec": {"selector": {"matchLabels": {"app": "v-b"}}},
+        "status": {"disruptionsAllowed": 0},
+    })
+    pending = make_pod(number=2, name="pending2")
+    pending["spec"]["priority"] = 100
+    res = VgpuPreempter(client).preempt({
+        "Pod": pending,
+        "NodeNameToVictims": {
+            "gpu-node-1": {"Pods": [client.get_pod("default", "v-a")],
+                           "NumPDBViolations": 0}},
+    })
+    assert res["NodeNameToMetaVictims"] == {}
+
+
+def test_leader_elector_exclusive_and_takeover():
+    from vgpu_manager_amd.client.lease import LeaderElector
+    c = FakeKubeClient()
+    a = LeaderElector(c, "kube-system", "vgpu-scheduler", identity="a",
+                      lease_duration=1.0, renew_deadline=0.5,
+                      retry_period=0.05)
+    b = LeaderElector(c, "kube-system", "vgpu-scheduler", identity="b",
+                      lease_duration=1.0, renew_deadline=0.5,
+                      retry_period=0.05)
+    assert a.try_acquire_or_renew()
+    assert not b.try_acquire_or_renew()   # a holds a fresh lease
+    assert a.try_acquire_or_renew()       # renew keeps holding
+    import time as _t
+    _t.sleep(1.1)                         # a's lease expires
+    assert b.try_acquire_or_renew()       # b takes over
+    assert not a.try_acquire_or_renew()   # a sees b's fresh lease
+    lease = c.get_lease("kube-system", "vgpu-scheduler")
+    assert lease["spec"]["holderIdentity"] == "b"
+    assert lease["spec"]["leaseTransitions"] == 1
+
+
 def test_http_routes(client):
     from starlette.testclient import TestClient
     from vgpu_manager_amd.scheduler.http import create_app

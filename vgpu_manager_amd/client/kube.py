@@ -63,6 +63,22 @@ class KubeClient:
                      message: str, etype: str = "Warning") -> None:
         pass
 
+    # PodDisruptionBudgets (preempt victim protection,
+    # reference preempt_predicate.go:692)
+    def list_pdbs(self, namespace: Optional[str] = None) -> List[dict]:
+        return []
+
+    # coordination.k8s.io Leases (scheduler leader election,
+    # reference cmd/device-scheduler lease.go)
+    def get_lease(self, namespace: str, name: str) -> dict:
+        raise KubeError("leases unsupported")
+
+    def create_lease(self, namespace: str, lease: dict) -> dict:
+        raise KubeError("leases unsupported")
+
+    def update_lease(self, namespace: str, name: str, lease: dict) -> dict:
+        raise KubeError("leases unsupported")
+
 
 def _match_labels(meta: dict, selector: Optional[Dict[str, str]]) -> bool:
     if not selector:
@@ -81,6 +97,8 @@ class FakeKubeClient(KubeClient):
         self.bindings: List[tuple] = []
         self.evictions: List[tuple] = []
         self.events: List[dict] = []
+        self.pdbs: List[dict] = []
+        self.leases: Dict[tuple, dict] = {}
 
     # -- test helpers --
     def add_node(self, node: dict) -> None:
@@ -166,6 +184,47 @@ class FakeKubeClient(KubeClient):
         with self._mu:
             self.events.append(dict(namespace=namespace, reason=reason,
                                     message=message, type=etype))
+
+    def add_pdb(self, pdb: dict) -> None:
+        with self._mu:
+            self.pdbs.append(pdb)
+
+    def list_pdbs(self, namespace=None):
+        with self._mu:
+            return [json.loads(json.dumps(p)) for p in self.pdbs
+                    if namespace is None or
+                    p.get("metadata", {}).get("namespace",
+                                              "default") == namespace]
+
+    def get_lease(self, namespace, name):
+        with self._mu:
+            lease = self.leases.get((namespace, name))
+            if lease is None:
+                raise KubeError(f"lease {namespace}/{name} not found")
+            return json.loads(json.dumps(lease))
+
+    def create_lease(self, namespace, lease):
+        with self._mu:
+            key = (namespace, lease["metadata"]["name"])
+            if key in self.leases:
+                raise KubeError("lease exists")
+            lease.setdefault("metadata", {})["resourceVersion"] = "1"
+            self.leases[key] = lease
+            return json.loads(json.dumps(lease))
+
+    def update_lease(self, namespace, name, lease):
+        with self._mu:
+            cur = self.leases.get((namespace, name))
+            if cur is None:
+                raise KubeError(f"lease {namespace}/{name} not found")
+            rv_cur = cur.get("metadata", {}).get("resourceVersion", "1")
+            rv_new = lease.get("metadata", {}).get("resourceVersion", rv_cur)
+            if rv_new != rv_cur:
+                raise KubeError("lease conflict")
+            lease.setdefault("metadata", {})["resourceVersion"] = \
+                str(int(rv_cur) + 1)
+            self.leases[(namespace, name)] = lease
+            return json.loads(json.dumps(lease))
 
 
 class RestKubeClient(KubeClient):
@@ -272,3 +331,27 @@ class RestKubeClient(KubeClient):
                       body)
         except KubeError:
             pass
+
+    def list_pdbs(self, namespace=None):
+        path = (f"/apis/policy/v1/namespaces/{namespace}"
+                "/poddisruptionbudgets" if namespace else
+                "/apis/policy/v1/poddisruptionbudgets")
+        return self._req("GET", path).get("items", [])
+
+    def get_lease(self, namespace, name):
+        return self._req(
+            "GET",
+            f"/apis/coordination.k8s.io/v1/namespaces/{namespace}"
+            f"/leases/{name}")
+
+    def create_lease(self, namespace, lease):
+        return self._req(
+            "POST",
+            f"/apis/coordination.k8s.io/v1/namespaces/{namespace}/leases",
+            lease)
+
+    def update_lease(self, namespace, name, lease):
+        return self._req(
+            "PUT",
+            f"/apis/coordination.k8s.io/v1/namespaces/{namespace}"
+            f"/leases/{name}", lease)

@@ -1,0 +1,147 @@
+"""Lease-based leader election (reference cmd/device-scheduler/lease.go:
+client-go leaderelection over coordination.k8s.io Leases).
+
+The extender runs active/passive replicas: only the leader serves
+mutating verbs.  `LeaderElector` implements the same
+acquire/renew/yield protocol client-go uses — optimistic concurrency on
+the Lease's resourceVersion, renewDeadline < leaseDuration, jittered
+retry — against our `KubeClient` Lease surface.
+"""
+from __future__ import annotations
+
+import logging
+import threading
+import time
+import uuid
+from typing import Callable, Optional
+
+from .kube import KubeClient, KubeError
+
+log = logging.getLogger("vgpu.client.lease")
+
+
+def _now() -> str:
+    return time.strftime("%Y-%m-%dT%H:%M:%S.000000Z", time.gmtime())
+
+
+def _parse(ts: str) -> float:
+    try:
+        return time.mktime(time.strptime(ts.split(".")[0],
+                                         "%Y-%m-%dT%H:%M:%S")) - \
+            time.timezone
+    except Exception:
+        return 0.0
+
+
+class LeaderElector:
+    def __init__(self, client: KubeClient, namespace: str, name: str,
+                 identity: Optional[str] = None,
+                 lease_duration: float = 15.0,
+                 renew_deadline: float = 10.0,
+                 retry_period: float = 2.0,
+                 on_started_leading: Optional[Callable[[], None]] = None,
+                 on_stopped_leading: Optional[Callable[[], None]] = None):
+        if renew_deadline >= lease_duration:
+            raise ValueError("renew_deadline must be < lease_duration")
+        self.client = client
+        self.namespace = namespace
+        self.name = name
+        self.identity = identity or f"{uuid.uuid4().hex[:12]}"
+        self.lease_duration = lease_duration
+        self.renew_deadline = renew_deadline
+        self.retry_period = retry_period
+        self.on_started_leading = on_started_leading
+        self.on_stopped_leading = on_stopped_leading
+        self._leading = False
+        self._stop = threading.Event()
+
+    @property
+    def leading(self) -> bool:
+        return self._leading
+
+    def stop(self) -> None:
+        self._stop.set()
+
+    # ---- one acquisition/renew attempt; returns True if we hold it ----
+    def try_acquire_or_renew(self) -> bool:
+        spec = {
+            "holderIdentity": self.identity,
+            "leaseDurationSeconds": max(1, int(round(self.lease_duration))),
+            "renewTime": _now(),
+        }
+        try:
+            lease = self.client.get_lease(self.namespace, self.name)
+        except KubeError:
+            body = {"apiVersion": "coordination.k8s.io/v1", "kind": "Lease",
+                    "metadata": {"name": self.name,
+                                 "namespace": self.namespace},
+                    "spec": dict(spec, acquireTime=_now(),
+                                 leaseTransitions=0)}
+            try:
+                self.client.create_lease(self.namespace, body)
+                return True
+            except KubeError:
+                return False
+
+        cur = lease.get("spec", {})
+        holder = cur.get("holderIdentity", "")
+        if holder and holder != self.identity:
+            renew = _parse(cur.get("renewTime", ""))
+            dur = cur.get("leaseDurationSeconds", self.lease_duration)
+            if renew and time.time() - renew < dur:
+                return False  # current holder still valid
+            spec["leaseTransitions"] = \
+                (cur.get("leaseTransitions", 0) or 0) + 1
+            spec["acquireTime"] = _now()
+        elif holder == self.identity:
+            spec["acquireTime"] = cur.get("acquireTime", _now())
+            spec["leaseTransitions"] = cur.get("leaseTransitions", 0)
+        else:
+            spec["acquireTime"] = _now()
+            spec["leaseTransitions"] = cur.get("leaseTransitions", 0)
+
+        lease["spec"] = spec
+        try:
+            self.client.update_lease(self.namespace, self.name, lease)
+            return True
+        except KubeError:
+            return False  # conflict: someone else raced us
+
+    def run(self) -> None:
+        """Block until stopped; maintains leadership when acquired."""
+        last_renew = 0.0
+        while not self._stop.is_set():
+            ok = self.try_acquire_or_renew()
+            now = time.time()
+            if ok:
+                last_renew = now
+                if not self._leading:
+                    self._leading = True
+                    log.info("became leader (%s)", self.identity)
+                    if self.on_started_leading:
+                        self.on_started_leading()
+            elif self._leading and now - last_renew > self.renew_deadline:
+                self._leading = False
+                log.warning("lost leadership (%s)", self.identity)
+                if self.on_stopped_leading:
+                    self.on_stopped_leading()
+            self._stop.wait(self.retry_period)
+        if self._leading:
+            self._leading = False
+            # best-effort release so the next replica acquires fast
+            try:
+                lease = self.client.get_lease(self.namespace, self.name)
+                if lease.get("spec", {}).get("holderIdentity") == \
+                        self.identity:
+                    lease["spec"]["holderIdentity"] = ""
+                    self.client.update_lease(self.namespace, self.name,
+                                             lease)
+            except KubeError:
+                pass
+            if self.on_stopped_leading:
+                self.on_stopped_leading()
+
+    def run_background(self) -> threading.Thread:
+        t = threading.Thread(target=self.run, daemon=True)
+        t.start()
+        return t

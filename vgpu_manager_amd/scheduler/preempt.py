@@ -30,6 +30,25 @@ def _pod_priority(pod: dict) -> int:
     return pod.get("spec", {}).get("priority", 0) or 0
 
 
+def _pdb_blocks(pod: dict, pdbs: List[dict]) -> bool:
+    """True if evicting `pod` would violate a PodDisruptionBudget
+    (reference preempt_predicate.go:692: extra victims must not break a
+    PDB; pre-proposed victims keep their upstream NumPDBViolations)."""
+    meta = pod.get("metadata", {})
+    ns = meta.get("namespace", "default")
+    labels = meta.get("labels", {}) or {}
+    for pdb in pdbs:
+        if pdb.get("metadata", {}).get("namespace", "default") != ns:
+            continue
+        sel = (pdb.get("spec", {}).get("selector", {})
+               or {}).get("matchLabels", {})
+        if not sel or not all(labels.get(k) == v for k, v in sel.items()):
+            continue
+        if (pdb.get("status", {}).get("disruptionsAllowed", 0) or 0) <= 0:
+            return True
+    return False
+
+
 class VgpuPreempter:
     def __init__(self, client: KubeClient):
         self.client = client
@@ -99,12 +118,18 @@ class VgpuPreempter:
         if can_allocate(set()):
             return victims
 
-        # search additional lower-priority vGPU victims
+        # search additional lower-priority vGPU victims; never pick one
+        # whose eviction would violate a PodDisruptionBudget
+        try:
+            pdbs = self.client.list_pdbs()
+        except KubeError:
+            pdbs = []
         pending_prio = _pod_priority(pending)
         extra_candidates = sorted(
             (p for p in pods
              if pod_claim_annotation(p)
              and _pod_priority(p) < pending_prio
+             and not _pdb_blocks(p, pdbs)
              and (p.get("metadata", {}).get("namespace", "default"),
                   p.get("metadata", {}).get("name")) not in victim_keys),
             key=_pod_priority)
