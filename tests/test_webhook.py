@@ -95,3 +95,73 @@ def test_admission_review_roundtrip():
                                   mutating=False)
     assert out["response"]["allowed"] is False
     assert out["response"]["status"]["code"] == 400
+
+
+def test_volcano_job_mutate_and_validate():
+    from vgpu_manager_amd.webhook.admission import (
+        mutate_volcano_job,
+        validate_volcano_job,
+    )
+    task_pod = make_pod(number=1)
+    job = {"metadata": {"name": "j"},
+           "spec": {"tasks": [
+               {"name": "t0", "replicas": 2,
+                "template": {"metadata": task_pod["metadata"],
+                             "spec": task_pod["spec"]}},
+               {"name": "t1",
+                "template": {"spec": {"containers": [
+                    {"name": "c", "resources": {}}]}}},
+           ]}}
+    patches = mutate_volcano_job(job)
+    assert any(p["path"] == "/spec/tasks/0/template/spec/schedulerName"
+               for p in patches)
+    # non-vgpu task untouched
+    assert not any(p["path"].startswith("/spec/tasks/1") for p in patches)
+    ok, _ = validate_volcano_job(job)
+    assert ok
+
+    bad = {"spec": {"tasks": [{"name": "t",
+                               "template": {"spec": make_pod(
+                                   number=99)["spec"]}}]}}
+    ok, msg = validate_volcano_job(bad)
+    assert not ok and "out of range" in msg
+
+
+def test_resource_claim_validate():
+    from vgpu_manager_amd.webhook.admission import validate_resource_claim
+    good = {"spec": {"devices": {
+        "requests": [{"name": "r", "deviceClassName": "vgpu-manager",
+                      "count": 2}],
+        "config": [{"opaque": {"parameters": {"cores": 50,
+                                              "memoryMiB": 4096}}}],
+    }}}
+    assert validate_resource_claim(good) == (True, "")
+    # foreign driver's class is never vetoed
+    foreign = {"spec": {"devices": {"requests": [
+        {"name": "r", "deviceClassName": "other.example.com",
+         "count": 9999}]}}}
+    assert validate_resource_claim(foreign)[0]
+    bad_count = {"spec": {"devices": {"requests": [
+        {"name": "r", "deviceClassName": "gpu-manager", "count": 99}]}}}
+    assert not validate_resource_claim(bad_count)[0]
+    bad_cores = {"spec": {"devices": {
+        "requests": [{"name": "r", "deviceClassName": "vgpu-manager"}],
+        "config": [{"opaque": {"parameters": {"cores": 150}}}]}}}
+    assert not validate_resource_claim(bad_cores)[0]
+
+
+def test_webhook_http_new_paths():
+    from starlette.testclient import TestClient
+    from vgpu_manager_amd.webhook.admission import create_app
+    tc = TestClient(create_app())
+    job = {"spec": {"tasks": [{"name": "t", "template": {
+        "spec": make_pod(number=1)["spec"]}}]}}
+    r = tc.post("/webhook/mutate-volcanojob",
+                json={"request": {"uid": "u", "object": job}})
+    assert r.json()["response"]["allowed"] is True
+    r = tc.post("/webhook/validate-resourceclaim",
+                json={"request": {"uid": "u", "object": {
+                    "spec": {"devices": {"requests": [
+                        {"name": "r", "deviceClassName": "gpu-manager",
+                         "count": 99}]}}}}})
+    assert r.json()["response"]["allowed"] is False

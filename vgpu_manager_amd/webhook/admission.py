@@ -159,22 +159,102 @@ def validate_pod(pod: dict) -> Tuple[bool, str]:
     return True, ""
 
 
+# ---- volcano Job (reference registry.go:60-63, volcano mutate/
+# validate run the pod logic over every task pod template) ----
+
+def mutate_volcano_job(job: dict, *,
+                       default_scheduler: str = VGPU_SCHEDULER_NAME
+                       ) -> List[dict]:
+    patches: List[dict] = []
+    tasks = (job.get("spec", {}) or {}).get("tasks") or []
+    for ti, task in enumerate(tasks):
+        tmpl = (task.get("template") or {})
+        pod = {"metadata": tmpl.get("metadata", {}) or {},
+               "spec": tmpl.get("spec", {}) or {}}
+        for p in mutate_pod(pod, default_scheduler=default_scheduler):
+            p = dict(p)
+            p["path"] = f"/spec/tasks/{ti}/template" + p["path"]
+            patches.append(p)
+    return patches
+
+
+def validate_volcano_job(job: dict) -> Tuple[bool, str]:
+    tasks = (job.get("spec", {}) or {}).get("tasks") or []
+    for task in tasks:
+        tmpl = task.get("template") or {}
+        pod = {"metadata": tmpl.get("metadata", {}) or {},
+               "spec": tmpl.get("spec", {}) or {}}
+        ok, msg = validate_pod(pod)
+        if not ok:
+            return False, f"task {task.get('name')}: {msg}"
+    return True, ""
+
+
+# ---- ResourceClaim validate (reference resourceclaim_validate path:
+# claims against our device classes must carry sane vgpu config) ----
+
+_VALID_DEVICE_CLASSES = {"gpu-manager", "vgpu-manager", "cpx-manager",
+                         "vfio-manager"}
+
+
+def validate_resource_claim(claim: dict) -> Tuple[bool, str]:
+    spec = (claim.get("spec", {}) or {}).get("devices", {}) or {}
+    for req in spec.get("requests") or []:
+        cls = req.get("deviceClassName", "")
+        exactly = req.get("exactly") or {}
+        cls = cls or exactly.get("deviceClassName", "")
+        if cls and cls not in _VALID_DEVICE_CLASSES:
+            # not ours — never veto foreign drivers' claims
+            continue
+        count = req.get("count", exactly.get("count", 1)) or 1
+        try:
+            count = int(count)
+        except (TypeError, ValueError):
+            return False, f"request {req.get('name')}: count not integer"
+        if not 1 <= count <= consts.MAX_DEVICE_COUNT:
+            return False, (f"request {req.get('name')}: count {count} out "
+                           f"of range 1..{consts.MAX_DEVICE_COUNT}")
+    for cfg in spec.get("config") or []:
+        opaque = (cfg.get("opaque") or {}).get("parameters") or {}
+        cores = opaque.get("cores")
+        if cores is not None:
+            try:
+                cr = int(cores)
+            except (TypeError, ValueError):
+                return False, f"config cores not integer: {cores!r}"
+            if not 0 <= cr <= consts.CORES_PER_GPU:
+                return False, f"config cores {cr} out of range"
+        mem = opaque.get("memoryMiB")
+        if mem is not None and int(mem) < 0:
+            return False, "config memoryMiB negative"
+    return True, ""
+
+
 # ---- AdmissionReview plumbing ----
 
 def handle_admission_review(body: dict, *, mutating: bool,
-                            dra_mode: bool = False) -> dict:
+                            dra_mode: bool = False,
+                            kind: str = "Pod") -> dict:
     req = body.get("request", {}) or {}
     uid = req.get("uid", "")
     obj = req.get("object", {}) or {}
     response = {"uid": uid, "allowed": True}
     if mutating:
-        patches = mutate_pod(obj, dra_mode=dra_mode)
+        if kind == "VolcanoJob":
+            patches = mutate_volcano_job(obj)
+        else:
+            patches = mutate_pod(obj, dra_mode=dra_mode)
         if patches:
             response["patchType"] = "JSONPatch"
             response["patch"] = base64.b64encode(
                 json.dumps(patches).encode()).decode()
     else:
-        ok, msg = validate_pod(obj)
+        if kind == "VolcanoJob":
+            ok, msg = validate_volcano_job(obj)
+        elif kind == "ResourceClaim":
+            ok, msg = validate_resource_claim(obj)
+        else:
+            ok, msg = validate_pod(obj)
         response["allowed"] = ok
         if not ok:
             response["status"] = {"message": msg, "code": 400}
@@ -209,7 +289,11 @@ def apply_json_patch(obj: dict, patches: List[dict]) -> dict:
 
 
 def create_app(dra_mode: bool = False):
+    # module-level import would make fastapi a hard dependency of every
+    # admission-logic consumer; but the Request annotation must resolve
+    # from module globals (PEP 563 strings) — so stash it there.
     from fastapi import FastAPI, Request
+    globals()["Request"] = Request
 
     app = FastAPI(title="vgpu-webhook")
 
@@ -222,6 +306,22 @@ def create_app(dra_mode: bool = False):
     async def validate(request: Request):
         return handle_admission_review(await request.json(),
                                        mutating=False)
+
+    @app.post("/webhook/mutate-volcanojob")
+    async def mutate_vcjob(request: Request):
+        return handle_admission_review(await request.json(),
+                                       mutating=True, kind="VolcanoJob")
+
+    @app.post("/webhook/validate-volcanojob")
+    async def validate_vcjob(request: Request):
+        return handle_admission_review(await request.json(),
+                                       mutating=False, kind="VolcanoJob")
+
+    @app.post("/webhook/validate-resourceclaim")
+    async def validate_claim(request: Request):
+        return handle_admission_review(await request.json(),
+                                       mutating=False,
+                                       kind="ResourceClaim")
 
     @app.get("/healthz")
     async def healthz():
