@@ -133,3 +133,148 @@ def test_cdi_spec_write(tmp_path):
     nodes = data["devices"][0]["containerEdits"]["deviceNodes"]
     assert {n["path"] for n in nodes} == {"/dev/kfd",
                                           "/dev/dri/renderD128"}
+
+
+# ---- sharing strategies (reference sharing.go analog) ----
+
+def test_sharing_time_slicing_equal_and_percents():
+    from vgpu_manager_amd.dra.sharing import apply_sharing_config
+    params = [VgpuClaimParams(uuid="GPU-fake-0000"),
+              VgpuClaimParams(uuid="GPU-fake-0000")]
+    dec = apply_sharing_config(params, {"strategy": "time-slicing"})
+    assert dec.core_limits == {0: 50, 1: 50}
+    dec = apply_sharing_config(params, {"strategy": "time-slicing",
+                                        "percents": [70, 30]})
+    assert dec.core_limits == {0: 70, 1: 30}
+
+
+def test_sharing_cu_partition_disjoint():
+    from vgpu_manager_amd.dra.sharing import (
+        SharingError,
+        apply_sharing_config,
+    )
+    params = [VgpuClaimParams(uuid="GPU-fake-0000") for _ in range(4)]
+    dec = apply_sharing_config(params, {"strategy": "cu-partition"})
+    seen = []
+    for idx in range(4):
+        assert len(dec.partitions[idx]) == 2  # 8 XCDs / 4 consumers
+        seen += dec.partitions[idx]
+    assert sorted(seen) == list(range(8))  # disjoint cover
+    with pytest.raises(SharingError):
+        apply_sharing_config(params, {"strategy": "cu-partition",
+                                      "partitionsPerConsumer": 3})
+    with pytest.raises(SharingError):
+        apply_sharing_config(params, {"strategy": "bogus"})
+
+
+def test_prepare_applies_sharing(state):
+    prepared = state.prepare(
+        "claim-ts",
+        [VgpuClaimParams(uuid="GPU-fake-0000", partition_key="a"),
+         VgpuClaimParams(uuid="GPU-fake-0000", partition_key="b")],
+        sharing_config={"strategy": "time-slicing"})
+    for key in ("a", "b"):
+        pdir = os.path.join(prepared.container_dir, key)
+        snap = VgpuConfigReader(
+            os.path.join(pdir, "config", "vgpu.config")).snapshot()
+        assert snap["devices"][0]["core_limit"] == 50
+
+
+# ---- VFIO passthrough (reference vfio-device.go analog) ----
+
+def _fake_pci(tmp_path, bdf="0000:03:00.0", driver="amdgpu", group="42"):
+    root = tmp_path / "sys"
+    dev = root / "bus" / "pci" / "devices" / bdf
+    dev.mkdir(parents=True)
+    for drv in (driver, "vfio-pci"):
+        d = root / "bus" / "pci" / "drivers" / drv
+        d.mkdir(parents=True, exist_ok=True)
+        (d / "unbind").write_text("")
+        (d / "bind").write_text("")
+    (root / "bus" / "pci" / "drivers_probe").write_text("")
+    (dev / "driver_override").write_text("")
+    os.symlink(str(root / "bus" / "pci" / "drivers" / driver),
+               str(dev / "driver"))
+    grp = root / "kernel" / "iommu_groups" / group
+    grp.mkdir(parents=True)
+    os.symlink(str(grp), str(dev / "iommu_group"))
+    return str(root), bdf
+
+
+def test_vfio_bind_unbind(tmp_path):
+    from vgpu_manager_amd.dra.vfio import VfioManager
+    root, bdf = _fake_pci(tmp_path)
+    m = VfioManager(sysfs_root=root)
+    assert m.current_driver(bdf) == "amdgpu"
+    node = m.bind_vfio(bdf)
+    assert node == "/dev/vfio/42"
+    # unbind wrote the bdf to amdgpu's unbind + set the override
+    unbind = open(os.path.join(root, "bus", "pci", "drivers",
+                               "amdgpu", "unbind")).read()
+    assert bdf in unbind
+    override = open(os.path.join(root, "bus", "pci", "devices", bdf,
+                                 "driver_override")).read()
+    assert "vfio-pci" in override
+    edits = m.container_edits(bdf, node)
+    assert {d["path"] for d in edits["deviceNodes"]} == \
+        {"/dev/vfio/42", "/dev/vfio/vfio"}
+    m.unbind_vfio(bdf)
+    override = open(os.path.join(root, "bus", "pci", "devices", bdf,
+                                 "driver_override")).read()
+    assert "vfio-pci" not in override
+
+
+def test_vfio_missing_device(tmp_path):
+    from vgpu_manager_amd.dra.vfio import VfioError, VfioManager
+    m = VfioManager(sysfs_root=str(tmp_path / "nosys"))
+    with pytest.raises(VfioError):
+        m.bind_vfio("0000:99:00.0")
+
+
+# ---- NRI-analog hook (reference nri/plugin.go) ----
+
+def test_nri_injects_partition_mounts(state):
+    from vgpu_manager_amd.dra.nri import (
+        ENV_CLAIM_UID,
+        ENV_PARTITION_KEY,
+        NriHook,
+    )
+    state.prepare("claim-n", [
+        VgpuClaimParams(uuid="GPU-fake-0000", partition_key="cont-a"),
+        VgpuClaimParams(uuid="GPU-fake-0001", partition_key="cont-b"),
+    ])
+    hook = NriHook(state)
+    pod = {"uid": "pu", "name": "p"}
+    cont = {"name": "cont-a",
+            "env": [f"{ENV_CLAIM_UID}=claim-n",
+                    f"{ENV_PARTITION_KEY}=cont-a"]}
+    adj = hook.create_container(pod, cont)
+    assert adj is not None
+    dests = {m["destination"] for m in adj.mounts}
+    assert "/tmp/.vgpu_lock" in dests and "/tmp/.sm_node" in dests
+    srcs = {m["source"] for m in adj.mounts}
+    assert all("cont-a" in s for s in srcs)  # ONLY its partition
+
+    # forged claim uid -> refused
+    forged = {"name": "x", "env": [f"{ENV_CLAIM_UID}=claim-evil"]}
+    assert hook.create_container(pod, forged) is None
+    # wrong partition key -> refused
+    wrong = {"name": "x", "env": [f"{ENV_CLAIM_UID}=claim-n",
+                                  f"{ENV_PARTITION_KEY}=nope"]}
+    assert hook.create_container(pod, wrong) is None
+    # non-claim container -> no adjustment
+    assert hook.create_container(pod, {"name": "y", "env": []}) is None
+    # dry-run observes, never injects
+    assert NriHook(state, dry_run=True).create_container(pod, cont) \
+        is None
+
+
+def test_nri_synchronize_cache(state):
+    from vgpu_manager_amd.dra.nri import ENV_CLAIM_UID, NriHook
+    hook = NriHook(state)
+    pods = [{"id": "sb1", "uid": "pu", "name": "p"}]
+    conts = [{"name": "c1", "pod_sandbox_id": "sb1",
+              "env": [f"{ENV_CLAIM_UID}=claim-n"]},
+             {"name": "c2", "pod_sandbox_id": "sb-unknown", "env": []}]
+    hook.synchronize(pods, conts)
+    assert hook._cache == {("pu", "c1"): "claim-n"}

@@ -1,0 +1,124 @@
+"""NRI-style per-container mount isolation (reference
+pkg/kubeletplugin/nri/plugin.go).
+
+DRA's CDI edits are per-CLAIM; when several containers share one claim
+with different partition keys, each container must see only ITS
+partition directory.  The reference solves this with an NRI plugin on
+containerd's ttrpc socket; this module implements the same logic as a
+transport-independent hook:
+
+  * `synchronize(pods, containers)` rebuilds the container cache;
+  * `create_container(pod, container)` validates the env-carried claim
+    UID against node prepared state and returns the mount + env
+    adjustments for that container's partition;
+  * dry-run observe mode reports what WOULD be injected.
+
+A containerd front-end (ttrpc framing) can wrap this object; every
+behavior is testable hermetically through the two methods, which is
+exactly how the reference tests its plugin (nri/plugin_test.go).
+"""
+from __future__ import annotations
+
+import logging
+import os
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+from ..util import consts
+from .state import DeviceState
+
+log = logging.getLogger("vgpu.dra.nri")
+
+# env the CDI claim edits plant so the hook can find the partition
+ENV_CLAIM_UID = "VGPU_CLAIM_UID"
+ENV_PARTITION_KEY = "VGPU_PARTITION_KEY"
+
+
+@dataclass
+class ContainerAdjustment:
+    mounts: List[dict] = field(default_factory=list)
+    env: Dict[str, str] = field(default_factory=dict)
+
+
+class NriHook:
+    def __init__(self, state: DeviceState, *, dry_run: bool = False):
+        self.state = state
+        self.dry_run = dry_run
+        # (pod_uid, container_name) -> claim_uid, rebuilt on synchronize
+        self._cache: Dict[Tuple[str, str], str] = {}
+
+    # ---- Synchronize (reference plugin.go:248) ----
+    def synchronize(self, pods: List[dict],
+                    containers: List[dict]) -> None:
+        self._cache.clear()
+        by_id = {p.get("id"): p for p in pods}
+        for c in containers:
+            pod = by_id.get(c.get("pod_sandbox_id"))
+            if pod is None:
+                continue
+            envs = _env_map(c)
+            claim = envs.get(ENV_CLAIM_UID)
+            if claim:
+                self._cache[(pod.get("uid", ""), c.get("name", ""))] = \
+                    claim
+        log.info("nri synchronize: %d tracked containers",
+                 len(self._cache))
+
+    # ---- CreateContainer (reference plugin.go:310-440) ----
+    def create_container(self, pod: dict, container: dict
+                         ) -> Optional[ContainerAdjustment]:
+        envs = _env_map(container)
+        claim_uid = envs.get(ENV_CLAIM_UID)
+        if not claim_uid:
+            return None  # not a vgpu-claim container
+        partition = envs.get(ENV_PARTITION_KEY,
+                             container.get("name", "default"))
+
+        # validate against node prepared state: an env forged by the
+        # pod author must not grant access to someone else's claim dir
+        entry = self.state.checkpoint.claims.get(claim_uid)
+        if entry is None:
+            log.warning("nri: claim %s not prepared on this node; "
+                        "refusing injection", claim_uid)
+            return None
+        base = entry["container_dir"]
+        pdir = os.path.join(base, partition)
+        if not os.path.isdir(pdir):
+            keys = sorted(os.listdir(base)) if os.path.isdir(base) else []
+            log.warning("nri: partition %r not in claim %s (have %s)",
+                        partition, claim_uid, keys)
+            return None
+
+        adj = ContainerAdjustment()
+        adj.mounts = [
+            {"source": os.path.join(pdir, "config"),
+             "destination": f"{consts.MANAGER_DIR}/config",
+             "options": ["bind", "ro"]},
+            {"source": os.path.join(pdir, "vgpu_lock"),
+             "destination": "/tmp/.vgpu_lock",
+             "options": ["bind", "rw"]},
+            {"source": os.path.join(pdir, "vmem_node"),
+             "destination": "/tmp/.vmem_node",
+             "options": ["bind", "rw"]},
+            {"source": os.path.join(pdir, "sm_node"),
+             "destination": "/tmp/.sm_node",
+             "options": ["bind", "rw"]},
+        ]
+        adj.env[ENV_PARTITION_KEY] = partition
+        if self.dry_run:
+            log.info("nri dry-run: would adjust %s/%s with %d mounts",
+                     pod.get("name"), container.get("name"),
+                     len(adj.mounts))
+            return None
+        return adj
+
+
+def _env_map(container: dict) -> Dict[str, str]:
+    out: Dict[str, str] = {}
+    for e in container.get("env") or []:
+        if isinstance(e, str) and "=" in e:
+            k, v = e.split("=", 1)
+            out[k] = v
+        elif isinstance(e, dict):
+            out[e.get("name", "")] = e.get("value", "")
+    return out

@@ -101,6 +101,9 @@ class VgpuClaimParams:
     cores: int = 0
     memory_mib: int = 0
     partition_key: str = "default"  # multi-container claim partitions
+    # cu-partition sharing: XCD/CPX partition indices pinned to this
+    # consumer (dra/sharing.py); empty = whole-GPU temporal sharing
+    cpx_partitions: List[int] = field(default_factory=list)
 
 
 @dataclass
@@ -191,14 +194,26 @@ class DeviceState:
 
     # ---- prepare ----
     def prepare(self, claim_uid: str, params: List[VgpuClaimParams],
-                pod_meta: Optional[dict] = None) -> PreparedDevice:
-        """Idempotent per claim (reference DeviceState.Prepare:299)."""
+                pod_meta: Optional[dict] = None,
+                sharing_config: Optional[dict] = None) -> PreparedDevice:
+        """Idempotent per claim (reference DeviceState.Prepare:299).
+        `sharing_config` is the claim's opaque sharing parameters
+        (reference sharing.go; see dra/sharing.py)."""
         with self._mu:
             existing = self.checkpoint.claims.get(claim_uid)
             if existing:
                 return PreparedDevice(
                     cdi_device_ids=existing["cdi_device_ids"],
                     container_dir=existing["container_dir"])
+
+            if sharing_config:
+                from .sharing import apply_sharing_config
+                decision = apply_sharing_config(params, sharing_config)
+                for i, p in enumerate(params):
+                    if i in decision.core_limits:
+                        p.cores = decision.core_limits[i]
+                    if i in decision.partitions:
+                        p.cpx_partitions = decision.partitions[i]
 
             by_partition: Dict[str, List[VgpuClaimParams]] = {}
             for p in params:
@@ -226,6 +241,9 @@ class DeviceState:
                     if p.cores:
                         envs[consts.ENV_CORE_LIMIT.format(k)] = \
                             str(p.cores)
+                    if p.cpx_partitions:
+                        envs[f"VGPU_CPX_PARTITIONS_{k}"] = ",".join(
+                            str(x) for x in p.cpx_partitions)
                     cdi_ids.append(cdi.qualified_name(p.uuid))
                 meta = pod_meta or {}
                 w = VgpuConfigWriter(

@@ -1,0 +1,100 @@
+"""VFIO passthrough preparation (reference pkg/kubeletplugin/
+vfio-device.go): hand a whole GPU to a VM-style consumer by unbinding
+it from amdgpu and binding it to vfio-pci.
+
+The mechanics are pure sysfs (driver_override + bind/unbind), so the
+device class works on MI355X unchanged from any PCI device.  The sysfs
+root is injectable so the whole flow is hermetically testable; the
+production root is `/sys`.
+"""
+from __future__ import annotations
+
+import logging
+import os
+from typing import List, Optional
+
+log = logging.getLogger("vgpu.dra.vfio")
+
+
+class VfioError(Exception):
+    pass
+
+
+class VfioManager:
+    def __init__(self, sysfs_root: str = "/sys"):
+        self.root = sysfs_root
+
+    # ---- paths ----
+    def _dev_dir(self, bdf: str) -> str:
+        return os.path.join(self.root, "bus", "pci", "devices", bdf)
+
+    def _drv_dir(self, driver: str) -> str:
+        return os.path.join(self.root, "bus", "pci", "drivers", driver)
+
+    def _write(self, path: str, value: str) -> None:
+        try:
+            with open(path, "w") as f:
+                f.write(value)
+        except OSError as e:
+            raise VfioError(f"write {path}: {e}") from e
+
+    # ---- queries ----
+    def current_driver(self, bdf: str) -> Optional[str]:
+        link = os.path.join(self._dev_dir(bdf), "driver")
+        try:
+            return os.path.basename(os.readlink(link))
+        except OSError:
+            return None
+
+    def iommu_group(self, bdf: str) -> Optional[str]:
+        link = os.path.join(self._dev_dir(bdf), "iommu_group")
+        try:
+            return os.path.basename(os.readlink(link))
+        except OSError:
+            return None
+
+    # ---- bind flow (reference vfio-device.go prepare) ----
+    def bind_vfio(self, bdf: str) -> str:
+        """Returns the /dev/vfio/<group> node the container needs."""
+        dev = self._dev_dir(bdf)
+        if not os.path.isdir(dev):
+            raise VfioError(f"no PCI device {bdf}")
+        cur = self.current_driver(bdf)
+        if cur == "vfio-pci":
+            pass  # idempotent
+        else:
+            if cur is not None:
+                self._write(os.path.join(self._drv_dir(cur), "unbind"),
+                            bdf)
+            self._write(os.path.join(dev, "driver_override"), "vfio-pci")
+            # drivers_probe re-runs driver matching with the override
+            self._write(os.path.join(self.root, "bus", "pci",
+                                     "drivers_probe"), bdf)
+        group = self.iommu_group(bdf)
+        if group is None:
+            raise VfioError(f"{bdf}: no iommu_group (IOMMU off?)")
+        log.info("vfio bind %s -> group %s", bdf, group)
+        return f"/dev/vfio/{group}"
+
+    def unbind_vfio(self, bdf: str, rebind_driver: str = "amdgpu") -> None:
+        """Return the device to the host driver (unprepare)."""
+        dev = self._dev_dir(bdf)
+        cur = self.current_driver(bdf)
+        if cur == "vfio-pci":
+            self._write(os.path.join(self._drv_dir("vfio-pci"),
+                                     "unbind"), bdf)
+        self._write(os.path.join(dev, "driver_override"), "\n")
+        self._write(os.path.join(self.root, "bus", "pci",
+                                 "drivers_probe"), bdf)
+        log.info("vfio unbind %s (rebind %s)", bdf, rebind_driver)
+
+    def container_edits(self, bdf: str, group_node: str) -> dict:
+        """CDI edits for a VFIO consumer: the group node + control node."""
+        return {
+            "deviceNodes": [
+                {"path": group_node, "type": "c"},
+                {"path": "/dev/vfio/vfio", "type": "c"},
+            ],
+            "env": [f"VFIO_GROUP={os.path.basename(group_node)}",
+                    f"VFIO_DEVICE_BDF={bdf}"],
+        }
