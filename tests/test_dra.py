@@ -278,3 +278,129 @@ def test_nri_synchronize_cache(state):
              {"name": "c2", "pod_sandbox_id": "sb-unknown", "env": []}]
     hook.synchronize(pods, conts)
     assert hook._cache == {("pu", "c1"): "claim-n"}
+
+
+# ---- claim resolution (reference pkg/claimresolve) ----
+
+def _alloc_claim(uid="cu-1", device="GPU-fake-0000", configs=None,
+                 driver=None):
+    from vgpu_manager_amd.dra.state import DRA_DRIVER_NAME
+    return {
+        "metadata": {"name": "c", "namespace": "default", "uid": uid},
+        "status": {"allocation": {"devices": {
+            "results": [{"request": "gpu",
+                         "driver": driver or DRA_DRIVER_NAME,
+                         "pool": "node-a", "device": device}],
+            "config": configs or [],
+        }}},
+    }
+
+
+def test_resolve_claim_basic_and_configs():
+    from vgpu_manager_amd.dra.resolve import resolve_claim
+    claim = _alloc_claim(configs=[
+        {"opaque": {"parameters": {"cores": 50, "memoryMiB": 4096}}},
+        {"requests": ["gpu"],
+         "opaque": {"parameters": {"partitionKey": "side"}}},
+    ])
+    params, sharing = resolve_claim(claim)
+    assert sharing is None
+    assert len(params) == 1
+    p = params[0]
+    assert (p.uuid, p.cores, p.memory_mib, p.partition_key) == \
+        ("GPU-fake-0000", 50, 4096, "side")
+
+
+def test_resolve_claim_cpx_and_foreign():
+    from vgpu_manager_amd.dra.resolve import resolve_claim
+    params, _ = resolve_claim(_alloc_claim(
+        device="GPU-fake-0000-cpx-3"))
+    assert params[0].uuid == "GPU-fake-0000"
+    assert params[0].cpx_partitions == [3]
+    params, _ = resolve_claim(_alloc_claim(driver="other.example.com"))
+    assert params == []
+
+
+def test_resolve_claim_sharing_config():
+    from vgpu_manager_amd.dra.resolve import resolve_claim
+    _, sharing = resolve_claim(_alloc_claim(configs=[
+        {"opaque": {"parameters": {"strategy": "time-slicing",
+                                   "slices": 4}}}]))
+    assert sharing["strategy"] == "time-slicing"
+
+
+# ---- DRA gRPC driver e2e over a unix socket ----
+
+def test_dra_driver_grpc_prepare_unprepare(state, tmp_path):
+    import grpc as _grpc
+    from vgpu_manager_amd.client.kube import FakeKubeClient
+    from vgpu_manager_amd.dra import api as dapi
+    from vgpu_manager_amd.dra.driver import DraDriver, DraDriverServer
+
+    client = FakeKubeClient()
+    client.add_resource_claim(_alloc_claim(uid="uid-1", configs=[
+        {"opaque": {"parameters": {"cores": 25, "memoryMiB": 2048}}}]))
+
+    endpoint = str(tmp_path / "plugins" / "drv" / "dra.sock")
+    driver = DraDriver(state, client, endpoint=endpoint)
+    server = DraDriverServer(
+        driver, plugins_dir=str(tmp_path / "plugins"),
+        plugins_registry=str(tmp_path / "registry"))
+    server.start()
+    try:
+        ch = _grpc.insecure_channel(f"unix://{endpoint}")
+        prep = ch.unary_unary(
+            f"/{dapi.DRA_SERVICE}/NodePrepareResources",
+            request_serializer=lambda m: m.encode(),
+            response_deserializer=dapi.NodePrepareResourcesResponse
+            .decode)
+        req = dapi.NodePrepareResourcesRequest(claims=[
+            dapi.Claim(namespace="default", uid="uid-1", name="c")])
+        resp = prep(req, timeout=10)
+        assert len(resp.claims) == 1
+        entry = resp.claims[0]
+        assert entry.key == "uid-1"
+        assert entry.value.error == ""
+        assert entry.value.devices[0].cdi_device_ids
+        assert "uid-1" in state.prepared_claims()
+
+        # unknown claim -> per-claim error, rpc still OK
+        resp = prep(dapi.NodePrepareResourcesRequest(claims=[
+            dapi.Claim(namespace="default", uid="uid-x", name="nope")]),
+            timeout=10)
+        assert resp.claims[0].value.error != ""
+
+        unprep = ch.unary_unary(
+            f"/{dapi.DRA_SERVICE}/NodeUnprepareResources",
+            request_serializer=lambda m: m.encode(),
+            response_deserializer=dapi.NodeUnprepareResourcesResponse
+            .decode)
+        resp = unprep(dapi.NodeUnprepareResourcesRequest(claims=[
+            dapi.Claim(namespace="default", uid="uid-1", name="c")]),
+            timeout=10)
+        assert resp.claims[0].value.error == ""
+        assert state.prepared_claims() == []
+
+        # registration service answers GetInfo
+        reg_sock = server.reg_socket
+        ch2 = _grpc.insecure_channel(f"unix://{reg_sock}")
+        info = ch2.unary_unary(
+            f"/{dapi.REGISTRATION_SERVICE}/GetInfo",
+            request_serializer=lambda m: m.encode(),
+            response_deserializer=dapi.PluginInfo.decode)(
+            dapi.InfoRequest(), timeout=10)
+        assert info.type == dapi.PLUGIN_TYPE_DRA
+        assert info.endpoint == endpoint
+        ch.close(); ch2.close()
+    finally:
+        server.stop()
+
+
+def test_dra_driver_publishes_slices(state):
+    from vgpu_manager_amd.client.kube import FakeKubeClient
+    from vgpu_manager_amd.dra.driver import DraDriver
+    client = FakeKubeClient()
+    d = DraDriver(state, client, endpoint="/nonexistent")
+    rs = d.publish_resource_slices()
+    assert rs["metadata"]["name"] in client.resource_slices
+    assert len(rs["spec"]["devices"]) == 2

@@ -68,6 +68,13 @@ class KubeClient:
     def list_pdbs(self, namespace: Optional[str] = None) -> List[dict]:
         return []
 
+    # resource.k8s.io (DRA driver surface)
+    def get_resource_claim(self, namespace: str, name: str) -> dict:
+        raise KubeError("resource claims unsupported")
+
+    def apply_resource_slice(self, rs: dict) -> None:
+        raise KubeError("resource slices unsupported")
+
     # coordination.k8s.io Leases (scheduler leader election,
     # reference cmd/device-scheduler lease.go)
     def get_lease(self, namespace: str, name: str) -> dict:
@@ -99,6 +106,8 @@ class FakeKubeClient(KubeClient):
         self.events: List[dict] = []
         self.pdbs: List[dict] = []
         self.leases: Dict[tuple, dict] = {}
+        self.resource_claims: Dict[tuple, dict] = {}
+        self.resource_slices: Dict[str, dict] = {}
 
     # -- test helpers --
     def add_node(self, node: dict) -> None:
@@ -188,6 +197,25 @@ class FakeKubeClient(KubeClient):
     def add_pdb(self, pdb: dict) -> None:
         with self._mu:
             self.pdbs.append(pdb)
+
+    def add_resource_claim(self, claim: dict) -> None:
+        with self._mu:
+            meta = claim["metadata"]
+            self.resource_claims[(meta.get("namespace", "default"),
+                                  meta["name"])] = claim
+
+    def get_resource_claim(self, namespace, name):
+        with self._mu:
+            c = self.resource_claims.get((namespace, name))
+            if c is None:
+                raise KubeError(
+                    f"resourceclaim {namespace}/{name} not found")
+            return json.loads(json.dumps(c))
+
+    def apply_resource_slice(self, rs):
+        with self._mu:
+            self.resource_slices[rs["metadata"]["name"]] = \
+                json.loads(json.dumps(rs))
 
     def list_pdbs(self, namespace=None):
         with self._mu:
@@ -337,6 +365,23 @@ class RestKubeClient(KubeClient):
                 "/poddisruptionbudgets" if namespace else
                 "/apis/policy/v1/poddisruptionbudgets")
         return self._req("GET", path).get("items", [])
+
+    def get_resource_claim(self, namespace, name):
+        return self._req(
+            "GET",
+            f"/apis/resource.k8s.io/v1beta1/namespaces/{namespace}"
+            f"/resourceclaims/{name}")
+
+    def apply_resource_slice(self, rs):
+        name = rs["metadata"]["name"]
+        try:
+            self._req("PUT",
+                      f"/apis/resource.k8s.io/v1beta1/resourceslices/"
+                      f"{name}", rs)
+        except KubeError:
+            self._req("POST",
+                      "/apis/resource.k8s.io/v1beta1/resourceslices",
+                      rs)
 
     def get_lease(self, namespace, name):
         return self._req(

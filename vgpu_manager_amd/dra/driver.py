@@ -1,0 +1,163 @@
+"""DRA kubelet-plugin driver (reference pkg/kubeletplugin/driver.go +
+cmd/kubelet-plugin): the gRPC servers kubelet talks to, the claim
+resolution path, and ResourceSlice publishing.
+
+Two unix sockets, as kubelet's plugin watcher expects:
+  <plugins_registry>/<driver>.sock  — Registration (GetInfo)
+  <plugins_dir>/<driver>/dra.sock   — DRAPlugin (Prepare/Unprepare)
+
+NodePrepareResources fetches each ResourceClaim from the apiserver,
+decodes the allocation result (dra/resolve.py) and hands it to
+DeviceState.prepare; errors are returned per-claim, never as gRPC
+failures (reference driver.go:446-520 contract).
+"""
+from __future__ import annotations
+
+import logging
+import os
+import threading
+from concurrent import futures
+from typing import Optional
+
+import grpc
+
+from ..client.kube import KubeClient, KubeError
+from .state import DRA_DRIVER_NAME, DeviceState, build_resource_slice
+from . import api
+from .resolve import resolve_claim
+
+log = logging.getLogger("vgpu.dra.driver")
+
+
+class DraDriver:
+    """The gRPC servicer for both services."""
+
+    def __init__(self, state: DeviceState, client: KubeClient, *,
+                 endpoint: str):
+        self.state = state
+        self.client = client
+        self.endpoint = endpoint
+        self.registered = threading.Event()
+
+    # ---- v1beta1.DRAPlugin ----
+    def NodePrepareResources(self, request, context):
+        resp = api.NodePrepareResourcesResponse()
+        for claim_ref in request.claims:
+            entry = api._PrepareEntry(key=claim_ref.uid)
+            try:
+                claim = self.client.get_resource_claim(
+                    claim_ref.namespace, claim_ref.name)
+                if claim.get("metadata", {}).get("uid") != claim_ref.uid:
+                    raise KubeError(
+                        f"claim uid mismatch for {claim_ref.name}")
+                params, sharing = resolve_claim(claim)
+                if not params:
+                    raise ValueError("no devices for driver "
+                                     f"{DRA_DRIVER_NAME} in claim")
+                pod_meta = {"uid": claim_ref.uid,
+                            "name": claim_ref.name,
+                            "namespace": claim_ref.namespace}
+                prepared = self.state.prepare(
+                    claim_ref.uid, params, pod_meta=pod_meta,
+                    sharing_config=sharing)
+                value = api.NodePrepareResourceResponse()
+                for p in params:
+                    value.devices.append(api.Device(
+                        request_names=[],
+                        pool_name=self.state.node_name,
+                        device_name=p.uuid,
+                        cdi_device_ids=prepared.cdi_device_ids))
+                entry.value = value
+            except (KubeError, ValueError, OSError) as e:
+                log.warning("prepare %s failed: %s", claim_ref.uid, e)
+                entry.value = api.NodePrepareResourceResponse(
+                    error=str(e))
+            resp.claims.append(entry)
+        return resp
+
+    def NodeUnprepareResources(self, request, context):
+        resp = api.NodeUnprepareResourcesResponse()
+        for claim_ref in request.claims:
+            entry = api._UnprepareEntry(
+                key=claim_ref.uid,
+                value=api.NodeUnprepareResourceResponse())
+            try:
+                self.state.unprepare(claim_ref.uid)
+            except OSError as e:
+                entry.value.error = str(e)
+            resp.claims.append(entry)
+        return resp
+
+    # ---- pluginregistration.Registration ----
+    def GetInfo(self, request, context):
+        return api.PluginInfo(
+            type=api.PLUGIN_TYPE_DRA, name=DRA_DRIVER_NAME,
+            endpoint=self.endpoint,
+            supported_versions=["v1beta1"])
+
+    def NotifyRegistrationStatus(self, request, context):
+        if request.plugin_registered:
+            log.info("kubelet registered driver %s", DRA_DRIVER_NAME)
+            self.registered.set()
+        else:
+            log.error("kubelet registration failed: %s", request.error)
+        return api.RegistrationStatusResponse()
+
+    # ---- ResourceSlice publishing (driver.go:276-397) ----
+    def publish_resource_slices(self, *, consumable_shares=False,
+                                cpx=False) -> dict:
+        rs = build_resource_slice(
+            self.state.node_name,
+            list(self.state.devices.values()),
+            consumable_shares=consumable_shares, cpx=cpx)
+        try:
+            self.client.apply_resource_slice(rs)
+        except KubeError as e:
+            log.warning("resource slice publish failed: %s", e)
+        return rs
+
+
+class DraDriverServer:
+    """Owns the two unix-socket gRPC servers."""
+
+    def __init__(self, driver: DraDriver, *,
+                 plugins_dir: str = api.PLUGINS_DIR,
+                 plugins_registry: str = api.PLUGINS_REGISTRY):
+        self.driver = driver
+        self.dra_socket = driver.endpoint
+        self.reg_socket = os.path.join(plugins_registry,
+                                       f"{DRA_DRIVER_NAME}.sock")
+        self._servers: list = []
+        self._plugins_dir = plugins_dir
+        self._plugins_registry = plugins_registry
+
+    def start(self) -> None:
+        os.makedirs(os.path.dirname(self.dra_socket), exist_ok=True)
+        os.makedirs(self._plugins_registry, exist_ok=True)
+        for path, handler in (
+                (self.dra_socket, api.dra_plugin_handler(self.driver)),
+                (self.reg_socket,
+                 api.registration_handler(self.driver))):
+            if os.path.exists(path):
+                os.unlink(path)
+            s = grpc.server(futures.ThreadPoolExecutor(max_workers=4))
+            s.add_generic_rpc_handlers((handler,))
+            s.add_insecure_port(f"unix://{path}")
+            s.start()
+            self._servers.append(s)
+        log.info("DRA driver serving on %s (registry %s)",
+                 self.dra_socket, self.reg_socket)
+
+    def stop(self) -> None:
+        for s in self._servers:
+            s.stop(grace=1.0)
+        self._servers.clear()
+        for path in (self.dra_socket, self.reg_socket):
+            try:
+                os.unlink(path)
+            except OSError:
+                pass
+
+
+def default_endpoint(plugins_dir: str = api.PLUGINS_DIR) -> str:
+    return os.path.join(plugins_dir, DRA_DRIVER_NAME, "dra.sock")

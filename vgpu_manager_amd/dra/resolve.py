@@ -1,0 +1,77 @@
+"""ResourceClaim allocation-result decoding (reference
+pkg/claimresolve/allocated_vgpu.go): turn a claim object's
+status.allocation back into the driver's `VgpuClaimParams` view plus
+the opaque sharing config, shared by the kubelet plugin, webhook and
+monitor.
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+from .state import DRA_DRIVER_NAME, VgpuClaimParams
+
+
+def claim_uid(claim: dict) -> str:
+    return (claim.get("metadata", {}) or {}).get("uid", "")
+
+
+def resolve_claim(claim: dict
+                  ) -> Tuple[List[VgpuClaimParams], Optional[dict]]:
+    """Returns ([params...], sharing_config|None).
+
+    Only results owned by OUR driver are resolved; a claim with mixed
+    drivers yields only our share of it.  The opaque config parameters
+    recognized: cores, memoryMiB, partitionKey (per-request via the
+    `requests` selector list), strategy/... (claim-wide sharing).
+    """
+    alloc = ((claim.get("status", {}) or {})
+             .get("allocation", {}) or {})
+    devices = alloc.get("devices", {}) or {}
+    results = devices.get("results") or []
+    configs = devices.get("config") or []
+
+    # request name -> opaque parameters (ours only)
+    per_request: dict = {}
+    claim_wide: dict = {}
+    sharing: Optional[dict] = None
+    for cfg in configs:
+        opaque = cfg.get("opaque") or {}
+        if opaque.get("driver") not in (None, "", DRA_DRIVER_NAME):
+            continue
+        params = opaque.get("parameters") or {}
+        if "strategy" in params:
+            sharing = params
+        reqs = cfg.get("requests") or []
+        if reqs:
+            for r in reqs:
+                per_request.setdefault(r, {}).update(params)
+        else:
+            claim_wide.update(params)
+
+    out: List[VgpuClaimParams] = []
+    for res in results:
+        if res.get("driver", DRA_DRIVER_NAME) != DRA_DRIVER_NAME:
+            continue
+        device = res.get("device", "")
+        if not device:
+            continue
+        params = dict(claim_wide)
+        params.update(per_request.get(res.get("request", ""), {}))
+        uuid = device
+        cpx: List[int] = []
+        # CPX partition device names are "<uuid>-cpx-<p>" (state.py
+        # build_resource_slice); map back to the parent GPU
+        if "-cpx-" in device:
+            uuid, _, p = device.rpartition("-cpx-")
+            try:
+                cpx = [int(p)]
+            except ValueError:
+                uuid = device
+        out.append(VgpuClaimParams(
+            uuid=uuid,
+            cores=int(params.get("cores", 0) or 0),
+            memory_mib=int(params.get("memoryMiB", 0) or 0),
+            partition_key=str(params.get("partitionKey", "default")),
+            cpx_partitions=cpx,
+        ))
+    return out, sharing
