@@ -69,3 +69,28 @@ def test_magics_match_header():
     assert abi.VGPU_VMEM_MAGIC == c_macro("VGPU_VMEM_MAGIC")
     assert abi.VGPU_SMND_MAGIC == c_macro("VGPU_SMND_MAGIC")
     assert abi.VGPU_PIDS_MAGIC == c_macro("VGPU_PIDS_MAGIC")
+
+
+def test_shim_exports_exactly_match_exports_map():
+    """hack/-style CI guard (reference hack/check_exported_symbols):
+    the shim must export ONLY dlsym + the hook surface; one stray
+    global symbol can shadow an unrelated library in every process
+    of the container."""
+    import re
+    import subprocess
+
+    from tests.conftest import LIB_DIR
+    so = os.path.join(LIB_DIR, "build", "libvgpu-control.so")
+    if not os.path.exists(so):
+        pytest.skip("library not built")
+    want = set()
+    for line in open(os.path.join(LIB_DIR, "exports.map")):
+        m = re.match(r"\s+([A-Za-z_][A-Za-z0-9_]*);", line)
+        if m:
+            want.add(m.group(1))
+    out = subprocess.run(["nm", "-D", "--defined-only", so],
+                         capture_output=True, text=True,
+                         check=True).stdout
+    got = {line.split()[-1] for line in out.splitlines() if line.strip()}
+    assert got == want, (f"exported-symbol drift: extra={got - want} "
+                         f"missing={want - got}")

@@ -107,3 +107,27 @@ def test_sampler_publishes_region(tmp_path):
     assert region.data.heartbeat_ns > 0
     region.close()
     sampler.stop()
+
+
+def test_dra_claim_collector(tmp_path):
+    from vgpu_manager_amd.device.types import fake_device
+    from vgpu_manager_amd.dra.state import DeviceState, VgpuClaimParams
+    from vgpu_manager_amd.monitor.collector import DraClaimCollector
+
+    cp = str(tmp_path / "checkpoint.json")
+    state = DeviceState("node-a", [fake_device(0)],
+                        claims_dir=str(tmp_path / "claims"),
+                        checkpoint_path=cp)
+    state.prepare("uid-9", [VgpuClaimParams(
+        uuid="GPU-fake-0000", cores=30, memory_mib=2048)])
+    fams = {f.name: f for f in
+            DraClaimCollector(cp, "node-a").collect()}
+    assert fams["dra_vgpu_claim_prepared"].samples[0].labels[
+        "claim_uid"] == "uid-9"
+    assert fams["dra_vgpu_claim_devices"].samples[0].value == 1
+    mem = fams["dra_vgpu_claim_memory_limit_mib"].samples[0]
+    assert mem.value == 2048 and mem.labels["uuid"] == "GPU-fake-0000"
+    assert fams["dra_vgpu_claim_core_limit"].samples[0].value == 30
+    # missing checkpoint -> no families, no crash
+    assert list(DraClaimCollector(
+        str(tmp_path / "nope.json"), "n").collect()) == []

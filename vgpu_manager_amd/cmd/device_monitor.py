@@ -23,6 +23,10 @@ def main(argv=None):
     ap.add_argument("--base-dir", default=consts.MANAGER_DIR)
     ap.add_argument("--shared-watcher", action="store_true",
                     help="run the shared utilization sampler too")
+    ap.add_argument("--dra-checkpoint",
+                    default="/var/lib/vgpu-manager/checkpoint.json",
+                    help="DRA driver checkpoint to export claim "
+                         "metrics from (if present)")
     args = ap.parse_args(argv)
     logging.basicConfig(level=logging.INFO)
 
@@ -35,6 +39,9 @@ def main(argv=None):
         manager = FakeDeviceManager(args.node_name, n_devices=0)
     REGISTRY.register(NodeVgpuCollector(manager, lister))
     REGISTRY.register(PhysicalGpuCollector())
+    from ..monitor.collector import DraClaimCollector
+    REGISTRY.register(DraClaimCollector(args.dra_checkpoint,
+                                        args.node_name))
 
     if args.shared_watcher:
         from ..monitor.sampler import AmdSmiSource, UtilSampler

@@ -137,3 +137,47 @@ class PhysicalGpuCollector(Collector):
             except Exception:
                 pass
         yield from (util, vram, vram_total)
+
+
+class DraClaimCollector(Collector):
+    """dra_vgpu_claim_*: prepared-claim view from the DRA driver's
+    checkpoint (reference collector/dra_gpu.go reads the same state
+    through the kubelet plugin)."""
+
+    def __init__(self, checkpoint_path: str, node_name: str = ""):
+        self.checkpoint_path = checkpoint_path
+        self.node_name = node_name
+
+    def collect(self):
+        import json as _json
+        prepared = GaugeMetricFamily(
+            "dra_vgpu_claim_prepared",
+            "1 per prepared DRA claim",
+            labels=["node", "claim_uid"])
+        devices = GaugeMetricFamily(
+            "dra_vgpu_claim_devices",
+            "Devices prepared for the claim",
+            labels=["node", "claim_uid"])
+        memory = GaugeMetricFamily(
+            "dra_vgpu_claim_memory_limit_mib",
+            "Per-device memory limit of the claim",
+            labels=["node", "claim_uid", "uuid", "partition"])
+        cores = GaugeMetricFamily(
+            "dra_vgpu_claim_core_limit",
+            "Per-device CU limit % of the claim",
+            labels=["node", "claim_uid", "uuid", "partition"])
+        try:
+            data = _json.load(open(self.checkpoint_path))
+            claims = data.get("payload", {}).get("claims", {})
+        except (OSError, ValueError):
+            return
+        for uid, entry in claims.items():
+            prepared.add_metric([self.node_name, uid], 1)
+            params = entry.get("params", [])
+            devices.add_metric([self.node_name, uid], len(params))
+            for p in params:
+                lbl = [self.node_name, uid, p.get("uuid", ""),
+                       p.get("partition_key", "default")]
+                memory.add_metric(lbl, p.get("memory_mib", 0))
+                cores.add_metric(lbl, p.get("cores", 0))
+        yield from (prepared, devices, memory, cores)
