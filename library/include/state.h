@@ -169,6 +169,9 @@ uint64_t alloc_registry_total(int dev);
 int  vmem_ledger_add(int dev, uint64_t dptr, uint64_t size, int kind);
 void vmem_ledger_remove(int idx);
 uint64_t vmem_ledger_used(int dev);
+void vmem_ledger_cleanup_self(void);  /* atexit: retire own charges   */
+int  vmem_ledger_sweep_dead(void);    /* reclaim dead-pid records     */
+void alloc_registry_clear(void);      /* fork child: not the owner    */
 void dev_hooked_add(int dev, int64_t delta);
 uint64_t dev_hooked_used(int dev);
 
@@ -177,6 +180,9 @@ uint64_t vgpu_account_used(int dev);
 
 /* atfork child handler hook-side reset (hip_hook.c) */
 void vgpu_hook_fork_child(void);
+
+/* register the atexit teardown (idempotent; hip_hook.c) */
+void vgpu_register_fini_atexit(void);
 
 /* amd-smi sampling (watcher side; dlopens libamd_smi lazily) */
 bool smi_available(void);

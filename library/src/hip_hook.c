@@ -361,10 +361,15 @@ static void hook_fini(void);
 /* atexit ordering: exit() runs atexit handlers BEFORE ELF destructors,
  * and a preloaded library's own destructor runs after everyone else's
  * — too late to stop the watcher from touching a finalized amd-smi.  */
-void vgpu_register_fini_atexit(void) { atexit(hook_fini); }
+void vgpu_register_fini_atexit(void) {
+    static int registered;
+    if (__atomic_exchange_n(&registered, 1, __ATOMIC_ACQ_REL) == 0)
+        atexit(hook_fini);
+}
 
 static void hook_fini(void) {
     __atomic_store_n(&g_shutdown, 1, __ATOMIC_RELEASE);
+    vmem_ledger_cleanup_self(); /* retire our shared-region charges    */
     if (!__atomic_load_n(&g_watcher_state, __ATOMIC_ACQUIRE))
         return; /* no watcher in THIS process (e.g. fork child idle)   */
     for (int i = 0; i < 50; i++) { /* <=500ms grace                    */
@@ -378,9 +383,12 @@ static void *watcher_main(void *arg) {
     (void)arg;
     /* absolute-time cadence: drift-free 100ms grid, overrun floor     */
     uint64_t next = mono_ns();
+    uint32_t cycle = 0;
     for (;;) {
         if (__atomic_load_n(&g_shutdown, __ATOMIC_ACQUIRE)) break;
         next += (uint64_t)WATCHER_CYCLE_MS * 1000000ull;
+        /* ~every 3.2s: reclaim spill records of SIGKILL'd siblings   */
+        if ((++cycle & 31u) == 0) vmem_ledger_sweep_dead();
         for (int dev = 0; dev < g_state.device_count; dev++) {
             if (cfg_dev(dev) < 0) continue;
             uint32_t flags = vgpu_device_flags(dev);

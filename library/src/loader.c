@@ -23,6 +23,7 @@
 #include <dlfcn.h>
 #include <errno.h>
 #include <pthread.h>
+#include <signal.h>
 #include <stdio.h>
 #include <stdlib.h>
 #include <string.h>
@@ -265,6 +266,7 @@ static void fork_child_handler(void) {
         pthread_mutex_init(&g_state.dev[i].gap_mu, NULL);
     }
     vgpu_hook_fork_child(); /* re-arm watcher start + reseed pools     */
+    alloc_registry_clear(); /* parent owns those allocations, not us   */
     vgpu_load_pid_set(&g_state.pids);
 }
 
@@ -368,6 +370,7 @@ static void do_init(void) {
         pthread_mutex_init(&g_state.dev[i].gap_mu, NULL);
 
     pthread_atfork(NULL, NULL, fork_child_handler);
+    vgpu_register_fini_atexit(); /* exit cleanup even without watcher */
 
     g_state.initialized = 1;
     g_init_rc = 0;
@@ -425,7 +428,11 @@ int alloc_registry_add(void *ptr, size_t size, int kind, int dev,
             g_reg[i].dev = dev;
             g_reg[i].vmem_idx = vmem_idx;
             g_reg[i].host_ptr = (uint64_t)(uintptr_t)host_ptr;
-            if (dev >= 0 && dev < MAX_DEVICE_COUNT)
+            /* the per-device total mirrors dev_hooked_used charges:
+             * only DEVICE/ASYNC kinds charged it (managed/hostspill
+             * bytes live in the vmem ledger, retired per record)     */
+            if (dev >= 0 && dev < MAX_DEVICE_COUNT &&
+                (kind == ALLOC_KIND_DEVICE || kind == ALLOC_KIND_ASYNC))
                 g_reg_dev_total[dev] += size;
             pthread_mutex_unlock(&g_reg_mu);
             return (int)i;
@@ -451,7 +458,9 @@ bool alloc_registry_remove(void *ptr, size_t *size, int *kind, int *dev,
             if (vmem_idx) *vmem_idx = g_reg[i].vmem_idx;
             if (host_ptr)
                 *host_ptr = (void *)(uintptr_t)g_reg[i].host_ptr;
-            if (g_reg[i].dev >= 0 && g_reg[i].dev < MAX_DEVICE_COUNT)
+            if (g_reg[i].dev >= 0 && g_reg[i].dev < MAX_DEVICE_COUNT &&
+                (g_reg[i].kind == ALLOC_KIND_DEVICE ||
+                 g_reg[i].kind == ALLOC_KIND_ASYNC))
                 g_reg_dev_total[g_reg[i].dev] -= g_reg[i].size;
             g_reg[i].ptr = 1; /* tombstone */
             pthread_mutex_unlock(&g_reg_mu);
@@ -468,6 +477,14 @@ uint64_t alloc_registry_total(int dev) {
     uint64_t t = g_reg_dev_total[dev];
     pthread_mutex_unlock(&g_reg_mu);
     return t;
+}
+
+/* fork child: the registry copy describes the PARENT's allocations —
+ * the child must not retire them at its own exit (double subtract).  */
+void alloc_registry_clear(void) {
+    pthread_mutex_init(&g_reg_mu, NULL);
+    memset(g_reg, 0, sizeof(g_reg));
+    memset(g_reg_dev_total, 0, sizeof(g_reg_dev_total));
 }
 
 /* ------------------------------------------------------------------ */
@@ -493,6 +510,9 @@ int vmem_ledger_add(int dev, uint64_t dptr, uint64_t size, int kind) {
             return (int)i;
         }
     }
+    /* full: reclaim slots owned by dead siblings, retry once          */
+    if (vmem_ledger_sweep_dead() > 0)
+        return vmem_ledger_add(dev, dptr, size, kind);
     LOGGER(LOG_WARN, "vmem ledger full");
     return -1;
 }
@@ -514,6 +534,78 @@ uint64_t vmem_ledger_used(int dev) {
     if (dev < 0 || dev >= MAX_DEVICE_COUNT) return 0;
     return __atomic_load_n(&g_state.vmem->counters[dev].vmem_used,
                            __ATOMIC_ACQUIRE);
+}
+
+/* Retire one LIVE record if (and only if) its pid matches `pid`.
+ * The BUSY claim serializes against a concurrent remove/sweep.       */
+static bool vmem_retire_if_pid(vmem_record_t *rec, vmem_region_t *r,
+                               int32_t pid) {
+    uint32_t st = VMEM_STATE_LIVE;
+    if (!__atomic_compare_exchange_n(&rec->state, &st, VMEM_STATE_BUSY,
+                                     true, __ATOMIC_ACQ_REL,
+                                     __ATOMIC_RELAXED))
+        return false;
+    if (rec->pid != pid) { /* raced: someone reused the slot          */
+        __atomic_store_n(&rec->state, VMEM_STATE_LIVE, __ATOMIC_RELEASE);
+        return false;
+    }
+    __atomic_fetch_sub(&r->counters[rec->device].vmem_used, rec->size,
+                       __ATOMIC_ACQ_REL);
+    __atomic_store_n(&rec->state, VMEM_STATE_FREE, __ATOMIC_RELEASE);
+    return true;
+}
+
+/* Exit cleanup (reference loader.c:2270-2364 exit/signal handlers):
+ * retire THIS process's spill records and un-charge its hooked device
+ * bytes from the shared counters — the HIP runtime frees the actual
+ * memory at process teardown, so leaving the charges would shrink
+ * every sibling's headroom forever.                                   */
+void vmem_ledger_cleanup_self(void) {
+    /* idempotent: hook_fini can be registered twice (init + watcher)
+     * and must not double-subtract the hooked totals                  */
+    static int done_pid;
+    if (!g_state.vmem_shared) return;
+    int32_t me = (int32_t)getpid();
+    if (__atomic_exchange_n(&done_pid, me, __ATOMIC_ACQ_REL) == me)
+        return; /* already ran in this process                        */
+    vmem_region_t *r = g_state.vmem;
+    int retired = 0;
+    for (uint32_t i = 0; i < MAX_VMEM_RECORDS; i++)
+        if (r->records[i].pid == me &&
+            vmem_retire_if_pid(&r->records[i], r, me))
+            retired++;
+    for (int dev = 0; dev < MAX_DEVICE_COUNT; dev++) {
+        uint64_t mine = alloc_registry_total(dev);
+        if (mine)
+            __atomic_fetch_sub(&r->counters[dev].dev_hooked_used, mine,
+                               __ATOMIC_ACQ_REL);
+    }
+    if (retired)
+        LOGGER(LOG_INFO, "exit cleanup retired %d vmem records", retired);
+}
+
+/* Dead-owner sweep: a SIGKILL'd process never runs the exit cleanup;
+ * its records are reclaimed by any sibling once the pid is gone.
+ * EPERM counts as alive (different-uid sibling); only ESRCH reclaims. */
+int vmem_ledger_sweep_dead(void) {
+    if (!g_state.vmem_shared) return 0;
+    vmem_region_t *r = g_state.vmem;
+    int swept = 0;
+    for (uint32_t i = 0; i < MAX_VMEM_RECORDS; i++) {
+        vmem_record_t *rec = &r->records[i];
+        if (__atomic_load_n(&rec->state, __ATOMIC_ACQUIRE) !=
+            VMEM_STATE_LIVE)
+            continue;
+        int32_t pid = rec->pid;
+        if (pid <= 0) continue;
+        if (kill((pid_t)pid, 0) == -1 && errno == ESRCH &&
+            vmem_retire_if_pid(rec, r, pid))
+            swept++;
+    }
+    if (swept)
+        LOGGER(LOG_WARN, "swept %d vmem records of dead processes",
+               swept);
+    return swept;
 }
 
 void dev_hooked_add(int dev, int64_t delta) {

@@ -161,6 +161,70 @@ static int scenario_fork(void) {
     return 0;
 }
 
+static int scenario_cleanup(void) {
+    /* env: VGPU_MEM_LIMIT_0=1m VGPU_MEM_OVERSOLD=1 VGPU_MEM_ACCOUNT_
+     * MODE=ledger VGPU_VMEM_PATH_OVERRIDE=<shared tmp file>.
+     *
+     * 1. A child that allocates past quota and EXITS NORMALLY must
+     *    leave no charge behind (atexit cleanup retires its device
+     *    bytes AND its spill records).
+     * 2. A child killed before atexit leaks its spill record; the
+     *    ledger-full sweep reclaims it once the pid is gone.         */
+    size_t freeb = 0, total = 0;
+    CHECK(hipMemGetInfo(&freeb, &total) == hipSuccess);
+    CHECK(total == 1024 * 1024 && freeb == total);
+
+    pid_t pid = fork();
+    CHECK(pid >= 0);
+    if (pid == 0) {
+        void *a = NULL, *b = NULL;
+        if (hipMalloc(&a, 900 * 1024) != hipSuccess) _exit(1);
+        if (hipMalloc(&b, 500 * 1024) != hipSuccess) _exit(1); /*spill*/
+        exit(0); /* NORMAL exit: atexit cleanup must retire charges   */
+    }
+    int st = 0;
+    CHECK(waitpid(pid, &st, 0) == pid);
+    CHECK(WIFEXITED(st) && WEXITSTATUS(st) == 0);
+    CHECK(hipMemGetInfo(&freeb, &total) == hipSuccess);
+    CHECK(freeb == total); /* nothing left behind                     */
+
+    /* 2: die WITHOUT cleanup (skip atexit via _exit)                 */
+    pid = fork();
+    CHECK(pid >= 0);
+    if (pid == 0) {
+        void *b = NULL;
+        void *a = NULL;
+        if (hipMalloc(&a, 900 * 1024) != hipSuccess) _exit(1);
+        if (hipMalloc(&b, 500 * 1024) != hipSuccess) _exit(1); /*spill*/
+        _exit(0); /* no atexit: leaks 900K hooked + 500K spill record */
+    }
+    CHECK(waitpid(pid, &st, 0) == pid);
+    CHECK(WIFEXITED(st) && WEXITSTATUS(st) == 0);
+    CHECK(hipMemGetInfo(&freeb, &total) == hipSuccess);
+    CHECK(freeb == 0); /* the leak is visible (1.4M charged > 1M)     */
+
+    /* fill the ledger: the add path must sweep the dead child's
+     * record instead of reporting ledger-full.  4095 slots remain;
+     * allocate 4K spills until the add would need the swept slot.    */
+    enum { FILL = 4096 };
+    static void *ptrs[FILL];
+    int got = 0;
+    for (int i = 0; i < FILL; i++) {
+        /* everything spills (we are far past quota)                  */
+        if (hipMalloc(&ptrs[got], 4096) != hipSuccess) break;
+        got++;
+    }
+    CHECK(got == FILL); /* only possible if the dead record was swept */
+    for (int i = 0; i < got; i++) CHECK(hipFree(ptrs[i]) == hipSuccess);
+    /* after the sweep + frees only the child's un-sweepable 900K of
+     * hooked device bytes remain charged (ledger mode cannot
+     * attribute those; amd-smi account modes self-heal them)         */
+    CHECK(hipMemGetInfo(&freeb, &total) == hipSuccess);
+    CHECK(total - freeb == 900 * 1024);
+    printf("PASS cleanup\n");
+    return 0;
+}
+
 int main(int argc, char **argv) {
     if (argc < 2) {
         fprintf(stderr, "usage: %s quota|oversold|launch|nolimit\n", argv[0]);
@@ -172,5 +236,6 @@ int main(int argc, char **argv) {
     if (strcmp(argv[1], "throttle") == 0) return scenario_throttle();
     if (strcmp(argv[1], "nolimit") == 0) return scenario_nolimit();
     if (strcmp(argv[1], "fork") == 0) return scenario_fork();
+    if (strcmp(argv[1], "cleanup") == 0) return scenario_cleanup();
     return 2;
 }

@@ -41,6 +41,16 @@ def test_hook_scenario(built_library, scenario, env):
     run_scenario(scenario, env)
 
 
+def test_hook_cleanup_and_sweep(built_library, tmp_path):
+    # shared vmem region: normal exit retires charges; _exit leaks a
+    # spill record that the ledger-full sweep reclaims (dead pid)
+    run_scenario("cleanup", {
+        "VGPU_MEM_LIMIT_0": "1m", "VGPU_MEM_OVERSOLD": "1",
+        "VGPU_MEM_ACCOUNT_MODE": "ledger",
+        "VGPU_VMEM_PATH_OVERRIDE": str(tmp_path / "vmem_node.config"),
+    })
+
+
 def test_hook_disabled(built_library):
     # DISABLE_VGPU_CONTROL must make limits inert
     run_scenario("nolimit", {"VGPU_MEM_LIMIT_0": "1m",

@@ -78,8 +78,10 @@ class GpuFilter:
             request.preferred_numa = preferred_numa
 
         lock_key = "global-filter"
+        t_lock0 = time.monotonic()
         if self.locker and not dry_run:
             self.locker.acquire(lock_key)
+        t_work0 = time.monotonic()
         try:
             candidates: List[Tuple[float, str, list]] = []
             for name in node_names:
@@ -132,10 +134,15 @@ class GpuFilter:
                     return self._result([], failed, error=str(e))
 
             metrics.observe("filter", time.monotonic() - t0, True)
+            metrics.observe_placement(
+                request.topology_mode or "none",
+                "numa" if preferred_numa is not None else "fresh")
             return self._result([chosen], failed)
         finally:
             if self.locker and not dry_run:
                 self.locker.release(lock_key)
+            metrics.observe_split("filter", t_work0 - t_lock0,
+                                  time.monotonic() - t_work0)
 
     @staticmethod
     def _candidate_nodes(args: dict) -> List[str]:
