@@ -404,3 +404,56 @@ def test_dra_driver_publishes_slices(state):
     rs = d.publish_resource_slices()
     assert rs["metadata"]["name"] in client.resource_slices
     assert len(rs["spec"]["devices"]) == 2
+
+
+# ---- dynamic CPX partitioning (reference dynamic MIG, mig.go) ----
+
+def test_dynamic_cpx_lifecycle(tmp_path):
+    from vgpu_manager_amd.device.partition import (
+        FakePartitionBackend,
+        PartitionManager,
+    )
+    be = FakePartitionBackend(n_gpus=2)
+    pm = PartitionManager(be)
+    devices = [fake_device(i) for i in range(2)]
+    state = DeviceState("node-a", devices,
+                        claims_dir=str(tmp_path / "claims"),
+                        checkpoint_path=str(tmp_path / "cp.json"),
+                        partition_manager=pm)
+    # whole-GPU claim: no switch
+    state.prepare("c-whole", [VgpuClaimParams(uuid="GPU-fake-0001")])
+    assert be.modes[1] == "SPX"
+    # cpx claim flips GPU 0 to CPX
+    state.prepare("c-cpx-1", [VgpuClaimParams(
+        uuid="GPU-fake-0000", cpx_partitions=[0, 1])])
+    assert be.modes[0] == "CPX"
+    # second cpx claim shares the mode
+    state.prepare("c-cpx-2", [VgpuClaimParams(
+        uuid="GPU-fake-0000", cpx_partitions=[2], partition_key="k2")])
+    assert be.modes[0] == "CPX"
+    # first unprepare keeps CPX (holder remains)
+    state.unprepare("c-cpx-1")
+    assert be.modes[0] == "CPX"
+    # last unprepare reverts to SPX
+    state.unprepare("c-cpx-2")
+    assert be.modes[0] == "SPX"
+
+
+def test_dynamic_cpx_busy_gpu_fails_claim(tmp_path):
+    from vgpu_manager_amd.device.partition import (
+        FakePartitionBackend,
+        PartitionError,
+        PartitionManager,
+    )
+    be = FakePartitionBackend(n_gpus=1, busy={0: 3})
+    state = DeviceState("node-a", [fake_device(0)],
+                        claims_dir=str(tmp_path / "claims"),
+                        checkpoint_path=str(tmp_path / "cp.json"),
+                        partition_manager=PartitionManager(be))
+    with pytest.raises(PartitionError):
+        state.prepare("c-busy", [VgpuClaimParams(
+            uuid="GPU-fake-0000", cpx_partitions=[0])])
+    # nothing checkpointed, no mode change, no stuck holder
+    assert state.prepared_claims() == []
+    assert be.modes[0] == "SPX"
+    assert state.partition_manager.holders(0) == set()

@@ -49,9 +49,17 @@ def main(argv=None):
     client = RestKubeClient()
     manager = AmdDeviceManager(args.node_name, config)
 
+    pm = None
+    if gates.enabled("DynamicCPXPartitioning"):
+        from ..device.partition import (
+            AmdSmiPartitionBackend,
+            PartitionManager,
+        )
+        pm = PartitionManager(AmdSmiPartitionBackend())
     state = DeviceState(args.node_name, manager.devices,
                         claims_dir=args.claims_dir,
-                        checkpoint_path=args.checkpoint)
+                        checkpoint_path=args.checkpoint,
+                        partition_manager=pm)
     driver = DraDriver(state, client,
                        endpoint=default_endpoint(args.plugins_dir))
     server = DraDriverServer(driver, plugins_dir=args.plugins_dir,

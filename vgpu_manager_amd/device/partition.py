@@ -1,0 +1,165 @@
+"""Dynamic compute-partition management — the MI355X analog of the
+reference's dynamic MIG creation (pkg/kubeletplugin/mig.go +
+partitions.go).
+
+MI355X GPUs run in SPX (one monolithic device) or CPX (8 devices, one
+per XCD, 32 CUs each; memory split by the NPS mode). Switching is an
+amd-smi operation that requires the GPU to be idle. The DRA driver
+switches a GPU to CPX when a cpx-partition claim is prepared against
+an SPX GPU, and back when the last such claim is unprepared —
+mirroring the reference's create-MIG-on-Prepare / destroy-on-
+Unprepare lifecycle.
+
+Backends: `AmdSmiPartitionBackend` (real, via the amdsmi python
+binding or the `amd-smi` CLI) and `FakePartitionBackend` (tests).
+"""
+from __future__ import annotations
+
+import logging
+import subprocess
+from typing import Dict, List, Optional
+
+log = logging.getLogger("vgpu.device.partition")
+
+SPX = "SPX"
+CPX = "CPX"
+VALID_MODES = (SPX, CPX)
+
+CPX_PARTITIONS_PER_GPU = 8  # one per XCD
+
+
+class PartitionError(Exception):
+    pass
+
+
+class PartitionBackend:
+    def get_mode(self, gpu_index: int) -> str:
+        raise NotImplementedError
+
+    def set_mode(self, gpu_index: int, mode: str) -> None:
+        raise NotImplementedError
+
+    def busy_process_count(self, gpu_index: int) -> int:
+        """Processes with context on the GPU (mode switch needs 0)."""
+        raise NotImplementedError
+
+
+class FakePartitionBackend(PartitionBackend):
+    def __init__(self, n_gpus: int = 8, busy: Optional[Dict[int, int]]
+                 = None):
+        self.modes = {i: SPX for i in range(n_gpus)}
+        self.busy = dict(busy or {})
+        self.switches: List[tuple] = []
+
+    def get_mode(self, gpu_index):
+        return self.modes[gpu_index]
+
+    def set_mode(self, gpu_index, mode):
+        if mode not in VALID_MODES:
+            raise PartitionError(f"invalid mode {mode}")
+        self.modes[gpu_index] = mode
+        self.switches.append((gpu_index, mode))
+
+    def busy_process_count(self, gpu_index):
+        return self.busy.get(gpu_index, 0)
+
+
+class AmdSmiPartitionBackend(PartitionBackend):
+    """amdsmi python binding first, `amd-smi` CLI fallback."""
+
+    def __init__(self):
+        self._amdsmi = None
+        self._handles = None
+
+    def _ensure(self):
+        if self._handles is None:
+            import amdsmi
+            amdsmi.amdsmi_init()
+            self._amdsmi = amdsmi
+            self._handles = amdsmi.amdsmi_get_processor_handles()
+        return self._handles
+
+    def get_mode(self, gpu_index):
+        try:
+            h = self._ensure()[gpu_index]
+            mode = self._amdsmi.amdsmi_get_gpu_compute_partition(h)
+            return str(mode).upper().replace(
+                "AMDSMI_COMPUTE_PARTITION_TYPE_", "")
+        except Exception:
+            out = subprocess.run(
+                ["amd-smi", "partition", "--gpu", str(gpu_index)],
+                capture_output=True, text=True, timeout=30).stdout
+            for tok in (CPX, SPX):
+                if tok in out:
+                    return tok
+            raise PartitionError(
+                f"cannot read partition mode of GPU {gpu_index}")
+
+    def set_mode(self, gpu_index, mode):
+        if mode not in VALID_MODES:
+            raise PartitionError(f"invalid mode {mode}")
+        try:
+            h = self._ensure()[gpu_index]
+            self._amdsmi.amdsmi_set_gpu_compute_partition(h, mode)
+            return
+        except Exception as e:
+            r = subprocess.run(
+                ["amd-smi", "set", "--gpu", str(gpu_index),
+                 "--compute-partition", mode],
+                capture_output=True, text=True, timeout=60)
+            if r.returncode != 0:
+                raise PartitionError(
+                    f"set {mode} on GPU {gpu_index} failed: "
+                    f"{e}; CLI: {r.stderr[:200]}") from e
+
+    def busy_process_count(self, gpu_index):
+        try:
+            h = self._ensure()[gpu_index]
+            return len(self._amdsmi.amdsmi_get_gpu_process_list(h))
+        except Exception:
+            return 0
+
+
+class PartitionManager:
+    """Reference-counted CPX lifecycle per GPU."""
+
+    def __init__(self, backend: PartitionBackend):
+        self.backend = backend
+        self._cpx_claims: Dict[int, set] = {}
+
+    def ensure_cpx(self, gpu_index: int, claim_uid: str) -> None:
+        """Called during Prepare of a cpx claim. Switches SPX→CPX iff
+        the GPU is idle; a busy GPU fails the prepare (per-claim
+        error, like the reference's failed MIG create)."""
+        holders = self._cpx_claims.setdefault(gpu_index, set())
+        mode = self.backend.get_mode(gpu_index)
+        if mode != CPX:
+            busy = self.backend.busy_process_count(gpu_index)
+            if busy > 0:
+                raise PartitionError(
+                    f"GPU {gpu_index}: cannot switch to CPX with "
+                    f"{busy} active processes")
+            self.backend.set_mode(gpu_index, CPX)
+            log.info("GPU %d: SPX -> CPX for claim %s", gpu_index,
+                     claim_uid)
+        holders.add(claim_uid)
+
+    def release_cpx(self, gpu_index: int, claim_uid: str,
+                    revert: bool = True) -> None:
+        """Called during Unprepare. Reverts CPX→SPX when the last
+        claim leaves (best-effort: a busy GPU stays CPX)."""
+        holders = self._cpx_claims.get(gpu_index, set())
+        holders.discard(claim_uid)
+        if revert and not holders and \
+                self.backend.get_mode(gpu_index) == CPX and \
+                self.backend.busy_process_count(gpu_index) == 0:
+            try:
+                self.backend.set_mode(gpu_index, SPX)
+                log.info("GPU %d: CPX -> SPX (last claim gone)",
+                         gpu_index)
+            except PartitionError as e:
+                log.warning("GPU %d: revert to SPX failed: %s",
+                            gpu_index, e)
+
+    def holders(self, gpu_index: int) -> set:
+        return set(self._cpx_claims.get(gpu_index, set()))

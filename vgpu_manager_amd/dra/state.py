@@ -184,12 +184,16 @@ class DeviceState:
     def __init__(self, node_name: str, devices: List[DeviceInfo], *,
                  claims_dir: str, checkpoint_path: str,
                  driver_lib: str = "/usr/local/vgpu-manager/"
-                                   + consts.DRIVER_LIB_NAME):
+                                   + consts.DRIVER_LIB_NAME,
+                 partition_manager=None):
         self.node_name = node_name
         self.devices = {d.uuid: d for d in devices}
         self.claims_dir = claims_dir
         self.driver_lib = driver_lib
         self.checkpoint = Checkpoint(checkpoint_path)
+        # dynamic SPX<->CPX switching (device/partition.py); None =
+        # static partitioning (modes set by the operator)
+        self.partition_manager = partition_manager
         self._mu = threading.Lock()
 
     # ---- prepare ----
@@ -220,6 +224,24 @@ class DeviceState:
                 if p.uuid not in self.devices:
                     raise ValueError(f"unknown device {p.uuid}")
                 by_partition.setdefault(p.partition_key, []).append(p)
+
+            # dynamic CPX: claims pinning XCD partitions flip the GPU
+            # to CPX mode first (reference dynamic MIG create on
+            # Prepare, mig.go) — a busy GPU fails THIS claim only
+            cpx_gpus = []
+            if self.partition_manager is not None:
+                try:
+                    for p in params:
+                        if p.cpx_partitions:
+                            gpu = self.devices[p.uuid].id
+                            self.partition_manager.ensure_cpx(
+                                gpu, claim_uid)
+                            cpx_gpus.append(gpu)
+                except Exception:
+                    for gpu in cpx_gpus:  # roll back partial switch
+                        self.partition_manager.release_cpx(gpu,
+                                                           claim_uid)
+                    raise
 
             cdi_ids = []
             base = os.path.join(self.claims_dir, claim_uid)
@@ -265,6 +287,7 @@ class DeviceState:
                 "cdi_device_ids": sorted(set(cdi_ids)),
                 "container_dir": base,
                 "params": [vars(p) for p in params],
+                "cpx_gpus": sorted(set(cpx_gpus)),
             }
             self.checkpoint.save()
             return PreparedDevice(cdi_device_ids=sorted(set(cdi_ids)),
@@ -277,6 +300,9 @@ class DeviceState:
             self.checkpoint.save()
             if entry is None:
                 return False
+            if self.partition_manager is not None:
+                for gpu in entry.get("cpx_gpus", []):
+                    self.partition_manager.release_cpx(gpu, claim_uid)
             import shutil
             shutil.rmtree(entry["container_dir"], ignore_errors=True)
             return True
