@@ -205,3 +205,33 @@ print("OK")
     r = run_py(code, {"VGPU_CONFIG_PATH_OVERRIDE": p,
                       "VGPU_MEM_ACCOUNT_MODE": "ledger"})
     assert r.returncode == 0 and "OK" in r.stdout, r.stdout + r.stderr
+
+
+def test_exit_cleanup_retires_shared_charges(tmp_path):
+    """A process that spills past quota and exits normally must leave
+    the shared vmem region uncharged for the next process (exit
+    cleanup, loader.c vmem_ledger_cleanup_self)."""
+    vmem = str(tmp_path / "vmem_node.config")
+    env = {"VGPU_MEM_LIMIT_0": "1g", "VGPU_MEM_OVERSOLD": "1",
+           "VGPU_MEM_ACCOUNT_MODE": "ledger",
+           "VGPU_VMEM_PATH_OVERRIDE": vmem,
+           "HSA_XNACK": os.environ.get("HSA_XNACK", "0")}
+    # process A: 900M device + 500M spill, exit with everything live
+    code_a = WK_PRELUDE + """
+a = wk.wk_malloc(900 << 20); assert a
+b = wk.wk_malloc(500 << 20); assert b   # spills past the 1G quota
+print("A-OK")
+"""
+    r = run_py(code_a, env, timeout=300)
+    assert r.returncode == 0 and "A-OK" in r.stdout, r.stdout + r.stderr
+    # process B: the full quota must be available again
+    code_b = WK_PRELUDE + """
+free = wk.wk_mem_free(); total = wk.wk_mem_total()
+assert total == 1 << 30, total
+assert total - free < (64 << 20), f"leaked {total - free} bytes"
+p = wk.wk_malloc(800 << 20)
+assert p, "quota headroom not restored"
+print("B-OK")
+"""
+    r = run_py(code_b, env, timeout=300)
+    assert r.returncode == 0 and "B-OK" in r.stdout, r.stdout + r.stderr
