@@ -214,3 +214,68 @@ def test_quantity_plugin():
     resp = qp.Allocate(api.AllocateRequest(container_requests=[
         api.ContainerAllocateRequest(devices_ids=["x"])]), None)
     assert len(resp.container_responses) == 1
+
+
+def test_cpx_plugin_devices_and_allocate():
+    from vgpu_manager_amd.deviceplugin import api
+    from vgpu_manager_amd.deviceplugin.server import CpxPlugin
+    from vgpu_manager_amd.device.manager import FakeDeviceManager
+
+    mgr = FakeDeviceManager(n_devices=2)
+    mgr.devices[0].cpx = True  # GPU 0 in CPX mode, GPU 1 SPX
+    plugin = CpxPlugin(mgr)
+    devs = plugin._devices()
+    assert len(devs) == 8  # only the CPX GPU contributes partitions
+    assert devs[0].ID.endswith("-cpx-0")
+
+    req = api.AllocateRequest(container_requests=[
+        api.ContainerAllocateRequest(devices_ids=[
+            devs[2].ID, devs[3].ID])])
+    resp = plugin.Allocate(req, None)
+    cr = resp.container_responses[0]
+    assert cr.envs["VGPU_CPX_PARTITIONS_0"] == "2,3"
+    paths = {d.container_path for d in cr.devices}
+    assert "/dev/kfd" in paths
+    assert any("renderD" in p for p in paths)
+
+
+def test_plugin_set_includes_cpx_when_present(tmp_path):
+    from vgpu_manager_amd.client.kube import FakeKubeClient
+    from vgpu_manager_amd.deviceplugin.server import PluginSet
+    from vgpu_manager_amd.device.manager import FakeDeviceManager
+
+    mgr = FakeDeviceManager(n_devices=2)
+    ps = PluginSet(mgr, FakeKubeClient(),
+                   plugins_dir=str(tmp_path))
+    assert not any("cpx" in s.endpoint_name for s in ps.servers)
+    mgr.devices[1].cpx = True
+    ps2 = PluginSet(mgr, FakeKubeClient(),
+                    plugins_dir=str(tmp_path))
+    assert any("cpx" in s.endpoint_name for s in ps2.servers)
+
+
+def test_dra_health_republish():
+    from vgpu_manager_amd.client.kube import FakeKubeClient
+    from vgpu_manager_amd.device.manager import FakeDeviceManager
+    from vgpu_manager_amd.dra.driver import DraDriver
+    from vgpu_manager_amd.dra.state import DeviceState
+
+    mgr = FakeDeviceManager(n_devices=2)
+    import tempfile
+    with tempfile.TemporaryDirectory() as td:
+        state = DeviceState("n", mgr.devices,
+                            claims_dir=td + "/claims",
+                            checkpoint_path=td + "/cp.json")
+        client = FakeKubeClient()
+        d = DraDriver(state, client, endpoint="/x")
+        d.watch_health(mgr)
+        d.publish_resource_slices()
+        name = list(client.resource_slices)[0]
+        devs = client.resource_slices[name]["spec"]["devices"]
+        assert all(x["basic"]["attributes"]["healthy"]["bool"]
+                   for x in devs)
+        mgr.set_health(0, False)  # triggers republish via callback
+        devs = client.resource_slices[name]["spec"]["devices"]
+        healthy = [x["basic"]["attributes"]["healthy"]["bool"]
+                   for x in devs]
+        assert healthy.count(False) == 1

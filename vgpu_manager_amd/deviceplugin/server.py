@@ -122,6 +122,11 @@ class PluginSet:
                 QuantityPlugin(consts.vgpu_memory_resource(),
                                max(total_mib // 1024, 1)),
                 "amd-vgpu-memory.sock", plugins_dir))
+        # per-partition passthrough plugin for any CPX-mode GPUs
+        # (reference factory.go registers per-MIG-profile plugins)
+        if any(getattr(d, "cpx", False) for d in manager.devices):
+            self.servers.append(PluginServer(
+                CpxPlugin(manager), "amd-cpx.sock", plugins_dir))
 
     def start_all(self, kubelet_socket: str = api.KUBELET_SOCKET,
                   register: bool = True) -> None:
@@ -161,3 +166,77 @@ def watch_kubelet_restart(plugin_set: PluginSet,
     t = threading.Thread(target=run, daemon=True, name="kubelet-watch")
     t.start()
     return t
+
+
+class CpxPlugin:
+    """Per-partition passthrough plugin for CPX-mode GPUs (reference
+    mig_plugin.go: per-MIG-profile plugins; the MI355X analog is one
+    32-CU XCD partition).  Device IDs are `<uuid>-cpx-<p>`; Allocate
+    injects the render nodes + pins the container to its partitions
+    via env — no fractional limits (passthrough semantics)."""
+
+    CPX_PER_GPU = 8
+
+    def __init__(self, manager, resource_name=None):
+        self.manager = manager
+        self.resource_name = resource_name or \
+            consts.cpx_resource_prefix() + "32cu"
+        self._stopped = False
+
+    def _devices(self):
+        out = []
+        for d in self.manager.devices:
+            if not getattr(d, "cpx", False):
+                continue
+            for p in range(self.CPX_PER_GPU):
+                out.append(api.Device(
+                    ID=f"{d.uuid}-cpx-{p}",
+                    health=api.HEALTHY if d.healthy else api.UNHEALTHY,
+                    topology=api.TopologyInfo(
+                        nodes=[api.NUMANode(ID=d.numa)])))
+        return out
+
+    def GetDevicePluginOptions(self, request, context):
+        return api.DevicePluginOptions()
+
+    def ListAndWatch(self, request, context):
+        while not self._stopped:
+            yield api.ListAndWatchResponse(devices=self._devices())
+            time.sleep(60)
+
+    def GetPreferredAllocation(self, request, context):
+        return api.PreferredAllocationResponse()
+
+    def Allocate(self, request, context):
+        by_uuid = {d.uuid: d for d in self.manager.devices}
+        resp = api.AllocateResponse()
+        for creq in request.container_requests:
+            cresp = api.ContainerAllocateResponse()
+            parts, uuids = [], []
+            for did in creq.devices_ids:
+                uuid, _, p = did.rpartition("-cpx-")
+                parts.append(p)
+                if uuid not in uuids:
+                    uuids.append(uuid)
+            cresp.envs = {
+                "VGPU_CPX_PARTITIONS_0": ",".join(parts),
+                consts.ENV_VISIBLE_DEVICES: ",".join(uuids),
+            }
+            cresp.devices.append(api.DeviceSpec(
+                container_path="/dev/kfd", host_path="/dev/kfd",
+                permissions="rw"))
+            for uuid in uuids:
+                d = by_uuid.get(uuid)
+                render = 128 + (d.id if d else 0)
+                cresp.devices.append(api.DeviceSpec(
+                    container_path=f"/dev/dri/renderD{render}",
+                    host_path=f"/dev/dri/renderD{render}",
+                    permissions="rw"))
+            resp.container_responses.append(cresp)
+        return resp
+
+    def PreStartContainer(self, request, context):
+        return api.PreStartContainerResponse()
+
+    def stop(self):
+        self._stopped = True

@@ -116,6 +116,17 @@ class DraDriver:
             log.warning("resource slice publish failed: %s", e)
         return rs
 
+    def watch_health(self, manager, **publish_kwargs) -> None:
+        """Republish slices whenever the device manager flips a
+        device's health — the DRA analog of device taints (reference
+        device_health.go:476: unhealthy devices must leave the
+        allocatable inventory promptly)."""
+        def on_change(dev_info):
+            log.warning("device %s healthy=%s; republishing slices",
+                        dev_info.uuid, dev_info.healthy)
+            self.publish_resource_slices(**publish_kwargs)
+        manager.on_health_change(on_change)
+
 
 class DraDriverServer:
     """Owns the two unix-socket gRPC servers."""
