@@ -1,0 +1,137 @@
+"""End-to-end control-plane plumbing on CPU — BASELINE config #1
+("device-plugin + scheduler extender on CPU-only cluster with fake-GPU
+registry") without kind: every real component, fake kube + fake GPUs.
+
+The loop: webhook mutate → node registration → extender filter →
+bind → kubelet GetPreferredAllocation → Allocate → written ABI
+regions → PreStartContainer → the REAL C shim enforcing exactly the
+written config against the stub HIP runtime.
+"""
+import json
+import os
+
+import pytest
+
+from vgpu_manager_amd.client.kube import FakeKubeClient
+from vgpu_manager_amd.device.manager import FakeDeviceManager
+from vgpu_manager_amd.deviceplugin import api
+from vgpu_manager_amd.deviceplugin.vnum_plugin import (
+    VnumPlugin,
+    fake_id,
+)
+from vgpu_manager_amd.scheduler.bind import NodeBinder
+from vgpu_manager_amd.scheduler.filter import GpuFilter
+from vgpu_manager_amd.util import consts
+from vgpu_manager_amd.webhook.admission import (
+    apply_json_patch,
+    mutate_pod,
+    validate_pod,
+)
+
+from tests.test_allocator import make_pod
+from tests.test_hook_cpu import run_scenario
+
+
+def test_full_pod_lifecycle(tmp_path):
+    client = FakeKubeClient()
+    manager = FakeDeviceManager("gpu-node-1", n_devices=2)
+    client.add_node({"metadata": {"name": "gpu-node-1",
+                                  "annotations": {}}})
+    manager.register(client)
+    node = client.get_node("gpu-node-1")
+    assert consts.node_register_ann() in node["metadata"]["annotations"]
+
+    # 1. admission: mutate + validate (1 GPU, 50 cores, 1 MiB memory
+    #    — tiny quota so the C shim scenario can verify it below)
+    pod = make_pod(number=1, cores=50, memory=1, name="w1")
+    ok, msg = validate_pod(pod)
+    assert ok, msg
+    pod = apply_json_patch(pod, mutate_pod(pod))
+    assert pod["spec"]["schedulerName"] == "vgpu-scheduler"
+    client.add_pod(pod)
+
+    # 2. extender filter: places the pod, writes pre-allocation
+    res = GpuFilter(client).filter(
+        {"Pod": pod, "NodeNames": ["gpu-node-1"]})
+    assert res["Error"] == "" and res["NodeNames"] == ["gpu-node-1"]
+    pod = client.get_pod("default", "w1")
+    ann = pod["metadata"]["annotations"]
+    assert ann[consts.predicate_node_ann()] == "gpu-node-1"
+    pre = ann[consts.pre_alloc_ann()]
+    assert pre.startswith("main[")
+
+    # 3. extender bind
+    res = NodeBinder(client).bind({
+        "PodName": "w1", "PodNamespace": "default",
+        "Node": "gpu-node-1"})
+    assert not res.get("Error")
+    assert client.get_pod("default", "w1")["spec"]["nodeName"] == \
+        "gpu-node-1"
+
+    # 4. kubelet: preferred allocation honours the pre-allocation
+    plugin = VnumPlugin(manager, client, base_dir=str(tmp_path))
+    avail = [d.ID for d in plugin.fake_devices()]
+    claimed_uuid = pre.split("[")[1].split("_")[1]
+    pref = plugin.GetPreferredAllocation(api.PreferredAllocationRequest(
+        container_requests=[api.ContainerPreferredAllocationRequest(
+            available_device_ids=avail, allocation_size=1)]), None)
+    chosen = pref.container_responses[0].device_ids
+    assert len(chosen) == 1 and chosen[0].startswith(claimed_uuid)
+
+    # 5. kubelet Allocate: envs + mounts + written regions
+    resp = plugin.Allocate(api.AllocateRequest(container_requests=[
+        api.ContainerAllocateRequest(devices_ids=chosen)]), None)
+    cr = resp.container_responses[0]
+    assert cr.envs[consts.ENV_MEM_LIMIT.format(0)] == str(1 << 20)
+    assert cr.envs[consts.ENV_CORE_LIMIT.format(0)] == "50"
+    mounts = {m.container_path: m.host_path for m in cr.mounts}
+    assert "/etc/ld.so.preload" in mounts
+    cdir = os.path.join(str(tmp_path), "uid-w1_main")
+    cfg = os.path.join(cdir, "config", "vgpu.config")
+    assert os.path.exists(cfg)
+    devjson = json.load(open(os.path.join(cdir, "devices.json")))
+    assert devjson["claims"][0].startswith(f"0_{claimed_uuid}")
+
+    # pod carries the real allocation + success phase
+    pod = client.get_pod("default", "w1")
+    ann = pod["metadata"]["annotations"]
+    assert ann[consts.real_alloc_ann()] == pre
+    assert pod["metadata"]["labels"][consts.assigned_phase_label()] == \
+        consts.PHASE_SUCCESS
+
+    # 6. PreStartContainer re-verifies
+    plugin.PreStartContainer(api.PreStartContainerRequest(
+        devices_ids=chosen), None)
+
+    # 7. the REAL C shim enforces exactly the written region: the
+    #    stub-runtime quota scenario expects a 1 MiB limit on dev 0
+    run_scenario("quota", {"VGPU_CONFIG_PATH_OVERRIDE": cfg})
+
+
+def test_second_pod_rejected_when_full(tmp_path):
+    from vgpu_manager_amd.device.types import NodeConfigInfo
+    client = FakeKubeClient()
+    manager = FakeDeviceManager(
+        "gpu-node-1", n_devices=1,
+        config=NodeConfigInfo(deviceSplitCount=1))
+    client.add_node({"metadata": {"name": "gpu-node-1",
+                                  "annotations": {}}})
+    manager.register(client)
+
+    p1 = make_pod(number=1, name="a")
+    client.add_pod(p1)
+    f = GpuFilter(client)
+    assert f.filter({"Pod": p1, "NodeNames": ["gpu-node-1"]})[
+        "NodeNames"] == ["gpu-node-1"]
+    # simulate it running there
+    pod = client.get_pod("default", "a")
+    pod["spec"]["nodeName"] = "gpu-node-1"
+    ann = pod["metadata"]["annotations"]
+    ann[consts.real_alloc_ann()] = ann[consts.pre_alloc_ann()]
+    client.add_pod(pod)
+
+    p2 = make_pod(number=1, name="b")
+    client.add_pod(p2)
+    res = f.filter({"Pod": p2, "NodeNames": ["gpu-node-1"]})
+    assert res["NodeNames"] == []
+    assert "gpu-node-1" in res["FailedNodes"]
