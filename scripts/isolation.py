@@ -1,0 +1,123 @@
+#!/usr/bin/env python3
+"""Throttle-isolation benchmark — BASELINE config #3: N "pods"
+oversubscribed on one MI355X, each hard-limited to an equal CU share.
+
+Each pod is a subprocess under LD_PRELOAD of the shim with its own
+core limit (own private bucket — separate fake containers). All run
+an identical launch-storm workload for a fixed wall time; completed-
+kernel counts measure the compute each actually received.
+
+Metrics:
+  * share_error_pct — max |pod_share − fair_share| / fair_share over
+    pods (fairness of isolation between equal tenants);
+  * aggregate_vs_unthrottled — sum of throttled throughputs vs one
+    unthrottled run (how much of the GPU the 4×25% tenants get
+    together; <1 reflects throttle headroom + controller error).
+
+Usage: python scripts/isolation.py [--pods 4] [--limit 25]
+       [--seconds 15] [--out gpurun_out/isolation.json]
+"""
+import argparse
+import ctypes
+import json
+import os
+import subprocess
+import sys
+import time
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+BUILD = os.path.join(REPO, "library", "build")
+SHIM = os.path.join(BUILD, "libvgpu-control.so")
+WORKLOAD = os.path.join(BUILD, "libworkload.so")
+
+WORKER = r"""
+import ctypes, json, sys, time
+wk = ctypes.CDLL(sys.argv[1])
+assert wk.wk_init(0) == 0
+seconds = float(sys.argv[2])
+wk.wk_launch_busy(4, 512, 256, ctypes.c_longlong(20000))
+wk.wk_sync()  # warmup + shim init
+count = 0
+t0 = time.perf_counter()
+while time.perf_counter() - t0 < seconds:
+    # 8 medium kernels per batch, then sync (storm regime: the token
+    # bucket is the binding constraint, not the GAP path)
+    wk.wk_launch_busy(8, 1024, 256, ctypes.c_longlong(40000))
+    wk.wk_sync()
+    count += 8
+el = time.perf_counter() - t0
+print(json.dumps({"kernels": count, "elapsed": el,
+                  "rate": count / el}))
+"""
+
+
+def run_pod(limit, seconds, extra_env=None):
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = SHIM
+    if limit:
+        env["VGPU_CORE_LIMIT_0"] = str(limit)
+    env.update(extra_env or {})
+    return subprocess.Popen(
+        [sys.executable, "-c", WORKER, WORKLOAD, str(seconds)],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+        text=True)
+
+
+def collect(proc):
+    out, err = proc.communicate(timeout=600)
+    if proc.returncode != 0:
+        raise RuntimeError(f"worker failed: {err[-400:]}")
+    return json.loads(out.strip().splitlines()[-1])
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--pods", type=int, default=4)
+    ap.add_argument("--limit", type=int, default=25)
+    ap.add_argument("--seconds", type=float, default=15.0)
+    ap.add_argument("--out", default=None)
+    args = ap.parse_args()
+
+    # unthrottled single-tenant baseline
+    base = collect(run_pod(0, args.seconds))
+
+    # N concurrent throttled pods.  In production each container has
+    # its OWN /tmp/.sm_node + /tmp/.vmem_node mounts; emulate that
+    # with per-pod region overrides (a shared /tmp here would merge
+    # their buckets into one container and void the isolation).
+    import tempfile
+    tdir = tempfile.mkdtemp(prefix="isolation-")
+    procs = [run_pod(args.limit, args.seconds, {
+        "VGPU_SM_NODE_PATH_OVERRIDE":
+            os.path.join(tdir, f"sm_node.{i}"),
+        "VGPU_VMEM_PATH_OVERRIDE":
+            os.path.join(tdir, f"vmem.{i}"),
+    }) for i in range(args.pods)]
+    results = [collect(p) for p in procs]
+
+    rates = [r["rate"] for r in results]
+    agg = sum(rates)
+    fair = agg / len(rates)
+    share_err = max(abs(r - fair) / fair for r in rates) * 100
+    out = {
+        "pods": args.pods,
+        "core_limit_pct": args.limit,
+        "seconds": args.seconds,
+        "unthrottled_rate": base["rate"],
+        "pod_rates": rates,
+        "aggregate_rate": agg,
+        "aggregate_vs_unthrottled": agg / base["rate"],
+        "share_error_pct": round(share_err, 2),
+        "per_pod_vs_limit": [
+            round(r / base["rate"] * 100, 2) for r in rates],
+    }
+    line = json.dumps(out)
+    print(line)
+    if args.out:
+        os.makedirs(os.path.dirname(args.out) or ".", exist_ok=True)
+        with open(args.out, "w") as f:
+            f.write(line + "\n")
+
+
+if __name__ == "__main__":
+    main()
