@@ -253,3 +253,17 @@ def test_http_routes(client):
                 json={"Pod": pod, "NodeNames": ["gpu-node-1"]})
     assert r.status_code == 200
     assert r.json()["NodeNames"] == ["gpu-node-1"]
+
+
+def test_filter_failure_emits_reason_event(client):
+    # pod needing 4 GPUs; nodes have 2 -> all rejected, Event emitted
+    pod = make_pod(number=4, name="too-big")
+    client.add_pod(pod)
+    res = GpuFilter(client).filter(
+        {"Pod": pod, "NodeNames": ["gpu-node-1", "gpu-node-2",
+                                   "cpu-node"]})
+    assert res["NodeNames"] == []
+    assert client.events, "rejection must produce a pod Event"
+    ev = client.events[-1]
+    assert ev["reason"] == "FilterFailed"
+    assert "gpu-node-1" in ev["message"]

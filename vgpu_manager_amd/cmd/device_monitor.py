@@ -23,6 +23,9 @@ def main(argv=None):
     ap.add_argument("--base-dir", default=consts.MANAGER_DIR)
     ap.add_argument("--shared-watcher", action="store_true",
                     help="run the shared utilization sampler too")
+    ap.add_argument("--stuck-grace-period", type=int, default=0,
+                    help="seconds after which a pre-allocated-but-"
+                         "never-bound pod is recovered (0 = off)")
     ap.add_argument("--dra-checkpoint",
                     default="/var/lib/vgpu-manager/checkpoint.json",
                     help="DRA driver checkpoint to export claim "
@@ -49,6 +52,25 @@ def main(argv=None):
             AmdSmiSource(),
             os.path.join(args.base_dir, "watcher", "sm_util.config"))
         sampler.start_background()
+
+    if args.stuck_grace_period > 0:
+        from ..client.kube import RestKubeClient
+        from ..monitor.stuck import recover_stuck_pods
+
+        def stuck_loop():
+            import time as _t
+            client = RestKubeClient()
+            while True:
+                try:
+                    recover_stuck_pods(
+                        client,
+                        default_grace_s=args.stuck_grace_period)
+                except Exception as e:
+                    logging.warning("stuck-pod pass failed: %s", e)
+                _t.sleep(max(args.stuck_grace_period / 2, 30))
+
+        threading.Thread(target=stuck_loop, daemon=True,
+                         name="stuck-pods").start()
 
     start_http_server(args.port)
     threading.Event().wait()

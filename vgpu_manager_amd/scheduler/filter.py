@@ -114,6 +114,25 @@ class GpuFilter:
 
             if chosen is None:
                 metrics.observe("filter", time.monotonic() - t0, False)
+                # aggregate reject reasons into a pod Event (reference
+                # reason.go: reason -> [nodes], node list truncated)
+                by_reason: Dict[str, List[str]] = {}
+                for n, r in failed.items():
+                    by_reason.setdefault(r, []).append(n)
+                parts = []
+                for r, nodes in sorted(by_reason.items()):
+                    shown = ",".join(sorted(nodes)[:3])
+                    more = len(nodes) - 3
+                    if more > 0:
+                        shown += f"(+{more} more)"
+                    parts.append(f"{r}: {shown}")
+                if not dry_run:
+                    self.client.create_event(
+                        namespace,
+                        {"kind": "Pod", "name": pod_name,
+                         "namespace": namespace,
+                         "uid": meta.get("uid", "")},
+                        "FilterFailed", "; ".join(parts) or "no nodes")
                 return self._result([], failed)
 
             if not dry_run:
