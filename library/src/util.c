@@ -164,7 +164,11 @@ static int cmp_i32(const void *a, const void *b) {
 
 /* load from pids.config region if present and fresh */
 static int load_pids_from_config(pid_set_t *set) {
-    void *ptr = vgpu_region_attach(VGPU_PIDS_PATH, sizeof(pids_data_t),
+    char buf[512];
+    const char *path = vgpu_getenv("VGPU_PIDS_PATH_OVERRIDE", buf,
+                                   sizeof(buf));
+    void *ptr = vgpu_region_attach(path ? path : VGPU_PIDS_PATH,
+                                   sizeof(pids_data_t),
                                    VGPU_PIDS_MAGIC, false, NULL);
     if (!ptr) return -1;
     pids_data_t *pd = (pids_data_t *)ptr;
@@ -206,6 +210,18 @@ static int load_pids_from_cgroup(pid_set_t *set) {
 
 int vgpu_load_pid_set(pid_set_t *set) {
     set->count = 0;
+    /* bare-process tenants (no container, shared cgroup): the cgroup
+     * walk would lump every co-tenant into one set and void per-pod
+     * attribution — self-only scopes it to this process tree root.  */
+    char sbuf[16];
+    const char *self = vgpu_getenv("VGPU_PIDS_SELF_ONLY", sbuf,
+                                   sizeof(sbuf));
+    if (self && *self == '1') {
+        set->pids[0] = (int32_t)getpid();
+        set->count = 1;
+        set->loaded_ns = mono_ns();
+        return 1;
+    }
     int n = load_pids_from_config(set);
     if (n <= 0) n = load_pids_from_cgroup(set);
     if (n < 0) {

@@ -92,6 +92,9 @@ def main():
             os.path.join(tdir, f"sm_node.{i}"),
         "VGPU_VMEM_PATH_OVERRIDE":
             os.path.join(tdir, f"vmem.{i}"),
+        # per-pod attribution: these tenants share one cgroup, so the
+        # cgroup pid walk would merge them into one "container"
+        "VGPU_PIDS_SELF_ONLY": "1",
     }) for i in range(args.pods)]
     results = [collect(p) for p in procs]
 
