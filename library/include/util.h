@@ -44,16 +44,31 @@ const dynamic_config_t *vgpu_dynconfig(void);
 #define MEM_ACCOUNT_MAX    2
 
 /* container PID set: loaded from pids.config if present, else built
- * from this process tree.  Sorted; membership by binary search.       */
+ * from this process tree.  Sorted; membership by binary search.
+ *
+ * `host_pids` mirrors `pids` translated to the HOST pid namespace:
+ * KFD sysfs and amd-smi report host pids, so a namespaced container
+ * can only attribute per-process data after translation.  The bridge
+ * is the PASID: our own /proc/<pid>/fdinfo carries "pasid: N" for
+ * amdgpu fds, and /sys/class/kfd/kfd/proc/<hostpid>/pasid is global —
+ * matching them maps ns pid -> host pid (the reference reaches host
+ * pids via its mounted .host_proc / peercred registry instead).       */
 typedef struct {
     int32_t pids[MAX_DEVICE_PIDS];
+    int32_t host_pids[MAX_DEVICE_PIDS]; /* sorted; 0 = unresolved     */
     int     count;
+    int     host_count;
     uint64_t loaded_ns;
 } pid_set_t;
 
 /* (re)load the container pid set; returns number of pids (>=0).       */
 int  vgpu_load_pid_set(pid_set_t *set);
 bool vgpu_pid_set_contains(const pid_set_t *set, int32_t pid);
+/* translate ns pid -> host pid via the KFD pasid bridge; returns the
+ * input when no translation is needed or possible.                    */
+int32_t vgpu_pid_to_host(int32_t ns_pid);
+/* refresh host_pids from pids (cheap when already fully resolved).    */
+void vgpu_pid_set_resolve_host(pid_set_t *set);
 
 #ifdef __cplusplus
 }

@@ -99,13 +99,20 @@ static bool sample_util(int dev, uint32_t *cont_permille,
             if (s0 & 1u) continue;
             uint32_t busy = u->dev_busy_permille;
             uint64_t ts = u->sample_ns;
-            uint32_t cont = 0;
+            uint32_t cont = 0, cont_cus = 0;
             uint32_t n = u->proc_count;
             if (n > MAX_UTIL_PROCS) n = MAX_UTIL_PROCS;
             for (uint32_t i = 0; i < n; i++)
-                if (vgpu_pid_set_contains(&g_state.pids, u->procs[i].pid))
+                if (vgpu_pid_set_contains(&g_state.pids,
+                                          u->procs[i].pid)) {
                     cont += u->procs[i].gfx_busy_permille;
+                    cont_cus += u->procs[i].cu_occupancy;
+                }
             if (!seq_read_valid(&u->seq, s0)) continue;
+            /* KFD reports no per-proc engine time on some stacks;
+             * CU occupancy is the working attribution there         */
+            if (cont == 0 && cont_cus > 0 && h->cu_count > 0)
+                cont = cont_cus * 1000u / (uint32_t)h->cu_count;
             if (mono_ns() - ts < 1000000000ull) { /* fresh within 1 s */
                 *cont_permille = cont > 1000 ? 1000 : cont;
                 *dev_permille = busy;
@@ -389,6 +396,9 @@ static void *watcher_main(void *arg) {
         next += (uint64_t)WATCHER_CYCLE_MS * 1000000ull;
         /* ~every 3.2s: reclaim spill records of SIGKILL'd siblings   */
         if ((++cycle & 31u) == 0) vmem_ledger_sweep_dead();
+        /* KFD/amd-smi report HOST pids; keep the host view of our
+         * pid set fresh (pasids appear only once contexts exist)     */
+        vgpu_pid_set_resolve_host(&g_state.pids);
         for (int dev = 0; dev < g_state.device_count; dev++) {
             if (cfg_dev(dev) < 0) continue;
             uint32_t flags = vgpu_device_flags(dev);

@@ -5,6 +5,7 @@
 #include "shm.h"
 
 #include <ctype.h>
+#include <dirent.h>
 #include <fcntl.h>
 #include <pthread.h>
 #include <stdio.h>
@@ -235,13 +236,96 @@ int vgpu_load_pid_set(pid_set_t *set) {
     return set->count;
 }
 
-bool vgpu_pid_set_contains(const pid_set_t *set, int32_t pid) {
-    int lo = 0, hi = set->count - 1;
+static bool sorted_contains(const int32_t *arr, int n, int32_t pid) {
+    int lo = 0, hi = n - 1;
     while (lo <= hi) {
         int mid = (lo + hi) / 2;
-        if (set->pids[mid] == pid) return true;
-        if (set->pids[mid] < pid) lo = mid + 1;
+        if (arr[mid] == pid) return true;
+        if (arr[mid] < pid) lo = mid + 1;
         else hi = mid - 1;
     }
     return false;
+}
+
+bool vgpu_pid_set_contains(const pid_set_t *set, int32_t pid) {
+    /* amd-smi / KFD / the external watcher report HOST pids; our set
+     * may be ns pids — accept a match in either namespace view.      */
+    return sorted_contains(set->pids, set->count, pid) ||
+           sorted_contains(set->host_pids, set->host_count, pid);
+}
+
+/* ---- ns pid -> host pid via the KFD pasid bridge ------------------ */
+
+#define KFD_PROC_DIR "/sys/class/kfd/kfd/proc"
+
+static uint32_t read_fdinfo_pasid(int32_t ns_pid) {
+    char dirp[64];
+    snprintf(dirp, sizeof(dirp), "/proc/%d/fdinfo", ns_pid);
+    DIR *d = opendir(dirp);
+    if (!d) return 0;
+    uint32_t pasid = 0;
+    struct dirent *e;
+    while (!pasid && (e = readdir(d)) != NULL) {
+        if (e->d_name[0] == '.') continue;
+        char fp[384];
+        snprintf(fp, sizeof(fp), "%s/%s", dirp, e->d_name);
+        FILE *f = fopen(fp, "re");
+        if (!f) continue;
+        char line[128];
+        bool amdgpu = false;
+        uint32_t p = 0;
+        while (fgets(line, sizeof(line), f)) {
+            if (strncmp(line, "drm-driver:", 11) == 0 &&
+                strstr(line, "amdgpu"))
+                amdgpu = true;
+            else if (strncmp(line, "pasid:", 6) == 0)
+                p = (uint32_t)strtoul(line + 6, NULL, 10);
+        }
+        fclose(f);
+        if (amdgpu && p) pasid = p;
+    }
+    closedir(d);
+    return pasid;
+}
+
+int32_t vgpu_pid_to_host(int32_t ns_pid) {
+    char path[384];
+    /* same namespace as KFD's view: no translation needed             */
+    snprintf(path, sizeof(path), KFD_PROC_DIR "/%d", ns_pid);
+    if (access(path, F_OK) == 0) return ns_pid;
+    uint32_t pasid = read_fdinfo_pasid(ns_pid);
+    if (!pasid) return ns_pid;
+    DIR *d = opendir(KFD_PROC_DIR);
+    if (!d) return ns_pid;
+    int32_t host = ns_pid;
+    struct dirent *e;
+    while ((e = readdir(d)) != NULL) {
+        if (e->d_name[0] < '0' || e->d_name[0] > '9') continue;
+        snprintf(path, sizeof(path), KFD_PROC_DIR "/%s/pasid",
+                 e->d_name);
+        FILE *f = fopen(path, "re");
+        if (!f) continue;
+        uint32_t p = 0;
+        if (fscanf(f, "%u", &p) != 1) p = 0;
+        fclose(f);
+        if (p == pasid) {
+            host = (int32_t)atoi(e->d_name);
+            break;
+        }
+    }
+    closedir(d);
+    return host;
+}
+
+void vgpu_pid_set_resolve_host(pid_set_t *set) {
+    /* re-resolve each call: a pasid only exists once the process has
+     * opened a GPU context, so early calls legitimately fail and the
+     * watcher retries every cycle (a handful of file reads).         */
+    int n = 0;
+    for (int i = 0; i < set->count && n < MAX_DEVICE_PIDS; i++) {
+        int32_t h = vgpu_pid_to_host(set->pids[i]);
+        if (h != 0) set->host_pids[n++] = h;
+    }
+    set->host_count = n;
+    qsort(set->host_pids, (size_t)n, sizeof(int32_t), cmp_i32);
 }
