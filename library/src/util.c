@@ -321,11 +321,21 @@ void vgpu_pid_set_resolve_host(pid_set_t *set) {
     /* re-resolve each call: a pasid only exists once the process has
      * opened a GPU context, so early calls legitimately fail and the
      * watcher retries every cycle (a handful of file reads).         */
-    int n = 0;
+    int n = 0, translated = 0;
     for (int i = 0; i < set->count && n < MAX_DEVICE_PIDS; i++) {
         int32_t h = vgpu_pid_to_host(set->pids[i]);
-        if (h != 0) set->host_pids[n++] = h;
+        if (h != 0) {
+            if (h != set->pids[i]) translated++;
+            set->host_pids[n++] = h;
+        }
     }
     set->host_count = n;
     qsort(set->host_pids, (size_t)n, sizeof(int32_t), cmp_i32);
+    static int last_logged = -1;
+    if (translated != last_logged) {
+        last_logged = translated;
+        LOGGER(LOG_DEBUG, "pid host-view: %d pids, %d ns->host "
+               "translations (self %d -> %d)", n, translated,
+               (int)getpid(), (int)vgpu_pid_to_host((int32_t)getpid()));
+    }
 }
