@@ -131,3 +131,33 @@ def test_dra_claim_collector(tmp_path):
     # missing checkpoint -> no families, no crash
     assert list(DraClaimCollector(
         str(tmp_path / "nope.json"), "n").collect()) == []
+
+
+def test_metrics_server_rate_limit_and_serve():
+    import urllib.request
+    from prometheus_client import CollectorRegistry, Counter
+    from vgpu_manager_amd.monitor.server import serve_metrics
+
+    reg = CollectorRegistry()
+    Counter("test_total", "t", registry=reg).inc()
+    srv = serve_metrics(0, registry=reg, rate=2.0, burst=2,
+                        bind="127.0.0.1")
+    port = srv.server_address[1]
+    try:
+        url = f"http://127.0.0.1:{port}/metrics"
+        body = urllib.request.urlopen(url).read().decode()
+        assert "test_total" in body
+        # burst exhausted -> 429
+        codes = []
+        for _ in range(4):
+            try:
+                urllib.request.urlopen(url)
+                codes.append(200)
+            except urllib.error.HTTPError as e:
+                codes.append(e.code)
+        assert 429 in codes
+        # healthz is never limited
+        assert urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/healthz").read() == b"ok"
+    finally:
+        srv.shutdown()

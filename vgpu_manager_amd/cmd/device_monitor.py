@@ -8,7 +8,7 @@ import os
 import sys
 import threading
 
-from prometheus_client import start_http_server, REGISTRY
+from prometheus_client import REGISTRY
 
 from ..monitor.collector import NodeVgpuCollector, PhysicalGpuCollector
 from ..monitor.lister import ContainerLister
@@ -23,6 +23,10 @@ def main(argv=None):
     ap.add_argument("--base-dir", default=consts.MANAGER_DIR)
     ap.add_argument("--shared-watcher", action="store_true",
                     help="run the shared utilization sampler too")
+    ap.add_argument("--tls-cert", default="")
+    ap.add_argument("--tls-key", default="")
+    ap.add_argument("--scrape-rate", type=float, default=2.0,
+                    help="max /metrics scrapes per second")
     ap.add_argument("--stuck-grace-period", type=int, default=0,
                     help="seconds after which a pre-allocated-but-"
                          "never-bound pod is recovered (0 = off)")
@@ -72,7 +76,10 @@ def main(argv=None):
         threading.Thread(target=stuck_loop, daemon=True,
                          name="stuck-pods").start()
 
-    start_http_server(args.port)
+    from ..monitor.server import serve_metrics
+    serve_metrics(args.port, certfile=args.tls_cert or None,
+                  keyfile=args.tls_key or None,
+                  rate=args.scrape_rate)
     threading.Event().wait()
     return 0
 
