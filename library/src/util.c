@@ -335,7 +335,8 @@ void vgpu_pid_set_resolve_host(pid_set_t *set) {
     if (translated != last_logged) {
         last_logged = translated;
         LOGGER(LOG_DEBUG, "pid host-view: %d pids, %d ns->host "
-               "translations (self %d -> %d)", n, translated,
-               (int)getpid(), (int)vgpu_pid_to_host((int32_t)getpid()));
+               "translations (self %d pasid %u -> %d)", n, translated,
+               (int)getpid(), read_fdinfo_pasid((int32_t)getpid()),
+               (int)vgpu_pid_to_host((int32_t)getpid()));
     }
 }
