@@ -258,16 +258,27 @@ bool vgpu_pid_set_contains(const pid_set_t *set, int32_t pid) {
 
 #define KFD_PROC_DIR "/sys/class/kfd/kfd/proc"
 
+/* test injection: fake /proc and KFD sysfs roots */
+static const char *proc_root(char *buf, size_t n) {
+    const char *v = vgpu_getenv("VGPU_PROC_DIR_OVERRIDE", buf, n);
+    return v ? v : "/proc";
+}
+static const char *kfd_proc_root(char *buf, size_t n) {
+    const char *v = vgpu_getenv("VGPU_KFD_PROC_DIR_OVERRIDE", buf, n);
+    return v ? v : KFD_PROC_DIR;
+}
+
 static uint32_t read_fdinfo_pasid(int32_t ns_pid) {
-    char dirp[64];
-    snprintf(dirp, sizeof(dirp), "/proc/%d/fdinfo", ns_pid);
+    char rbuf[256], dirp[320];
+    snprintf(dirp, sizeof(dirp), "%s/%d/fdinfo",
+             proc_root(rbuf, sizeof(rbuf)), ns_pid);
     DIR *d = opendir(dirp);
     if (!d) return 0;
     uint32_t pasid = 0;
     struct dirent *e;
     while (!pasid && (e = readdir(d)) != NULL) {
         if (e->d_name[0] == '.') continue;
-        char fp[384];
+        char fp[640];
         snprintf(fp, sizeof(fp), "%s/%s", dirp, e->d_name);
         FILE *f = fopen(fp, "re");
         if (!f) continue;
@@ -289,20 +300,20 @@ static uint32_t read_fdinfo_pasid(int32_t ns_pid) {
 }
 
 int32_t vgpu_pid_to_host(int32_t ns_pid) {
-    char path[384];
+    char kbuf[256], path[640];
+    const char *kfd = kfd_proc_root(kbuf, sizeof(kbuf));
     /* same namespace as KFD's view: no translation needed             */
-    snprintf(path, sizeof(path), KFD_PROC_DIR "/%d", ns_pid);
+    snprintf(path, sizeof(path), "%s/%d", kfd, ns_pid);
     if (access(path, F_OK) == 0) return ns_pid;
     uint32_t pasid = read_fdinfo_pasid(ns_pid);
     if (!pasid) return ns_pid;
-    DIR *d = opendir(KFD_PROC_DIR);
+    DIR *d = opendir(kfd);
     if (!d) return ns_pid;
     int32_t host = ns_pid;
     struct dirent *e;
     while ((e = readdir(d)) != NULL) {
         if (e->d_name[0] < '0' || e->d_name[0] > '9') continue;
-        snprintf(path, sizeof(path), KFD_PROC_DIR "/%s/pasid",
-                 e->d_name);
+        snprintf(path, sizeof(path), "%s/%s/pasid", kfd, e->d_name);
         FILE *f = fopen(path, "re");
         if (!f) continue;
         uint32_t p = 0;

@@ -72,3 +72,40 @@ def test_hook_config_file(built_library, tmp_path):
     # scenario "quota" expects a 1 MiB limit on device 0 — now sourced
     # from the region file instead of env
     run_scenario("quota", {"VGPU_CONFIG_PATH_OVERRIDE": p})
+
+
+def test_pid_host_translation_via_pasid(built_library, tmp_path):
+    """Cross-pid-namespace attribution: the shim maps its ns pid to
+    the host pid through the KFD pasid bridge (fake /proc + KFD sysfs
+    trees; util.c vgpu_pid_to_host)."""
+    import subprocess as sp
+
+    fake_kfd = tmp_path / "kfd_proc"
+    (fake_kfd / "987654").mkdir(parents=True)
+    (fake_kfd / "987654" / "pasid").write_text("777\n")
+    fake_proc = tmp_path / "proc"
+    fake_proc.mkdir()
+
+    build = os.path.join(LIB_DIR, "build")
+    env = dict(os.environ)
+    env.update({
+        "VGPU_CORE_LIMIT_0": "50",
+        "VGPU_LOGGER_LEVEL": "5",
+        "VGPU_PIDS_SELF_ONLY": "1",
+        "VGPU_PROC_DIR_OVERRIDE": str(fake_proc),
+        "VGPU_KFD_PROC_DIR_OVERRIDE": str(fake_kfd),
+        "LD_PRELOAD": os.path.join(build, "libvgpu-control.so"),
+        "LD_LIBRARY_PATH": os.path.join(build, "stub"),
+        "VGPU_REAL_HIP_PATH": os.path.join(build, "stub",
+                                           "libamdhip64.so.7"),
+    })
+    p = sp.Popen([os.path.join(build, "test_hook_cpu"), "throttle"],
+                 env=env, stdout=sp.PIPE, stderr=sp.PIPE, text=True)
+    # fake fdinfo for the just-spawned pid: an amdgpu fd with pasid 777
+    fdinfo = fake_proc / str(p.pid) / "fdinfo"
+    fdinfo.mkdir(parents=True)
+    (fdinfo / "3").write_text(
+        "pos:\t0\ndrm-driver:\tamdgpu\npasid:\t777\n")
+    out, err = p.communicate(timeout=120)
+    assert p.returncode == 0, out + err
+    assert f"-> 987654" in err, f"no ns->host translation logged:\n{err}"
