@@ -209,6 +209,48 @@ static int load_pids_from_cgroup(pid_set_t *set) {
     return n;
 }
 
+/* Host-proc mode (reference .host_proc mount, Appendix B + the
+ * cgroup/host-proc PID filtering of cuda_hook.c:2191-2340): the node
+ * agent bind-mounts host /proc read-only at
+ * /etc/vgpu-manager/.host_proc; we walk it and keep HOST pids whose
+ * cgroup path names our pod UID — host pids are what KFD/amd-smi
+ * report, so this gives working attribution without a pid-ns bridge. */
+static int load_pids_from_host_proc(pid_set_t *set) {
+    char rbuf[256];
+    const char *root = vgpu_getenv("VGPU_HOST_PROC_DIR_OVERRIDE", rbuf,
+                                   sizeof(rbuf));
+    if (!root) root = VGPU_MANAGER_DIR "/.host_proc";
+    char ubuf[128];
+    const char *uid = vgpu_getenv("VGPU_POD_UID", ubuf, sizeof(ubuf));
+    if (!uid || !*uid) return -1;
+    DIR *d = opendir(root);
+    if (!d) return -1;
+    /* cgroup paths spell the UID with '_' on some runtimes           */
+    char uid_us[128];
+    snprintf(uid_us, sizeof(uid_us), "%s", uid);
+    for (char *p = uid_us; *p; p++)
+        if (*p == '-') *p = '_';
+    int n = 0;
+    struct dirent *e;
+    while (n < MAX_DEVICE_PIDS && (e = readdir(d)) != NULL) {
+        if (e->d_name[0] < '0' || e->d_name[0] > '9') continue;
+        char path[640];
+        snprintf(path, sizeof(path), "%s/%s/cgroup", root, e->d_name);
+        FILE *f = fopen(path, "re");
+        if (!f) continue;
+        char line[512];
+        bool mine = false;
+        while (!mine && fgets(line, sizeof(line), f))
+            if (strstr(line, uid) || strstr(line, uid_us))
+                mine = true;
+        fclose(f);
+        if (mine) set->pids[n++] = (int32_t)atoi(e->d_name);
+    }
+    closedir(d);
+    set->count = n;
+    return n > 0 ? n : -1;
+}
+
 int vgpu_load_pid_set(pid_set_t *set) {
     set->count = 0;
     /* bare-process tenants (no container, shared cgroup): the cgroup
@@ -224,6 +266,7 @@ int vgpu_load_pid_set(pid_set_t *set) {
         return 1;
     }
     int n = load_pids_from_config(set);
+    if (n <= 0) n = load_pids_from_host_proc(set);
     if (n <= 0) n = load_pids_from_cgroup(set);
     if (n < 0) {
         /* last resort: just this process */

@@ -109,3 +109,38 @@ def test_pid_host_translation_via_pasid(built_library, tmp_path):
     out, err = p.communicate(timeout=120)
     assert p.returncode == 0, out + err
     assert f"-> 987654" in err, f"no ns->host translation logged:\n{err}"
+
+
+def test_pid_set_from_host_proc(built_library, tmp_path):
+    """Production attribution: host /proc mounted at .host_proc, pids
+    selected by pod UID appearing in their host cgroup path."""
+    host_proc = tmp_path / "host_proc"
+    uid = "abc-123-def"
+    for pid, cg in [
+        (500100, f"0::/kubepods/poduid_{uid.replace('-', '_')}/c1\n"),
+        (500101, f"0::/kubepods/pod{uid}/c1\n"),
+        (500999, "0::/kubepods/podother-uid/c9\n"),
+        (1, "0::/init.scope\n"),
+    ]:
+        d = host_proc / str(pid)
+        d.mkdir(parents=True)
+        (d / "cgroup").write_text(cg)
+
+    build = os.path.join(LIB_DIR, "build")
+    env = dict(os.environ)
+    env.update({
+        "VGPU_CORE_LIMIT_0": "50",
+        "VGPU_LOGGER_LEVEL": "5",
+        "VGPU_POD_UID": uid,
+        "VGPU_HOST_PROC_DIR_OVERRIDE": str(host_proc),
+        "LD_PRELOAD": os.path.join(build, "libvgpu-control.so"),
+        "LD_LIBRARY_PATH": os.path.join(build, "stub"),
+        "VGPU_REAL_HIP_PATH": os.path.join(build, "stub",
+                                           "libamdhip64.so.7"),
+    })
+    import subprocess as sp
+    r = sp.run([os.path.join(build, "test_hook_cpu"), "launch"],
+               env=env, capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stdout + r.stderr
+    # both spellings of the pod uid matched; foreign pids excluded
+    assert "host-view: 2 pids" in r.stderr, r.stderr

@@ -58,7 +58,8 @@ def main(argv=None):
     plugin_set = PluginSet(
         manager, client, open_vcore=config.openVCore,
         open_vmemory=config.openVMemory, driver_lib=args.driver_lib,
-        shared_watcher=gates.enabled("SharedSMUtilizationWatcher"))
+        shared_watcher=gates.enabled("SharedSMUtilizationWatcher"),
+        client_mode=gates.enabled("DevicePluginClientMode"))
     plugin_set.start_all(args.kubelet_socket)
     watch_kubelet_restart(plugin_set, args.kubelet_socket)
 

@@ -105,6 +105,7 @@ class PluginSet:
     def __init__(self, manager, client, *, open_vcore=False,
                  open_vmemory=False,
                  plugins_dir: str = api.PLUGINS_DIR, **vnum_kwargs):
+        # vnum_kwargs may carry client_mode/shared_watcher/driver_lib
         from .vnum_plugin import VnumPlugin
         self.servers: List[PluginServer] = []
         vnum = VnumPlugin(manager, client, **vnum_kwargs)

@@ -50,12 +50,14 @@ class VnumPlugin:
                  base_dir: str = consts.MANAGER_DIR,
                  driver_lib: str = "/usr/local/vgpu-manager/"
                                    + consts.DRIVER_LIB_NAME,
-                 shared_watcher: bool = False):
+                 shared_watcher: bool = False,
+                 client_mode: bool = False):
         self.manager = manager
         self.client = client
         self.base_dir = base_dir
         self.driver_lib = driver_lib
         self.shared_watcher = shared_watcher
+        self.client_mode = client_mode
         self.resource_name = consts.vgpu_number_resource()
         self._lw_cond = threading.Condition()
         self._stopped = False
@@ -271,6 +273,12 @@ class VnumPlugin:
         if self.shared_watcher:
             mounts.append((f"{consts.MANAGER_DIR}/watcher",
                            os.path.join(self.base_dir, "watcher"), ro))
+        if not self.client_mode:
+            # host /proc (ro): the shim resolves its HOST pids by pod
+            # UID in host cgroup paths — KFD/amd-smi report host pids
+            # (reference .host_proc mount, Appendix B)
+            mounts.append((f"{consts.MANAGER_DIR}/.host_proc",
+                           "/proc", ro))
         for cpath, hpath, read_only in mounts:
             resp.mounts.append(api.Mount(container_path=cpath,
                                          host_path=hpath,
