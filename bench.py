@@ -204,6 +204,18 @@ def main():
                   env)
 
     # ---- measurement ----
+    # all ranks move through the phases in lockstep so no rank's
+    # untimed phases overlap another rank's timed region
+    dist = None
+    # gloo lockstep only with a real GPU: importing torch would load
+    # the REAL libamdhip64 and break the stub's soname resolution in
+    # the CPU smoke (each rank then free-runs; rank 0 still reports)
+    if world > 1 and gpu:
+        import torch.distributed as tdist
+        dist = tdist
+        dist.init_process_group(backend="gloo")
+        dist.barrier()
+
     # 1. bare reference (subprocess, untimed region)
     bare = spawn_inner("bare", device, args.steps, args.warmup, {},
                        preload=False)
@@ -243,11 +255,8 @@ def main():
         quota_error_pct = max(errs) if errs else None
 
     # 3. hooked timed region (the contract steps)
-    dist = None
-    if world > 1:
-        import torch.distributed as tdist
-        dist = tdist
-        dist.init_process_group(backend="gloo")
+    if dist:
+        dist.barrier()
     wk = load_workload(device)
     run_steps(wk, args.warmup)
     if dist:
