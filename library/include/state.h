@@ -192,6 +192,8 @@ bool smi_sample_device(int host_index, uint32_t *busy_permille,
                        uint64_t *container_vram, uint32_t *container_cus,
                        const pid_set_t *pids);
 uint64_t smi_container_vram(int host_index, const pid_set_t *pids);
+/* last-resort ns->host self identification by VRAM probe; 0 = fail  */
+int32_t smi_self_host_pid(int dev);
 
 /* hook implementations (exported) live in hip_hook.c / smi_hook.c     */
 

@@ -58,6 +58,7 @@ typedef struct {
     int32_t host_pids[MAX_DEVICE_PIDS]; /* sorted; 0 = unresolved     */
     int     count;
     int     host_count;
+    int32_t self_host_pid;  /* VRAM-probe result (smi_self_host_pid)  */
     uint64_t loaded_ns;
 } pid_set_t;
 
@@ -69,6 +70,9 @@ bool vgpu_pid_set_contains(const pid_set_t *set, int32_t pid);
 int32_t vgpu_pid_to_host(int32_t ns_pid);
 /* refresh host_pids from pids (cheap when already fully resolved).    */
 void vgpu_pid_set_resolve_host(pid_set_t *set);
+/* true when the set already has a real ns->host mapping (translated
+ * entry or a probed self_host_pid) — i.e. no further probe needed.    */
+bool vgpu_pid_set_translated(const pid_set_t *set);
 
 #ifdef __cplusplus
 }

@@ -47,6 +47,7 @@ static inline int cfg_dev(int dev) {
  * but NOT the watcher thread — the flag is re-armed by the atfork
  * child handler so the child's first throttled launch restarts it.    */
 static int g_watcher_state; /* 0 = not running, 1 = started            */
+static int g_self_probe_tries; /* vram-probe self host-pid attempts    */
 
 static void dev_hot_init(int dev) {
     dev_hot_t *h = &g_state.dev[dev];
@@ -399,6 +400,17 @@ static void *watcher_main(void *arg) {
         /* KFD/amd-smi report HOST pids; keep the host view of our
          * pid set fresh (pasids appear only once contexts exist)     */
         vgpu_pid_set_resolve_host(&g_state.pids);
+        /* kernels whose kfd pasid files read 0 defeat the sysfs
+         * bridge; identify OUR host pid by a one-shot VRAM probe so
+         * per-process attribution still works (max 3 attempts;
+         * re-armed for fork children in vgpu_hook_fork_child)       */
+        if (!vgpu_pid_set_translated(&g_state.pids) &&
+            g_self_probe_tries < 3) {
+            g_self_probe_tries++;
+            int32_t hp = smi_self_host_pid(0);
+            if (hp > 0 && hp != (int32_t)getpid())
+                g_state.pids.self_host_pid = hp;
+        }
         for (int dev = 0; dev < g_state.device_count; dev++) {
             if (cfg_dev(dev) < 0) continue;
             uint32_t flags = vgpu_device_flags(dev);
@@ -439,6 +451,7 @@ static void start_watcher(void) {
  * empty bucket nobody refills (reference cuda_hook.c:260-315 analog). */
 void vgpu_hook_fork_child(void) {
     __atomic_store_n(&g_watcher_state, 0, __ATOMIC_RELEASE);
+    g_self_probe_tries = 0; /* the child is a NEW host pid            */
     __atomic_store_n(&g_watcher_parked, 0, __ATOMIC_RELEASE);
     __atomic_store_n(&g_shutdown, 0, __ATOMIC_RELEASE);
     for (int i = 0; i < MAX_DEVICE_COUNT; i++) {

@@ -17,6 +17,8 @@
 
 #include <amd_smi/amdsmi.h>
 #include <dlfcn.h>
+#include <time.h>
+#include <unistd.h>
 #include <pthread.h>
 #include <stdlib.h>
 #include <string.h>
@@ -320,4 +322,73 @@ void *vgpu_smi_lookup_hook(const char *name) {
     for (const smi_hook_entry_t *e = g_smi_hooks; e->name; e++)
         if (strcmp(e->name, name) == 0) return e->fn;
     return NULL;
+}
+
+/* ------------------------------------------------------------------ */
+/* self host-pid identification by VRAM probe                          */
+/*                                                                     */
+/* Last-resort pid-namespace bridge: KFD sysfs/amd-smi report HOST     */
+/* pids, and on kernels whose kfd proc `pasid` files read 0 the sysfs  */
+/* bridge (util.c) cannot match.  Here we allocate a distinctive VRAM  */
+/* amount through the REAL runtime and find the host pid whose VRAM    */
+/* grew by exactly that much.  Retried by the caller if a co-tenant's  */
+/* concurrent allocation makes the delta ambiguous.                    */
+/* ------------------------------------------------------------------ */
+int32_t smi_self_host_pid(int dev) {
+    if (!smi_available() || !real_hip.hipMalloc || !real_hip.hipFree)
+        return 0;
+    amdsmi_processor_handle h = handle_for(dev);
+    if (!h || !real_smi.amdsmi_get_gpu_process_list) return 0;
+
+    enum { MAXP = 128 };
+    amdsmi_proc_info_t before[MAXP], after[MAXP];
+    uint32_t nb = MAXP, na = MAXP;
+    memset(before, 0, sizeof(before));
+    if (real_smi.amdsmi_get_gpu_process_list(h, &nb, before) !=
+        AMDSMI_STATUS_SUCCESS)
+        return 0;
+    if (nb > MAXP) nb = MAXP;
+
+    /* distinctive size: 48 MiB + pid-salted page count               */
+    size_t probe = (48u << 20) + ((size_t)(getpid() & 0xff) << 12);
+    void *p = NULL;
+    if (real_hip.hipMalloc(&p, probe) != hipSuccess || !p) return 0;
+
+    int32_t found = 0;
+    for (int attempt = 0; attempt < 10 && !found; attempt++) {
+        struct timespec ts = {0, 50000000L}; /* 50ms: let smi refresh */
+        nanosleep(&ts, NULL);
+        na = MAXP;
+        memset(after, 0, sizeof(after));
+        if (real_smi.amdsmi_get_gpu_process_list(h, &na, after) !=
+            AMDSMI_STATUS_SUCCESS)
+            break;
+        if (na > MAXP) na = MAXP;
+        int candidates = 0;
+        int32_t cand = 0;
+        for (uint32_t i = 0; i < na; i++) {
+            uint64_t prev = 0;
+            for (uint32_t j = 0; j < nb; j++)
+                if (before[j].pid == after[i].pid) {
+                    prev = before[j].memory_usage.vram_mem;
+                    break;
+                }
+            uint64_t now = after[i].memory_usage.vram_mem;
+            if (now > prev) {
+                uint64_t delta = now - prev;
+                /* within 4 MiB of the probe (allocator granularity)  */
+                if (delta >= probe - (4u << 20) &&
+                    delta <= probe + (4u << 20)) {
+                    candidates++;
+                    cand = (int32_t)after[i].pid;
+                }
+            }
+        }
+        if (candidates == 1) found = cand;
+    }
+    real_hip.hipFree(p);
+    if (found)
+        LOGGER(LOG_INFO, "self host pid identified by vram probe: %d",
+               found);
+    return found;
 }

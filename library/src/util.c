@@ -158,6 +158,8 @@ const dynamic_config_t *vgpu_dynconfig(void) {
 
 /* ---------------- pid sets ---------------- */
 
+static bool sorted_contains(const int32_t *arr, int n, int32_t pid);
+
 static int cmp_i32(const void *a, const void *b) {
     int32_t x = *(const int32_t *)a, y = *(const int32_t *)b;
     return (x > y) - (x < y);
@@ -251,8 +253,18 @@ static int load_pids_from_host_proc(pid_set_t *set) {
     return n > 0 ? n : -1;
 }
 
+bool vgpu_pid_set_translated(const pid_set_t *set) {
+    if (set->self_host_pid > 0) return true;
+    for (int i = 0; i < set->host_count; i++)
+        if (!sorted_contains(set->pids, set->count,
+                             set->host_pids[i]))
+            return true;
+    return false;
+}
+
 int vgpu_load_pid_set(pid_set_t *set) {
     set->count = 0;
+    set->self_host_pid = 0; /* a fork child must re-identify          */
     /* bare-process tenants (no container, shared cgroup): the cgroup
      * walk would lump every co-tenant into one set and void per-pod
      * attribution — self-only scopes it to this process tree root.  */
@@ -382,6 +394,10 @@ void vgpu_pid_set_resolve_host(pid_set_t *set) {
             if (h != set->pids[i]) translated++;
             set->host_pids[n++] = h;
         }
+    }
+    if (set->self_host_pid > 0 && n < MAX_DEVICE_PIDS) {
+        set->host_pids[n++] = set->self_host_pid;
+        translated++;
     }
     set->host_count = n;
     qsort(set->host_pids, (size_t)n, sizeof(int32_t), cmp_i32);

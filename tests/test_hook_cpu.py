@@ -139,7 +139,9 @@ def test_pid_set_from_host_proc(built_library, tmp_path):
                                            "libamdhip64.so.7"),
     })
     import subprocess as sp
-    r = sp.run([os.path.join(build, "test_hook_cpu"), "launch"],
+    # throttle runs >=0.15s: several watcher cycles, so the host-view
+    # resolution definitely runs before exit
+    r = sp.run([os.path.join(build, "test_hook_cpu"), "throttle"],
                env=env, capture_output=True, text=True, timeout=120)
     assert r.returncode == 0, r.stdout + r.stderr
     # both spellings of the pod uid matched; foreign pids excluded
