@@ -29,6 +29,24 @@ BUILD = os.path.join(REPO, "library", "build")
 SHIM = os.path.join(BUILD, "libvgpu-control.so")
 WORKLOAD = os.path.join(BUILD, "libworkload.so")
 
+# GAP-path workload: ONE big synchronous kernel at a time with idle
+# gaps — the regime that defeats any token bucket (reference
+# sm_core_limit_gap_throttle_design.md:29: observed ~100% busy at
+# hard_core=30 without the GAP path)
+GAP_WORKER_CODE = r"""
+import ctypes, sys, time
+wk = ctypes.CDLL(%r)
+wk.wk_launch_busy.argtypes = [ctypes.c_int]*3 + [ctypes.c_longlong]
+assert wk.wk_init(0) == 0
+deadline = time.time() + %f
+while time.time() < deadline:
+    # ~300ms single kernel, then sync: sparse-launch regime
+    wk.wk_launch_busy(1, 8192, 256, 3000000)
+    wk.wk_sync()
+    time.sleep(0.25)   # >200ms idle gap re-arms the GAP detector
+print("worker done")
+"""
+
 WORKER_CODE = r"""
 import ctypes, sys, time
 wk = ctypes.CDLL(%r)
@@ -56,17 +74,21 @@ def sample_busy(samples, stop, period=0.1):
         stop.wait(period)
 
 
-def run_case(controller, target, duration, warm_frac=0.35):
+def run_case(controller, target, duration, warm_frac=0.35,
+             gap_mode=None):
     env = dict(os.environ)
     env["LD_PRELOAD"] = SHIM
     env["VGPU_CORE_LIMIT_0"] = str(target)
     env["VGPU_CU_CONTROLLER"] = controller
+    if gap_mode == "off":
+        env["VGPU_GAP_DISABLE"] = "1"
     samples = []
     stop = threading.Event()
     t = threading.Thread(target=sample_busy, args=(samples, stop),
                          daemon=True)
     t.start()
-    code = WORKER_CODE % (WORKLOAD, float(duration))
+    code = (GAP_WORKER_CODE if gap_mode else WORKER_CODE) % (
+        WORKLOAD, float(duration))
     r = subprocess.run([sys.executable, "-c", code], env=env,
                        capture_output=True, text=True,
                        timeout=duration * 8 + 120)
@@ -85,6 +107,7 @@ def run_case(controller, target, duration, warm_frac=0.35):
     mae = statistics.mean(abs(b - target) for b in window)
     mean_busy = statistics.mean(window)
     return dict(controller=controller, target=target,
+                gap=gap_mode,
                 mae=round(mae, 2), mean_busy=round(mean_busy, 1),
                 n_samples=len(window),
                 wall_s=round(t_end - t0, 1))
@@ -96,14 +119,22 @@ def main():
     ap.add_argument("--controllers", default="delta,aimd")
     ap.add_argument("--duration", type=float, default=20.0)
     ap.add_argument("--out", default="gpurun_out/ablation.json")
+    ap.add_argument("--gap-test", action="store_true",
+                    help="big-sync-kernel regime: GAP path on vs off")
     args = ap.parse_args()
 
     results = []
-    for ctl in args.controllers.split(","):
-        for tgt in args.targets.split(","):
-            res = run_case(ctl.strip(), int(tgt), args.duration)
+    if args.gap_test:
+        for mode in ("on", "off"):
+            res = run_case("aimd", 30, args.duration, gap_mode=mode)
             print(json.dumps(res), flush=True)
             results.append(res)
+    else:
+        for ctl in args.controllers.split(","):
+            for tgt in args.targets.split(","):
+                res = run_case(ctl.strip(), int(tgt), args.duration)
+                print(json.dumps(res), flush=True)
+                results.append(res)
     os.makedirs(os.path.dirname(args.out) or ".", exist_ok=True)
     with open(args.out, "w") as f:
         json.dump({"results": results,
