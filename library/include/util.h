@@ -73,6 +73,8 @@ void vgpu_pid_set_resolve_host(pid_set_t *set);
 /* true when the set already has a real ns->host mapping (translated
  * entry or a probed self_host_pid) — i.e. no further probe needed.    */
 bool vgpu_pid_set_translated(const pid_set_t *set);
+/* instantaneous CUs occupied by the set's host pids (KFD sysfs).      */
+uint32_t vgpu_kfd_cu_occupancy_sum(const pid_set_t *set);
 
 #ifdef __cplusplus
 }

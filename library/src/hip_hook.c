@@ -158,6 +158,17 @@ static bool sample_util(int dev, uint32_t *cont_permille,
         h->prev_sample_ns = now;
         if (cont == 0 && cus > 0 && h->cu_count > 0)
             cont = cus * 1000u / (uint32_t)h->cu_count;
+        if (cont == 0 && h->cu_count > 0) {
+            /* KFD sysfs CU occupancy of our host pids: a point
+             * sample, smoothed with an EWMA (alpha 1/4) across
+             * cycles so sync gaps do not read as idleness          */
+            uint32_t occ = vgpu_kfd_cu_occupancy_sum(&g_state.pids);
+            uint32_t inst = occ * 1000u / (uint32_t)h->cu_count;
+            if (inst > 1000) inst = 1000;
+            int32_t delta = (int32_t)inst - (int32_t)h->occ_ema;
+            h->occ_ema = (uint32_t)((int32_t)h->occ_ema + delta / 4);
+            cont = h->occ_ema;
+        }
         *cont_permille = cont > 1000 ? 1000 : cont;
         *dev_permille = busy;
         /* publish for siblings when we own the shared bucket          */

@@ -410,3 +410,34 @@ void vgpu_pid_set_resolve_host(pid_set_t *set) {
                (int)vgpu_pid_to_host((int32_t)getpid()));
     }
 }
+
+/* Sum of kfd proc stats cu_occupancy files over
+ * the set's host pids: CUs our processes occupy RIGHT NOW (0..CUs per
+ * GPU; gfx950 reports 0..256).  Point samples — callers smooth with
+ * an EWMA.  This is the per-process compute-attribution source on
+ * kernels whose amd-smi reports no per-process engine time.          */
+uint32_t vgpu_kfd_cu_occupancy_sum(const pid_set_t *set) {
+    char kbuf[256];
+    const char *kfd = kfd_proc_root(kbuf, sizeof(kbuf));
+    uint32_t total = 0;
+    for (int i = 0; i < set->host_count; i++) {
+        char dirp[640];
+        snprintf(dirp, sizeof(dirp), "%s/%d", kfd, set->host_pids[i]);
+        DIR *d = opendir(dirp);
+        if (!d) continue;
+        struct dirent *e;
+        while ((e = readdir(d)) != NULL) {
+            if (strncmp(e->d_name, "stats_", 6) != 0) continue;
+            char fp[960];
+            snprintf(fp, sizeof(fp), "%s/%s/cu_occupancy", dirp,
+                     e->d_name);
+            FILE *f = fopen(fp, "re");
+            if (!f) continue;
+            unsigned v = 0;
+            if (fscanf(f, "%u", &v) == 1) total += v;
+            fclose(f);
+        }
+        closedir(d);
+    }
+    return total;
+}
