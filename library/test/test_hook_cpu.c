@@ -225,6 +225,41 @@ static int scenario_cleanup(void) {
     return 0;
 }
 
+static int scenario_graph(void) {
+    /* env: VGPU_CORE_LIMIT_0=50.  Graph launches must be charged the
+     * SUM of their kernel nodes' grids (hook graph-cost accounting) —
+     * a graph-heavy app must not tunnel under the token bucket.      */
+    hipGraph_t graph = NULL;
+    CHECK(hipGraphCreate(&graph, 0) == hipSuccess);
+    hipKernelNodeParams p;
+    memset(&p, 0, sizeof(p));
+    p.blockDim = (dim3){256, 1, 1};
+    for (int i = 0; i < 8; i++) {
+        p.gridDim = (dim3){4096, 1, 1}; /* 8 x 4096 grids per launch */
+        hipGraphNode_t node;
+        CHECK(hipGraphAddKernelNode(&node, graph, NULL, 0, &p) ==
+              hipSuccess);
+    }
+    hipGraphExec_t exec = NULL;
+    CHECK(hipGraphInstantiate(&exec, graph, NULL, NULL, 0) ==
+          hipSuccess);
+
+    struct timespec t0, t1;
+    clock_gettime(CLOCK_MONOTONIC, &t0);
+    uint64_t before = stub_count_launch();
+    for (int i = 0; i < 100; i++) /* 3.3M grid tokens total          */
+        CHECK(hipGraphLaunch(exec, NULL) == hipSuccess);
+    clock_gettime(CLOCK_MONOTONIC, &t1);
+    CHECK(stub_count_launch() - before == 800); /* 100 x 8 nodes     */
+    double el = (double)(t1.tv_sec - t0.tv_sec) +
+                (double)(t1.tv_nsec - t0.tv_nsec) / 1e9;
+    /* the bucket must have stalled us at least one refill cycle      */
+    CHECK(el >= 0.15);
+    CHECK(hipGraphExecDestroy(exec) == hipSuccess);
+    printf("PASS graph (%.2fs)\n", el);
+    return 0;
+}
+
 int main(int argc, char **argv) {
     if (argc < 2) {
         fprintf(stderr, "usage: %s quota|oversold|launch|nolimit\n", argv[0]);
@@ -237,5 +272,6 @@ int main(int argc, char **argv) {
     if (strcmp(argv[1], "nolimit") == 0) return scenario_nolimit();
     if (strcmp(argv[1], "fork") == 0) return scenario_fork();
     if (strcmp(argv[1], "cleanup") == 0) return scenario_cleanup();
+    if (strcmp(argv[1], "graph") == 0) return scenario_graph();
     return 2;
 }

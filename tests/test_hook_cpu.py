@@ -36,6 +36,8 @@ def run_scenario(scenario, env_extra):
     ("throttle", {"VGPU_CORE_LIMIT_0": "50"}),
     # fork: child must restart the watcher (atfork re-arm) or it hangs
     ("fork", {"VGPU_CORE_LIMIT_0": "50"}),
+    # graph launches debit the sum of their kernel nodes' grids
+    ("graph", {"VGPU_CORE_LIMIT_0": "50"}),
 ])
 def test_hook_scenario(built_library, scenario, env):
     run_scenario(scenario, env)
