@@ -111,9 +111,17 @@ static bool sample_util(int dev, uint32_t *cont_permille,
                 }
             if (!seq_read_valid(&u->seq, s0)) continue;
             /* KFD reports no per-proc engine time on some stacks;
-             * CU occupancy is the working attribution there         */
-            if (cont == 0 && cont_cus > 0 && h->cu_count > 0)
-                cont = cont_cus * 1000u / (uint32_t)h->cu_count;
+             * CU occupancy is the working attribution there (point
+             * samples -> EWMA, same as the local source)           */
+            if (cont == 0 && cont_cus > 0 && h->cu_count > 0) {
+                uint32_t inst = cont_cus * 1000u /
+                                (uint32_t)h->cu_count;
+                if (inst > 1000) inst = 1000;
+                int32_t delta = (int32_t)inst - (int32_t)h->occ_ema;
+                h->occ_ema = (uint32_t)((int32_t)h->occ_ema +
+                                        delta / 4);
+                cont = h->occ_ema;
+            }
             if (mono_ns() - ts < 1000000000ull) { /* fresh within 1 s */
                 *cont_permille = cont > 1000 ? 1000 : cont;
                 *dev_permille = busy;
@@ -156,13 +164,14 @@ static bool sample_util(int dev, uint32_t *cont_permille,
         }
         h->prev_proc_gfx_ns = gfx_ns;
         h->prev_sample_ns = now;
-        if (cont == 0 && cus > 0 && h->cu_count > 0)
-            cont = cus * 1000u / (uint32_t)h->cu_count;
         if (cont == 0 && h->cu_count > 0) {
-            /* KFD sysfs CU occupancy of our host pids: a point
-             * sample, smoothed with an EWMA (alpha 1/4) across
-             * cycles so sync gaps do not read as idleness          */
-            uint32_t occ = vgpu_kfd_cu_occupancy_sum(&g_state.pids);
+            /* occupancy attribution (amd-smi per-proc cu_occupancy,
+             * else KFD sysfs): POINT samples — a storm workload
+             * reads full occupancy or zero depending on sync phase,
+             * so feed the controller an EWMA, not the raw sample   */
+            uint32_t occ = cus;
+            if (occ == 0)
+                occ = vgpu_kfd_cu_occupancy_sum(&g_state.pids);
             uint32_t inst = occ * 1000u / (uint32_t)h->cu_count;
             if (inst > 1000) inst = 1000;
             int32_t delta = (int32_t)inst - (int32_t)h->occ_ema;

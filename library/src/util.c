@@ -120,7 +120,7 @@ static void dyncfg_init(void) {
         }
     }
     c->aimd_ai_base_div = env_int("VGPU_CU_AIMD_AI_BASE_DIV", 64);
-    c->aimd_deadband_permille = env_int("VGPU_CU_AIMD_DEADBAND_PERMILLE", 20);
+    c->aimd_deadband_permille = env_int("VGPU_CU_AIMD_DEADBAND_PERMILLE", 60);
     c->aimd_md_cooldown = env_int("VGPU_CU_AIMD_MD_COOLDOWN_CYCLES", 2);
     c->auto_debounce_cycles = env_int("VGPU_CU_AUTO_DEBOUNCE_CYCLES", 3);
     c->auto_ext_util_threshold =
