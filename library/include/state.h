@@ -119,7 +119,8 @@ typedef struct {
     uint64_t prev_launch_count;
     int64_t  waiting;           /* launchers parked in the rate limiter */
     uint32_t low_cycles;        /* idle-bypass hysteresis               */
-    uint32_t occ_ema;           /* EWMA of KFD CU-occupancy permille    */
+    uint32_t occ_ema;           /* EWMA of OUR CU-occupancy permille    */
+    uint32_t oth_ema;           /* EWMA of other tenants' occupancy     */
 } dev_hot_t;
 
 /* ---- global library state ---- */

@@ -75,6 +75,9 @@ void vgpu_pid_set_resolve_host(pid_set_t *set);
 bool vgpu_pid_set_translated(const pid_set_t *set);
 /* instantaneous CUs occupied by the set's host pids (KFD sysfs).      */
 uint32_t vgpu_kfd_cu_occupancy_sum(const pid_set_t *set);
+/* ours + other tenants' instantaneous CU occupancy.                   */
+void vgpu_kfd_cu_occupancy2(const pid_set_t *set, uint32_t *ours,
+                            uint32_t *others);
 
 #ifdef __cplusplus
 }

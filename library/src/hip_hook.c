@@ -100,7 +100,7 @@ static bool sample_util(int dev, uint32_t *cont_permille,
             if (s0 & 1u) continue;
             uint32_t busy = u->dev_busy_permille;
             uint64_t ts = u->sample_ns;
-            uint32_t cont = 0, cont_cus = 0;
+            uint32_t cont = 0, cont_cus = 0, other_cus = 0;
             uint32_t n = u->proc_count;
             if (n > MAX_UTIL_PROCS) n = MAX_UTIL_PROCS;
             for (uint32_t i = 0; i < n; i++)
@@ -108,19 +108,24 @@ static bool sample_util(int dev, uint32_t *cont_permille,
                                           u->procs[i].pid)) {
                     cont += u->procs[i].gfx_busy_permille;
                     cont_cus += u->procs[i].cu_occupancy;
+                } else {
+                    other_cus += u->procs[i].cu_occupancy;
                 }
             if (!seq_read_valid(&u->seq, s0)) continue;
-            /* KFD reports no per-proc engine time on some stacks;
-             * CU occupancy is the working attribution there (point
-             * samples -> EWMA, same as the local source)           */
-            if (cont == 0 && cont_cus > 0 && h->cu_count > 0) {
+            /* same sole-tenant/co-tenant policy as the local source */
+            if (cont == 0 && h->cu_count > 0) {
                 uint32_t inst = cont_cus * 1000u /
                                 (uint32_t)h->cu_count;
                 if (inst > 1000) inst = 1000;
-                int32_t delta = (int32_t)inst - (int32_t)h->occ_ema;
+                uint32_t oth = other_cus * 1000u /
+                               (uint32_t)h->cu_count;
+                if (oth > 1000) oth = 1000;
                 h->occ_ema = (uint32_t)((int32_t)h->occ_ema +
-                                        delta / 4);
-                cont = h->occ_ema;
+                    ((int32_t)inst - (int32_t)h->occ_ema) / 4);
+                h->oth_ema = (uint32_t)((int32_t)h->oth_ema +
+                    ((int32_t)oth - (int32_t)h->oth_ema) / 4);
+                if (h->oth_ema >= 20)
+                    cont = h->occ_ema ? h->occ_ema : 1;
             }
             if (mono_ns() - ts < 1000000000ull) { /* fresh within 1 s */
                 *cont_permille = cont > 1000 ? 1000 : cont;
@@ -165,18 +170,25 @@ static bool sample_util(int dev, uint32_t *cont_permille,
         h->prev_proc_gfx_ns = gfx_ns;
         h->prev_sample_ns = now;
         if (cont == 0 && h->cu_count > 0) {
-            /* occupancy attribution (amd-smi per-proc cu_occupancy,
-             * else KFD sysfs): POINT samples — a storm workload
-             * reads full occupancy or zero depending on sync phase,
-             * so feed the controller an EWMA, not the raw sample   */
-            uint32_t occ = cus;
-            if (occ == 0)
-                occ = vgpu_kfd_cu_occupancy_sum(&g_state.pids);
-            uint32_t inst = occ * 1000u / (uint32_t)h->cu_count;
+            /* occupancy attribution: POINT samples (full-or-zero by
+             * sync phase) -> EWMA both OUR occupancy and the other
+             * tenants'.  Sole tenant (others ~0): report cont=0 so
+             * the controller falls back to whole-device busy — the
+             * EXACT signal there.  Co-tenants: our EWMA is the only
+             * attributable signal.                                  */
+            uint32_t ours = 0, others = 0;
+            vgpu_kfd_cu_occupancy2(&g_state.pids, &ours, &others);
+            if (ours < cus) ours = cus; /* amd-smi view as floor     */
+            uint32_t inst = ours * 1000u / (uint32_t)h->cu_count;
             if (inst > 1000) inst = 1000;
-            int32_t delta = (int32_t)inst - (int32_t)h->occ_ema;
-            h->occ_ema = (uint32_t)((int32_t)h->occ_ema + delta / 4);
-            cont = h->occ_ema;
+            uint32_t oth = others * 1000u / (uint32_t)h->cu_count;
+            if (oth > 1000) oth = 1000;
+            h->occ_ema = (uint32_t)((int32_t)h->occ_ema +
+                ((int32_t)inst - (int32_t)h->occ_ema) / 4);
+            h->oth_ema = (uint32_t)((int32_t)h->oth_ema +
+                ((int32_t)oth - (int32_t)h->oth_ema) / 4);
+            if (h->oth_ema >= 20) /* sharing: attribute            */
+                cont = h->occ_ema ? h->occ_ema : 1;
         }
         *cont_permille = cont > 1000 ? 1000 : cont;
         *dev_permille = busy;

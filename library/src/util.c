@@ -416,28 +416,56 @@ void vgpu_pid_set_resolve_host(pid_set_t *set) {
  * GPU; gfx950 reports 0..256).  Point samples — callers smooth with
  * an EWMA.  This is the per-process compute-attribution source on
  * kernels whose amd-smi reports no per-process engine time.          */
+static uint32_t kfd_pid_occupancy(const char *kfd, const char *pid) {
+    char dirp[640];
+    snprintf(dirp, sizeof(dirp), "%s/%s", kfd, pid);
+    DIR *d = opendir(dirp);
+    if (!d) return 0;
+    uint32_t total = 0;
+    struct dirent *e;
+    while ((e = readdir(d)) != NULL) {
+        if (strncmp(e->d_name, "stats_", 6) != 0) continue;
+        char fp[960];
+        snprintf(fp, sizeof(fp), "%s/%s/cu_occupancy", dirp,
+                 e->d_name);
+        FILE *f = fopen(fp, "re");
+        if (!f) continue;
+        unsigned v = 0;
+        if (fscanf(f, "%u", &v) == 1) total += v;
+        fclose(f);
+    }
+    closedir(d);
+    return total;
+}
+
 uint32_t vgpu_kfd_cu_occupancy_sum(const pid_set_t *set) {
+    uint32_t ours = 0, others = 0;
+    vgpu_kfd_cu_occupancy2(set, &ours, &others);
+    return ours;
+}
+
+/* ours + everyone-else's instantaneous CU occupancy (all GPUs the
+ * pids touch).  The "others" figure is what tells a controller
+ * whether it is effectively the sole tenant (use exact whole-device
+ * busy) or sharing (use own attribution).                            */
+void vgpu_kfd_cu_occupancy2(const pid_set_t *set, uint32_t *ours,
+                            uint32_t *others) {
+    *ours = 0;
+    *others = 0;
     char kbuf[256];
     const char *kfd = kfd_proc_root(kbuf, sizeof(kbuf));
-    uint32_t total = 0;
-    for (int i = 0; i < set->host_count; i++) {
-        char dirp[640];
-        snprintf(dirp, sizeof(dirp), "%s/%d", kfd, set->host_pids[i]);
-        DIR *d = opendir(dirp);
-        if (!d) continue;
-        struct dirent *e;
-        while ((e = readdir(d)) != NULL) {
-            if (strncmp(e->d_name, "stats_", 6) != 0) continue;
-            char fp[960];
-            snprintf(fp, sizeof(fp), "%s/%s/cu_occupancy", dirp,
-                     e->d_name);
-            FILE *f = fopen(fp, "re");
-            if (!f) continue;
-            unsigned v = 0;
-            if (fscanf(f, "%u", &v) == 1) total += v;
-            fclose(f);
-        }
-        closedir(d);
+    DIR *d = opendir(kfd);
+    if (!d) return;
+    struct dirent *e;
+    while ((e = readdir(d)) != NULL) {
+        if (e->d_name[0] < '0' || e->d_name[0] > '9') continue;
+        uint32_t v = kfd_pid_occupancy(kfd, e->d_name);
+        if (!v) continue;
+        int32_t pid = (int32_t)atoi(e->d_name);
+        if (vgpu_pid_set_contains(set, pid))
+            *ours += v;
+        else
+            *others += v;
     }
-    return total;
+    closedir(d);
 }
