@@ -416,7 +416,13 @@ void vgpu_pid_set_resolve_host(pid_set_t *set) {
  * GPU; gfx950 reports 0..256).  Point samples — callers smooth with
  * an EWMA.  This is the per-process compute-attribution source on
  * kernels whose amd-smi reports no per-process engine time.          */
-static uint32_t kfd_pid_occupancy(const char *kfd, const char *pid) {
+#define MAX_OWN_GPUS 16
+
+/* occupancy of one pid, optionally filtered to a gpu-id set; when
+ * `collect` is set, the pid's gpu ids are appended to the set.       */
+static uint32_t kfd_pid_occupancy(const char *kfd, const char *pid,
+                                  unsigned *gpu_ids, int *n_gpu_ids,
+                                  int collect) {
     char dirp[640];
     snprintf(dirp, sizeof(dirp), "%s/%s", kfd, pid);
     DIR *d = opendir(dirp);
@@ -425,13 +431,30 @@ static uint32_t kfd_pid_occupancy(const char *kfd, const char *pid) {
     struct dirent *e;
     while ((e = readdir(d)) != NULL) {
         if (strncmp(e->d_name, "stats_", 6) != 0) continue;
+        unsigned gid = (unsigned)strtoul(e->d_name + 6, NULL, 10);
+        if (!collect && gpu_ids && n_gpu_ids) {
+            int mine = 0;
+            for (int i = 0; i < *n_gpu_ids; i++)
+                if (gpu_ids[i] == gid) { mine = 1; break; }
+            if (!mine) continue; /* other tenants on OTHER GPUs do
+                                  * not make us a co-tenant          */
+        }
         char fp[960];
         snprintf(fp, sizeof(fp), "%s/%s/cu_occupancy", dirp,
                  e->d_name);
         FILE *f = fopen(fp, "re");
         if (!f) continue;
         unsigned v = 0;
-        if (fscanf(f, "%u", &v) == 1) total += v;
+        if (fscanf(f, "%u", &v) == 1) {
+            total += v;
+            if (collect && gpu_ids && n_gpu_ids &&
+                *n_gpu_ids < MAX_OWN_GPUS) {
+                int seen = 0;
+                for (int i = 0; i < *n_gpu_ids; i++)
+                    if (gpu_ids[i] == gid) { seen = 1; break; }
+                if (!seen) gpu_ids[(*n_gpu_ids)++] = gid;
+            }
+        }
         fclose(f);
     }
     closedir(d);
@@ -454,18 +477,27 @@ void vgpu_kfd_cu_occupancy2(const pid_set_t *set, uint32_t *ours,
     *others = 0;
     char kbuf[256];
     const char *kfd = kfd_proc_root(kbuf, sizeof(kbuf));
+    /* pass 1: our pids — sum occupancy AND learn which gpu ids are
+     * ours (a pid dir holds stats_<gpuid> only for GPUs it uses)     */
+    unsigned gpu_ids[MAX_OWN_GPUS];
+    int n_gpu_ids = 0;
+    char pidstr[16];
+    for (int i = 0; i < set->host_count; i++) {
+        snprintf(pidstr, sizeof(pidstr), "%d", set->host_pids[i]);
+        *ours += kfd_pid_occupancy(kfd, pidstr, gpu_ids, &n_gpu_ids, 1);
+    }
+    /* even idle we must know our gpus; keep stats dirs as presence   */
+    if (n_gpu_ids == 0) return; /* no context yet: no tenancy info    */
+    /* pass 2: everyone else, counted ONLY on our gpus                */
     DIR *d = opendir(kfd);
     if (!d) return;
     struct dirent *e;
     while ((e = readdir(d)) != NULL) {
         if (e->d_name[0] < '0' || e->d_name[0] > '9') continue;
-        uint32_t v = kfd_pid_occupancy(kfd, e->d_name);
-        if (!v) continue;
         int32_t pid = (int32_t)atoi(e->d_name);
-        if (vgpu_pid_set_contains(set, pid))
-            *ours += v;
-        else
-            *others += v;
+        if (vgpu_pid_set_contains(set, pid)) continue;
+        *others += kfd_pid_occupancy(kfd, e->d_name, gpu_ids,
+                                     &n_gpu_ids, 0);
     }
     closedir(d);
 }

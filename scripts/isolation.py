@@ -74,6 +74,9 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--pods", type=int, default=4)
     ap.add_argument("--limit", type=int, default=25)
+    ap.add_argument("--limits", default=None,
+                    help="per-pod limits, e.g. 10,20,30,40 (asymmetric"
+                         " shares prove the throttle, not contention)")
     ap.add_argument("--seconds", type=float, default=15.0)
     ap.add_argument("--out", default=None)
     args = ap.parse_args()
@@ -94,7 +97,10 @@ def main():
     # their buckets into one container and void the isolation).
     import tempfile
     tdir = tempfile.mkdtemp(prefix="isolation-")
-    procs = [run_pod(args.limit, args.seconds, {
+    limits = ([int(x) for x in args.limits.split(",")]
+              if args.limits else [args.limit] * args.pods)
+    assert len(limits) == args.pods
+    procs = [run_pod(limits[i], args.seconds, {
         "VGPU_SM_NODE_PATH_OVERRIDE":
             os.path.join(tdir, f"sm_node.{i}"),
         "VGPU_VMEM_PATH_OVERRIDE":
@@ -107,11 +113,14 @@ def main():
 
     rates = [r["rate"] for r in results]
     agg = sum(rates)
-    fair = agg / len(rates)
-    share_err = max(abs(r - fair) / fair for r in rates) * 100
+    # error vs the CONFIGURED share of each pod (asymmetric-aware)
+    tot_lim = sum(limits)
+    fair_rates = [agg * l / tot_lim for l in limits]
+    share_err = max(abs(r - f) / f
+                    for r, f in zip(rates, fair_rates)) * 100
     out = {
         "pods": args.pods,
-        "core_limit_pct": args.limit,
+        "core_limit_pct": limits,
         "seconds": args.seconds,
         "unthrottled_single_rate": base["rate"],
         "unthrottled_concurrent_capacity": capacity,
