@@ -59,6 +59,7 @@ typedef struct {
     int     count;
     int     host_count;
     int32_t self_host_pid;  /* VRAM-probe result (smi_self_host_pid)  */
+    int     host_native;    /* pids came from the host side already    */
     uint64_t loaded_ns;
 } pid_set_t;
 

@@ -254,6 +254,7 @@ static int load_pids_from_host_proc(pid_set_t *set) {
 }
 
 bool vgpu_pid_set_translated(const pid_set_t *set) {
+    if (set->host_native) return true; /* pids.config / .host_proc   */
     if (set->self_host_pid > 0) return true;
     for (int i = 0; i < set->host_count; i++)
         if (!sorted_contains(set->pids, set->count,
@@ -265,6 +266,7 @@ bool vgpu_pid_set_translated(const pid_set_t *set) {
 int vgpu_load_pid_set(pid_set_t *set) {
     set->count = 0;
     set->self_host_pid = 0; /* a fork child must re-identify          */
+    set->host_native = 0;
     /* bare-process tenants (no container, shared cgroup): the cgroup
      * walk would lump every co-tenant into one set and void per-pod
      * attribution — self-only scopes it to this process tree root.  */
@@ -279,6 +281,7 @@ int vgpu_load_pid_set(pid_set_t *set) {
     }
     int n = load_pids_from_config(set);
     if (n <= 0) n = load_pids_from_host_proc(set);
+    if (n > 0) set->host_native = 1; /* both sources yield host pids */
     if (n <= 0) n = load_pids_from_cgroup(set);
     if (n < 0) {
         /* last resort: just this process */
