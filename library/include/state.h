@@ -121,6 +121,8 @@ typedef struct {
     uint32_t low_cycles;        /* idle-bypass hysteresis               */
     uint32_t occ_ema;           /* EWMA of OUR CU-occupancy permille    */
     uint32_t oth_ema;           /* EWMA of other tenants' occupancy     */
+    uint32_t attrib_mode;       /* 1 = occupancy attribution active     */
+    uint32_t _rsvd2;
 } dev_hot_t;
 
 /* ---- global library state ---- */

@@ -49,6 +49,13 @@ static inline int cfg_dev(int dev) {
 static int g_watcher_state; /* 0 = not running, 1 = started            */
 static int g_self_probe_tries; /* vram-probe self host-pid attempts    */
 
+/* CU-occupancy sub-sampling: the watcher samples KFD every ~10ms
+ * during its inter-cycle sleep; the per-cycle MEAN is an unbiased
+ * duty-cycle estimate (a single point sample reads full-or-zero
+ * depending on sync phase and whipsaws the controller).              */
+static uint64_t g_occ_sum, g_oth_sum;
+static uint32_t g_occ_n;
+
 static void dev_hot_init(int dev) {
     dev_hot_t *h = &g_state.dev[dev];
     if (h->pool) return;
@@ -170,25 +177,40 @@ static bool sample_util(int dev, uint32_t *cont_permille,
         h->prev_proc_gfx_ns = gfx_ns;
         h->prev_sample_ns = now;
         if (cont == 0 && h->cu_count > 0) {
-            /* occupancy attribution: POINT samples (full-or-zero by
-             * sync phase) -> EWMA both OUR occupancy and the other
-             * tenants'.  Sole tenant (others ~0): report cont=0 so
-             * the controller falls back to whole-device busy — the
-             * EXACT signal there.  Co-tenants: our EWMA is the only
-             * attributable signal.                                  */
-            uint32_t ours = 0, others = 0;
-            vgpu_kfd_cu_occupancy2(&g_state.pids, &ours, &others);
-            if (ours < cus) ours = cus; /* amd-smi view as floor     */
-            uint32_t inst = ours * 1000u / (uint32_t)h->cu_count;
+            /* occupancy attribution from the ~10ms sub-samples the
+             * watcher took during its sleep: the per-cycle mean is
+             * an unbiased duty estimate.  Sole tenant (others ~0):
+             * report cont=0 so the controller uses whole-device
+             * busy — the EXACT signal there.                        */
+            uint32_t inst, oth;
+            if (g_occ_n > 0) {
+                inst = (uint32_t)(g_occ_sum * 1000ull /
+                                  ((uint64_t)g_occ_n *
+                                   (uint64_t)h->cu_count));
+                oth = (uint32_t)(g_oth_sum * 1000ull /
+                                 ((uint64_t)g_occ_n *
+                                  (uint64_t)h->cu_count));
+                g_occ_sum = g_oth_sum = 0;
+                g_occ_n = 0;
+            } else {
+                uint32_t ours = cus, others = 0;
+                vgpu_kfd_cu_occupancy2(&g_state.pids, &ours, &others);
+                if (ours < cus) ours = cus;
+                inst = ours * 1000u / (uint32_t)h->cu_count;
+                oth = others * 1000u / (uint32_t)h->cu_count;
+            }
             if (inst > 1000) inst = 1000;
-            uint32_t oth = others * 1000u / (uint32_t)h->cu_count;
             if (oth > 1000) oth = 1000;
             h->occ_ema = (uint32_t)((int32_t)h->occ_ema +
-                ((int32_t)inst - (int32_t)h->occ_ema) / 4);
+                ((int32_t)inst - (int32_t)h->occ_ema) / 2);
             h->oth_ema = (uint32_t)((int32_t)h->oth_ema +
-                ((int32_t)oth - (int32_t)h->oth_ema) / 4);
-            if (h->oth_ema >= 20) /* sharing: attribute            */
+                ((int32_t)oth - (int32_t)h->oth_ema) / 2);
+            if (h->oth_ema >= 20) { /* sharing: attribute            */
                 cont = h->occ_ema ? h->occ_ema : 1;
+                h->attrib_mode = 1;
+            } else {
+                h->attrib_mode = 0;
+            }
         }
         *cont_permille = cont > 1000 ? 1000 : cont;
         *dev_permille = busy;
@@ -230,8 +252,13 @@ static int64_t ctl_aimd(const dynamic_config_t *c, dev_hot_t *h,
                         int64_t share, uint32_t target, uint32_t obs) {
     /* AIMD with deadband + MD cooldown (reference sm_controller_aimd:
      * naive AIMD sawtooths; the deadband and cooldown remove the
-     * steady-state oscillation that cost +1/3 walltime there).        */
-    uint32_t hi = target + target * c->aimd_deadband_permille / 1000;
+     * steady-state oscillation that cost +1/3 walltime there).
+     * The configured (wide) deadband applies in occupancy-attribution
+     * mode where the signal is inherently noisier; the exact
+     * whole-busy mode keeps a tight 20-permille band.                 */
+    uint32_t db = h->attrib_mode ? (uint32_t)c->aimd_deadband_permille
+                                 : 20u;
+    uint32_t hi = target + target * db / 1000;
     uint32_t lo = (uint32_t)((uint64_t)target * c->aimd_eff_num /
                              c->aimd_eff_den);
     if (h->aimd_cooldown > 0) h->aimd_cooldown--;
@@ -454,9 +481,19 @@ static void *watcher_main(void *arg) {
         uint64_t now = mono_ns();
         if (next <= now + 10000000ull) /* 10ms overrun floor           */
             next = now + 10000000ull;
-        struct timespec ts = {(time_t)((next - now) / 1000000000ull),
-                              (long)((next - now) % 1000000000ull)};
-        nanosleep(&ts, NULL);
+        /* sleep in ~10ms ticks, sampling occupancy each tick          */
+        while ((now = mono_ns()) < next) {
+            uint64_t left = next - now;
+            struct timespec ts = {0, left > 10000000ull
+                                         ? 10000000L
+                                         : (long)left};
+            nanosleep(&ts, NULL);
+            uint32_t ours = 0, others = 0;
+            vgpu_kfd_cu_occupancy2(&g_state.pids, &ours, &others);
+            g_occ_sum += ours;
+            g_oth_sum += others;
+            g_occ_n++;
+        }
     }
     __atomic_store_n(&g_watcher_parked, 1, __ATOMIC_RELEASE);
     return NULL;
