@@ -235,3 +235,30 @@ print("B-OK")
 """
     r = run_py(code_b, env, timeout=300)
     assert r.returncode == 0 and "B-OK" in r.stdout, r.stdout + r.stderr
+
+
+def test_torch_throttle_slows_matmul():
+    """A PyTorch matmul loop under a 20% CU limit must run markedly
+    slower than unthrottled (real-framework throttle evidence)."""
+    code = """
+import time, torch
+assert torch.cuda.is_available()
+a = torch.randn(4096, 4096, device="cuda")
+b = torch.randn(4096, 4096, device="cuda")
+for _ in range(3):
+    (a @ b).sum().item()        # warmup + shim init
+t0 = time.perf_counter()
+for _ in range(40):
+    c = a @ b
+torch.cuda.synchronize()
+print("ELAPSED", time.perf_counter() - t0)
+"""
+    r0 = run_py(code, {"VGPU_PIDS_SELF_ONLY": "1"}, timeout=600)
+    assert r0.returncode == 0, r0.stdout + r0.stderr
+    base = float(r0.stdout.strip().splitlines()[-1].split()[-1])
+    r1 = run_py(code, {"VGPU_CORE_LIMIT_0": "20",
+                       "VGPU_PIDS_SELF_ONLY": "1"}, timeout=600)
+    assert r1.returncode == 0, r1.stdout + r1.stderr
+    lim = float(r1.stdout.strip().splitlines()[-1].split()[-1])
+    assert lim > base * 1.5, \
+        f"torch throttle ineffective: {base:.3f}s -> {lim:.3f}s"
