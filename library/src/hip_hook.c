@@ -362,9 +362,17 @@ static int64_t control_cycle(int dev) {
     int ctl = c->controller == 3 ? 2 : c->controller; /* auto -> aimd  */
     share = ctl == 1 ? ctl_delta(c, h, share, eff_target, obs)
                      : ctl_aimd(c, h, share, eff_target, obs);
-    int64_t smin = h->pool / 100000;
+    /* feedforward bounds: the utilization loop trims INSIDE a band
+     * around the limit-proportional grant.  An attribution failure
+     * can then neither starve a pod below half its share nor let it
+     * free-run past 3x — the error of any misobservation is bounded
+     * by construction (tokens/cycle for eff_target permille):        */
+    int64_t nominal = h->pool * (int64_t)eff_target / 1000 /
+                      (1000 / WATCHER_CYCLE_MS);
+    int64_t smin = nominal / 2;
     if (smin < 1) smin = 1;
-    int64_t smax = h->pool;
+    int64_t smax = nominal * 3;
+    if (smax > h->pool) smax = h->pool;
     if (share < smin) share = smin;
     if (share > smax) share = smax;
     h->cur_share = share;
@@ -464,7 +472,10 @@ static void *watcher_main(void *arg) {
          * per-process attribution still works (max 3 attempts;
          * re-armed for fork children in vgpu_hook_fork_child)       */
         if (!vgpu_pid_set_translated(&g_state.pids) &&
-            g_self_probe_tries < 3) {
+            g_self_probe_tries < 3 &&
+            cycle > ((uint32_t)getpid() & 7u)) {
+            /* pid-staggered start so simultaneously-launched sibling
+             * pods do not probe in the same instant                 */
             g_self_probe_tries++;
             int32_t hp = smi_self_host_pid(0);
             if (hp > 0 && hp != (int32_t)getpid())

@@ -349,11 +349,15 @@ int32_t smi_self_host_pid(int dev) {
         return 0;
     if (nb > MAXP) nb = MAXP;
 
-    /* distinctive size: 48 MiB + pid-salted page count               */
-    size_t probe = (48u << 20) + ((size_t)(getpid() & 0xff) << 12);
+    /* distinctive size: 2 MiB salt spacing >> the 1 MiB match
+     * tolerance, so CONCURRENT probes by sibling pods (adjacent
+     * pids) can never alias each other                              */
+    size_t probe = (16u << 20) +
+                   ((size_t)((unsigned)getpid() % 61u) << 21);
     void *p = NULL;
     if (real_hip.hipMalloc(&p, probe) != hipSuccess || !p) return 0;
 
+    const uint64_t tol = 1u << 20;
     int32_t found = 0;
     for (int attempt = 0; attempt < 10 && !found; attempt++) {
         struct timespec ts = {0, 50000000L}; /* 50ms: let smi refresh */
@@ -376,9 +380,7 @@ int32_t smi_self_host_pid(int dev) {
             uint64_t now = after[i].memory_usage.vram_mem;
             if (now > prev) {
                 uint64_t delta = now - prev;
-                /* within 4 MiB of the probe (allocator granularity)  */
-                if (delta >= probe - (4u << 20) &&
-                    delta <= probe + (4u << 20)) {
+                if (delta >= probe - tol && delta <= probe + tol) {
                     candidates++;
                     cand = (int32_t)after[i].pid;
                 }
@@ -387,6 +389,37 @@ int32_t smi_self_host_pid(int dev) {
         if (candidates == 1) found = cand;
     }
     real_hip.hipFree(p);
+    if (found) {
+        /* verification: after the free, the candidate's VRAM must
+         * drop back by ~probe — rejects coincidental growth         */
+        int verified = 0;
+        for (int attempt = 0; attempt < 6 && !verified; attempt++) {
+            struct timespec ts = {0, 50000000L};
+            nanosleep(&ts, NULL);
+            na = MAXP;
+            memset(after, 0, sizeof(after));
+            if (real_smi.amdsmi_get_gpu_process_list(h, &na, after) !=
+                AMDSMI_STATUS_SUCCESS)
+                break;
+            if (na > MAXP) na = MAXP;
+            for (uint32_t j = 0; j < nb; j++) {
+                if (before[j].pid != (uint32_t)found) continue;
+                uint64_t base = before[j].memory_usage.vram_mem;
+                for (uint32_t i = 0; i < na; i++)
+                    if (after[i].pid == (uint32_t)found &&
+                        after[i].memory_usage.vram_mem <= base + tol)
+                        verified = 1;
+            }
+            /* a pid absent from `before` (new process = us) counts
+             * as verified once its vram is back near the probe-free
+             * level; approximate: accept after one recheck          */
+            int in_before = 0;
+            for (uint32_t j = 0; j < nb; j++)
+                if (before[j].pid == (uint32_t)found) in_before = 1;
+            if (!in_before) verified = 1;
+        }
+        if (!verified) found = 0;
+    }
     if (found)
         LOGGER(LOG_INFO, "self host pid identified by vram probe: %d",
                found);
