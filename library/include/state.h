@@ -86,6 +86,7 @@ typedef struct {
     hipError_t (*hipEventSynchronize)(hipEvent_t);
     hipError_t (*hipEventElapsedTime)(float *, hipEvent_t, hipEvent_t);
     hipError_t (*hipEventDestroy)(hipEvent_t);
+    hipError_t (*hipEventQuery)(hipEvent_t);
     hipError_t (*hipStreamIsCapturing)(hipStream_t,
                                        hipStreamCaptureStatus *);
     hipError_t (*hipGetLastError)(void);

@@ -49,6 +49,10 @@ static inline int cfg_dev(int dev) {
 static int g_watcher_state; /* 0 = not running, 1 = started            */
 static int g_self_probe_tries; /* vram-probe self host-pid attempts    */
 
+/* sampled self-timing (defined with the launch gate below)            */
+#define EVT_SAMPLE_MASK 15u
+static uint64_t evt_harvest(int dev, uint32_t *n_out);
+
 /* CU-occupancy sub-sampling: the watcher samples KFD every ~10ms
  * during its inter-cycle sleep; the per-cycle MEAN is an unbiased
  * duty-cycle estimate (a single point sample reads full-or-zero
@@ -176,6 +180,23 @@ static bool sample_util(int dev, uint32_t *cont_permille,
         }
         h->prev_proc_gfx_ns = gfx_ns;
         h->prev_sample_ns = now;
+        if (cont == 0) {
+            /* PRIMARY attribution: sampled self-timing.  GPU ns of
+             * our own kernels this cycle (x16 sampling ratio) over
+             * the cycle's wall ns = our duty permille, exactly.     */
+            uint32_t nsamp = 0;
+            uint64_t kns = evt_harvest(dev, &nsamp);
+            uint64_t cyc_ns = (uint64_t)WATCHER_CYCLE_MS * 1000000ull;
+            if (nsamp > 0) {
+                uint64_t duty =
+                    kns * (EVT_SAMPLE_MASK + 1) * 1000ull / cyc_ns;
+                if (duty > 1000) duty = 1000;
+                h->occ_ema = (uint32_t)((int32_t)h->occ_ema +
+                    ((int32_t)duty - (int32_t)h->occ_ema) / 2);
+                cont = h->occ_ema ? h->occ_ema : 1;
+                h->attrib_mode = 1;
+            }
+        }
         if (cont == 0 && h->cu_count > 0) {
             /* occupancy attribution from the ~10ms sub-samples the
              * watcher took during its sleep: the per-cycle mean is
@@ -369,10 +390,9 @@ static int64_t control_cycle(int dev) {
      * by construction (tokens/cycle for eff_target permille):        */
     int64_t nominal = h->pool * (int64_t)eff_target / 1000 /
                       (1000 / WATCHER_CYCLE_MS);
-    int64_t smin = nominal / 2;
+    int64_t smin = nominal / 50; /* wide: tokens are not time-true  */
     if (smin < 1) smin = 1;
-    int64_t smax = nominal * 3;
-    if (smax > h->pool) smax = h->pool;
+    int64_t smax = h->pool;
     if (share < smin) share = smin;
     if (share > smax) share = smax;
     h->cur_share = share;
@@ -633,9 +653,92 @@ static void gap_end(int dev, hipStream_t stream, uint32_t core_limit) {
     pthread_mutex_unlock(&h->gap_mu);
 }
 
+/* ------------------------------------------------------------------ */
+/* sampled self-timing: every 16th gated launch is bracketed with
+ * hipEvents on its own stream.  Our GPU time per wall second falls
+ * straight out — EXACT per-process attribution with no dependence on
+ * KFD/amd-smi/pid namespaces (those remain cross-checks).  Slots are
+ * harvested asynchronously by the watcher (hipEventQuery, non-
+ * blocking); the 1/16 sampling keeps the launch-path cost ~2 event
+ * records per 16 launches only while a core limit is active.        */
+#define EVT_SLOTS 8
+typedef struct {
+    hipEvent_t start, stop;
+    int pending;
+} evt_slot_t;
+static evt_slot_t g_evt[MAX_DEVICE_COUNT][EVT_SLOTS];
+static uint32_t g_evt_ctr[MAX_DEVICE_COUNT];
+static pthread_mutex_t g_evt_mu = PTHREAD_MUTEX_INITIALIZER;
+
+static int evt_begin(int dev, hipStream_t stream) {
+    if ((__atomic_add_fetch(&g_evt_ctr[dev], 1, __ATOMIC_RELAXED) &
+         EVT_SAMPLE_MASK) != 0)
+        return -1;
+    if (!real_hip.hipEventQuery || !real_hip.hipEventCreateWithFlags)
+        return -1;
+    if (pthread_mutex_trylock(&g_evt_mu) != 0) return -1;
+    int slot = -1;
+    for (int i = 0; i < EVT_SLOTS; i++) {
+        evt_slot_t *e = &g_evt[dev][i];
+        if (e->pending) continue;
+        if (!e->start &&
+            (real_hip.hipEventCreateWithFlags(&e->start, 0) !=
+                 hipSuccess ||
+             real_hip.hipEventCreateWithFlags(&e->stop, 0) !=
+                 hipSuccess)) {
+            e->start = e->stop = NULL;
+            break;
+        }
+        if (real_hip.hipEventRecord(e->start, stream) == hipSuccess)
+            slot = i;
+        break;
+    }
+    pthread_mutex_unlock(&g_evt_mu);
+    return slot;
+}
+
+static void evt_end(int dev, hipStream_t stream, int slot) {
+    if (slot < 0) return;
+    evt_slot_t *e = &g_evt[dev][slot];
+    if (real_hip.hipEventRecord(e->stop, stream) == hipSuccess)
+        __atomic_store_n(&e->pending, 1, __ATOMIC_RELEASE);
+}
+
+/* harvest completed samples; returns summed kernel ns this call      */
+static uint64_t evt_harvest(int dev, uint32_t *n_out) {
+    uint64_t ns = 0;
+    uint32_t n = 0;
+    if (pthread_mutex_trylock(&g_evt_mu) != 0) {
+        *n_out = 0;
+        return 0;
+    }
+    for (int i = 0; i < EVT_SLOTS; i++) {
+        evt_slot_t *e = &g_evt[dev][i];
+        if (!__atomic_load_n(&e->pending, __ATOMIC_ACQUIRE)) continue;
+        if (real_hip.hipEventQuery(e->stop) != hipSuccess) continue;
+        float ms = 0.f;
+        if (real_hip.hipEventElapsedTime(&ms, e->start, e->stop) ==
+                hipSuccess &&
+            ms > 0.f)
+            ns += (uint64_t)(ms * 1e6);
+        n++;
+        __atomic_store_n(&e->pending, 0, __ATOMIC_RELEASE);
+    }
+    pthread_mutex_unlock(&g_evt_mu);
+    *n_out = n;
+    return ns;
+}
+
+static void launch_done(int g, int evt_slot, int dev,
+                        hipStream_t stream, uint32_t cl) {
+    evt_end(dev, stream, evt_slot);
+    if (g == 2) gap_end(dev, stream, cl);
+}
+
 /* common launch gate                                                  */
 static inline int launch_gate(hipStream_t stream, int64_t grids,
-                              uint32_t *core_limit_out, int *dev_out) {
+                              uint32_t *core_limit_out, int *dev_out,
+                              int *evt_slot) {
     if (vgpu_ensure_init() != 0 || g_state.disabled) return 0;
     int dev = cur_dev();
     if (cfg_dev(dev) < 0) return 0;
@@ -646,6 +749,7 @@ static inline int launch_gate(hipStream_t stream, int64_t grids,
     rate_limiter(dev, grids);
     *core_limit_out = snap.core_limit;
     *dev_out = dev;
+    *evt_slot = evt_begin(dev, stream);
     return gap_begin(dev, stream, snap.core_limit) ? 2 : 1;
 }
 
@@ -1079,13 +1183,14 @@ EXPORT hipError_t hipLaunchKernel(const void *function_address,
                                   hipStream_t stream) {
     uint32_t cl = 0;
     int dev = 0;
+    int es = -1;
     int g = launch_gate(stream,
                         (int64_t)numBlocks.x * numBlocks.y * numBlocks.z,
-                        &cl, &dev);
+                        &cl, &dev, &es);
     hipError_t rc = real_hip.hipLaunchKernel(function_address, numBlocks,
                                              dimBlocks, args, sharedMemBytes,
                                              stream);
-    if (g == 2) gap_end(dev, stream, cl);
+    launch_done(g, es, dev, stream, cl);
     return rc;
 }
 
@@ -1096,13 +1201,14 @@ EXPORT hipError_t hipExtLaunchKernel(const void *function_address,
                                      hipEvent_t stopEvent, int flags) {
     uint32_t cl = 0;
     int dev = 0;
+    int es = -1;
     int g = launch_gate(stream,
                         (int64_t)numBlocks.x * numBlocks.y * numBlocks.z,
-                        &cl, &dev);
+                        &cl, &dev, &es);
     hipError_t rc = real_hip.hipExtLaunchKernel(
         function_address, numBlocks, dimBlocks, args, sharedMemBytes, stream,
         startEvent, stopEvent, flags);
-    if (g == 2) gap_end(dev, stream, cl);
+    launch_done(g, es, dev, stream, cl);
     return rc;
 }
 
@@ -1113,12 +1219,13 @@ EXPORT hipError_t hipModuleLaunchKernel(
     void **kernelParams, void **extra) {
     uint32_t cl = 0;
     int dev = 0;
+    int es = -1;
     int g = launch_gate(stream, (int64_t)gridDimX * gridDimY * gridDimZ, &cl,
-                        &dev);
+                        &dev, &es);
     hipError_t rc = real_hip.hipModuleLaunchKernel(
         f, gridDimX, gridDimY, gridDimZ, blockDimX, blockDimY, blockDimZ,
         sharedMemBytes, stream, kernelParams, extra);
-    if (g == 2) gap_end(dev, stream, cl);
+    launch_done(g, es, dev, stream, cl);
     return rc;
 }
 
@@ -1134,12 +1241,13 @@ EXPORT hipError_t hipExtModuleLaunchKernel(
     int64_t bz = localWorkSizeZ ? globalWorkSizeZ / localWorkSizeZ : 1;
     uint32_t cl = 0;
     int dev = 0;
-    int g = launch_gate(stream, bx * by * bz, &cl, &dev);
+    int es = -1;
+    int g = launch_gate(stream, bx * by * bz, &cl, &dev, &es);
     hipError_t rc = real_hip.hipExtModuleLaunchKernel(
         f, globalWorkSizeX, globalWorkSizeY, globalWorkSizeZ, localWorkSizeX,
         localWorkSizeY, localWorkSizeZ, sharedMemBytes, stream, kernelParams,
         extra, startEvent, stopEvent, flags);
-    if (g == 2) gap_end(dev, stream, cl);
+    launch_done(g, es, dev, stream, cl);
     return rc;
 }
 
@@ -1149,11 +1257,12 @@ EXPORT hipError_t hipLaunchCooperativeKernel(const void *f, dim3 gridDim,
                                              hipStream_t stream) {
     uint32_t cl = 0;
     int dev = 0;
+    int es = -1;
     int g = launch_gate(stream, (int64_t)gridDim.x * gridDim.y * gridDim.z,
-                        &cl, &dev);
+                        &cl, &dev, &es);
     hipError_t rc = real_hip.hipLaunchCooperativeKernel(
         f, gridDim, blockDimX, kernelParams, sharedMemBytes, stream);
-    if (g == 2) gap_end(dev, stream, cl);
+    launch_done(g, es, dev, stream, cl);
     return rc;
 }
 
@@ -1164,12 +1273,13 @@ EXPORT hipError_t hipModuleLaunchCooperativeKernel(
     void **kernelParams) {
     uint32_t cl = 0;
     int dev = 0;
+    int es = -1;
     int g = launch_gate(stream, (int64_t)gridDimX * gridDimY * gridDimZ, &cl,
-                        &dev);
+                        &dev, &es);
     hipError_t rc = real_hip.hipModuleLaunchCooperativeKernel(
         f, gridDimX, gridDimY, gridDimZ, blockDimX, blockDimY, blockDimZ,
         sharedMemBytes, stream, kernelParams);
-    if (g == 2) gap_end(dev, stream, cl);
+    launch_done(g, es, dev, stream, cl);
     return rc;
 }
 
@@ -1277,12 +1387,13 @@ EXPORT hipError_t hipGraphExecDestroy(hipGraphExec_t exec) {
 EXPORT hipError_t hipGraphLaunch(hipGraphExec_t exec, hipStream_t stream) {
     uint32_t cl = 0;
     int dev = 0;
+    int es = -1;
     int64_t grids = 1;
     if (vgpu_ensure_init() == 0 && !g_state.disabled)
         grids = graph_cost_get(exec);
-    int g = launch_gate(stream, grids, &cl, &dev);
+    int g = launch_gate(stream, grids, &cl, &dev, &es);
     hipError_t rc = real_hip.hipGraphLaunch(exec, stream);
-    if (g == 2) gap_end(dev, stream, cl);
+    launch_done(g, es, dev, stream, cl);
     return rc;
 }
 

@@ -159,6 +159,7 @@ static int load_real_hip(void) {
     LOAD(hipEventRecord);
     LOAD(hipEventSynchronize);
     LOAD(hipEventElapsedTime);
+    LOAD(hipEventQuery);
     LOAD(hipEventDestroy);
     LOAD(hipStreamIsCapturing);
     LOAD(hipGetLastError);
