@@ -170,6 +170,10 @@ EXPORT hipError_t hipEventSynchronize(hipEvent_t e) {
     (void)e;
     return hipSuccess;
 }
+EXPORT hipError_t hipEventQuery(hipEvent_t e) {
+    (void)e;
+    return hipSuccess;
+}
 EXPORT hipError_t hipEventElapsedTime(float *ms, hipEvent_t a, hipEvent_t b) {
     (void)a; (void)b;
     *ms = 0.f;
