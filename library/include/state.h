@@ -124,6 +124,8 @@ typedef struct {
     uint32_t oth_ema;           /* EWMA of other tenants' occupancy     */
     uint32_t attrib_mode;       /* 1 = occupancy attribution active     */
     uint32_t _rsvd2;
+    uint64_t evt_mean_ns;       /* EWMA of sampled kernel duration      */
+    uint64_t evt_prev_launches; /* launch counter at last estimation    */
 } dev_hot_t;
 
 /* ---- global library state ---- */

@@ -181,15 +181,27 @@ static bool sample_util(int dev, uint32_t *cont_permille,
         h->prev_proc_gfx_ns = gfx_ns;
         h->prev_sample_ns = now;
         if (cont == 0) {
-            /* PRIMARY attribution: sampled self-timing.  GPU ns of
-             * our own kernels this cycle (x16 sampling ratio) over
-             * the cycle's wall ns = our duty permille, exactly.     */
+            /* PRIMARY attribution: sampled self-timing.  The mean
+             * sampled kernel duration x the EXACT number of gated
+             * launches this cycle = our GPU ns, unbiased regardless
+             * of how many samples landed in the cycle.              */
             uint32_t nsamp = 0;
             uint64_t kns = evt_harvest(dev, &nsamp);
-            uint64_t cyc_ns = (uint64_t)WATCHER_CYCLE_MS * 1000000ull;
             if (nsamp > 0) {
+                uint64_t mean = kns / nsamp;
+                h->evt_mean_ns = h->evt_mean_ns
+                                     ? (h->evt_mean_ns + mean) / 2
+                                     : mean;
+            }
+            uint64_t launches =
+                __atomic_load_n(&h->launch_count, __ATOMIC_RELAXED);
+            uint64_t dl = launches - h->evt_prev_launches;
+            h->evt_prev_launches = launches;
+            if (h->evt_mean_ns > 0) {
+                uint64_t cyc_ns =
+                    (uint64_t)WATCHER_CYCLE_MS * 1000000ull;
                 uint64_t duty =
-                    kns * (EVT_SAMPLE_MASK + 1) * 1000ull / cyc_ns;
+                    h->evt_mean_ns * dl * 1000ull / cyc_ns;
                 if (duty > 1000) duty = 1000;
                 h->occ_ema = (uint32_t)((int32_t)h->occ_ema +
                     ((int32_t)duty - (int32_t)h->occ_ema) / 2);
