@@ -688,6 +688,14 @@ static int evt_begin(int dev, hipStream_t stream) {
         return -1;
     if (!real_hip.hipEventQuery || !real_hip.hipEventCreateWithFlags)
         return -1;
+    /* never record sampling events into a graph capture: they would
+     * be replayed (wrong data) and the slot would pend forever       */
+    if (real_hip.hipStreamIsCapturing) {
+        hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+        if (real_hip.hipStreamIsCapturing(stream, &st) == hipSuccess &&
+            st != hipStreamCaptureStatusNone)
+            return -1;
+    }
     if (pthread_mutex_trylock(&g_evt_mu) != 0) return -1;
     int slot = -1;
     for (int i = 0; i < EVT_SLOTS; i++) {
