@@ -180,41 +180,18 @@ static bool sample_util(int dev, uint32_t *cont_permille,
         }
         h->prev_proc_gfx_ns = gfx_ns;
         h->prev_sample_ns = now;
-        if (cont == 0) {
-            /* PRIMARY attribution: sampled self-timing.  The mean
-             * sampled kernel duration x the EXACT number of gated
-             * launches this cycle = our GPU ns, unbiased regardless
-             * of how many samples landed in the cycle.              */
-            uint32_t nsamp = 0;
-            uint64_t kns = evt_harvest(dev, &nsamp);
-            if (nsamp > 0) {
-                uint64_t mean = kns / nsamp;
-                h->evt_mean_ns = h->evt_mean_ns
-                                     ? (h->evt_mean_ns + mean) / 2
-                                     : mean;
-            }
-            uint64_t launches =
-                __atomic_load_n(&h->launch_count, __ATOMIC_RELAXED);
-            uint64_t dl = launches - h->evt_prev_launches;
-            h->evt_prev_launches = launches;
-            if (h->evt_mean_ns > 0) {
-                uint64_t cyc_ns =
-                    (uint64_t)WATCHER_CYCLE_MS * 1000000ull;
-                uint64_t duty =
-                    h->evt_mean_ns * dl * 1000ull / cyc_ns;
-                if (duty > 1000) duty = 1000;
-                h->occ_ema = (uint32_t)((int32_t)h->occ_ema +
-                    ((int32_t)duty - (int32_t)h->occ_ema) / 2);
-                cont = h->occ_ema ? h->occ_ema : 1;
-                h->attrib_mode = 1;
-            }
-        }
         if (cont == 0 && h->cu_count > 0) {
-            /* occupancy attribution from the ~10ms sub-samples the
-             * watcher took during its sleep: the per-cycle mean is
-             * an unbiased duty estimate.  Sole tenant (others ~0):
-             * report cont=0 so the controller uses whole-device
-             * busy — the EXACT signal there.                        */
+            /* Attribution ladder, measured on MI355X:
+             *  - CO-TENANT: our share of GPU compute = mean CU
+             *    occupancy over the cycle (10ms sub-samples the
+             *    watcher took during its sleep).  Contention-
+             *    correct: a CU belongs to exactly one wave.
+             *  - SOLE TENANT: sampled self-timing — mean event-
+             *    timed kernel duration x the exact gated-launch
+             *    count = our duty; falls back to whole-device busy
+             *    (also exact when alone).  Event wall-durations are
+             *    NOT used when sharing: contention stretches them
+             *    past our real share.                              */
             uint32_t inst, oth;
             if (g_occ_n > 0) {
                 inst = (uint32_t)(g_occ_sum * 1000ull /
@@ -238,11 +215,34 @@ static bool sample_util(int dev, uint32_t *cont_permille,
                 ((int32_t)inst - (int32_t)h->occ_ema) / 2);
             h->oth_ema = (uint32_t)((int32_t)h->oth_ema +
                 ((int32_t)oth - (int32_t)h->oth_ema) / 2);
-            if (h->oth_ema >= 20) { /* sharing: attribute            */
+
+            /* keep the self-timing estimator warm in either mode   */
+            uint32_t nsamp = 0;
+            uint64_t kns = evt_harvest(dev, &nsamp);
+            if (nsamp > 0) {
+                uint64_t mean = kns / nsamp;
+                h->evt_mean_ns = h->evt_mean_ns
+                                     ? (h->evt_mean_ns + mean) / 2
+                                     : mean;
+            }
+            uint64_t launches =
+                __atomic_load_n(&h->launch_count, __ATOMIC_RELAXED);
+            uint64_t dl = launches - h->evt_prev_launches;
+            h->evt_prev_launches = launches;
+
+            if (h->oth_ema >= 20) { /* sharing: occupancy share     */
                 cont = h->occ_ema ? h->occ_ema : 1;
                 h->attrib_mode = 1;
-            } else {
+            } else if (h->evt_mean_ns > 0) { /* alone: self-timing  */
+                uint64_t cyc_ns =
+                    (uint64_t)WATCHER_CYCLE_MS * 1000000ull;
+                uint64_t duty =
+                    h->evt_mean_ns * dl * 1000ull / cyc_ns;
+                if (duty > 1000) duty = 1000;
+                cont = (uint32_t)duty ? (uint32_t)duty : 1;
                 h->attrib_mode = 0;
+            } else {
+                h->attrib_mode = 0; /* busy fallback in control     */
             }
         }
         *cont_permille = cont > 1000 ? 1000 : cont;
