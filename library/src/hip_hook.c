@@ -135,8 +135,14 @@ static bool sample_util(int dev, uint32_t *cont_permille,
                     ((int32_t)inst - (int32_t)h->occ_ema) / 4);
                 h->oth_ema = (uint32_t)((int32_t)h->oth_ema +
                     ((int32_t)oth - (int32_t)h->oth_ema) / 4);
-                if (h->oth_ema >= 20)
-                    cont = h->occ_ema ? h->occ_ema : 1;
+                if (h->oth_ema >= 20) {
+                    uint64_t denom = h->occ_ema + h->oth_ema;
+                    uint32_t c = denom
+                        ? (uint32_t)((uint64_t)busy * h->occ_ema /
+                                     denom)
+                        : 0;
+                    cont = c ? c : 1;
+                }
             }
             if (mono_ns() - ts < 1000000000ull) { /* fresh within 1 s */
                 *cont_permille = cont > 1000 ? 1000 : cont;
@@ -230,8 +236,17 @@ static bool sample_util(int dev, uint32_t *cont_permille,
             uint64_t dl = launches - h->evt_prev_launches;
             h->evt_prev_launches = launches;
 
-            if (h->oth_ema >= 20) { /* sharing: occupancy share     */
-                cont = h->occ_ema ? h->occ_ema : 1;
+            if (h->oth_ema >= 20) {
+                /* sharing: our FRACTION of total residency times the
+                 * whole-device duty.  cu_occupancy counts residency
+                 * (co-resident waves double-count), so the absolute
+                 * figure overreads under sharing — the ratio does
+                 * not, and `busy` supplies the time normalization. */
+                uint64_t denom = h->occ_ema + h->oth_ema;
+                uint32_t c = denom ? (uint32_t)((uint64_t)busy *
+                                                h->occ_ema / denom)
+                                   : 0;
+                cont = c ? c : 1;
                 h->attrib_mode = 1;
             } else if (h->evt_mean_ns > 0) { /* alone: self-timing  */
                 uint64_t cyc_ns =
