@@ -64,6 +64,10 @@ typedef struct {
                                                    unsigned int, unsigned int,
                                                    unsigned int, unsigned int,
                                                    hipStream_t, void **);
+    hipError_t (*hipLaunchKernelExC)(const hipLaunchConfig_t *,
+                                     const void *, void **);
+    hipError_t (*hipDrvLaunchKernelEx)(const HIP_LAUNCH_CONFIG *,
+                                       hipFunction_t, void **, void **);
 
     hipError_t (*hipGraphLaunch)(hipGraphExec_t, hipStream_t);
     hipError_t (*hipGraphInstantiate)(hipGraphExec_t *, hipGraph_t,

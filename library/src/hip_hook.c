@@ -1318,6 +1318,44 @@ EXPORT hipError_t hipModuleLaunchCooperativeKernel(
     return rc;
 }
 
+EXPORT hipError_t hipLaunchKernelExC(const hipLaunchConfig_t *config,
+                                     const void *fPtr, void **args) {
+    /* the extensible launch used by modern ROCm PyTorch — without
+     * this hook torch tunnels under the token bucket entirely       */
+    uint32_t cl = 0;
+    int dev = 0;
+    int es = -1;
+    hipStream_t stream = config ? config->stream : NULL;
+    int64_t grids = config ? (int64_t)config->gridDim.x *
+                                 config->gridDim.y * config->gridDim.z
+                           : 1;
+    int g = launch_gate(stream, grids, &cl, &dev, &es);
+    hipError_t rc = real_hip.hipLaunchKernelExC
+                        ? real_hip.hipLaunchKernelExC(config, fPtr, args)
+                        : hipErrorNotSupported;
+    launch_done(g, es, dev, stream, cl);
+    return rc;
+}
+
+EXPORT hipError_t hipDrvLaunchKernelEx(const HIP_LAUNCH_CONFIG *config,
+                                       hipFunction_t f, void **params,
+                                       void **extra) {
+    uint32_t cl = 0;
+    int dev = 0;
+    int es = -1;
+    hipStream_t stream = config ? config->hStream : NULL;
+    int64_t grids = config ? (int64_t)config->gridDimX *
+                                 config->gridDimY * config->gridDimZ
+                           : 1;
+    int g = launch_gate(stream, grids, &cl, &dev, &es);
+    hipError_t rc = real_hip.hipDrvLaunchKernelEx
+                        ? real_hip.hipDrvLaunchKernelEx(config, f,
+                                                        params, extra)
+                        : hipErrorNotSupported;
+    launch_done(g, es, dev, stream, cl);
+    return rc;
+}
+
 /* ------------------------------------------------------------------ */
 /* graph cost accounting                                               */
 /* ------------------------------------------------------------------ */

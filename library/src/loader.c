@@ -143,6 +143,8 @@ static int load_real_hip(void) {
     LOAD(hipExtModuleLaunchKernel);
     LOAD(hipLaunchCooperativeKernel);
     LOAD(hipModuleLaunchCooperativeKernel);
+    LOAD(hipLaunchKernelExC);
+    LOAD(hipDrvLaunchKernelEx);
     LOAD(hipGraphLaunch);
     LOAD(hipGraphInstantiate);
     LOAD(hipGraphInstantiateWithFlags);
