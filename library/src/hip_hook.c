@@ -276,6 +276,65 @@ static bool sample_util(int dev, uint32_t *cont_permille,
         }
         return true;
     }
+
+    /* 4. no amd-smi at all (it can fail to load inside some runtimes,
+     * e.g. under torch): the KFD occupancy sub-samples and the
+     * hipEvent self-timing need NOTHING from amd-smi, so attribution
+     * still works — only whole-device busy is approximated.          */
+    if (h->cu_count > 0) {
+        uint32_t inst = 0, oth = 0;
+        if (g_occ_n > 0) {
+            inst = (uint32_t)(g_occ_sum * 1000ull /
+                              ((uint64_t)g_occ_n *
+                               (uint64_t)h->cu_count));
+            oth = (uint32_t)(g_oth_sum * 1000ull /
+                             ((uint64_t)g_occ_n *
+                              (uint64_t)h->cu_count));
+            g_occ_sum = g_oth_sum = 0;
+            g_occ_n = 0;
+        }
+        if (inst > 1000) inst = 1000;
+        if (oth > 1000) oth = 1000;
+        h->occ_ema = (uint32_t)((int32_t)h->occ_ema +
+            ((int32_t)inst - (int32_t)h->occ_ema) / 2);
+        h->oth_ema = (uint32_t)((int32_t)h->oth_ema +
+            ((int32_t)oth - (int32_t)h->oth_ema) / 2);
+
+        uint32_t nsamp = 0;
+        uint64_t kns = evt_harvest(dev, &nsamp);
+        if (nsamp > 0) {
+            uint64_t mean = kns / nsamp;
+            h->evt_mean_ns = h->evt_mean_ns
+                                 ? (h->evt_mean_ns + mean) / 2
+                                 : mean;
+        }
+        uint64_t launches =
+            __atomic_load_n(&h->launch_count, __ATOMIC_RELAXED);
+        uint64_t dl = launches - h->evt_prev_launches;
+        h->evt_prev_launches = launches;
+
+        if (h->oth_ema >= 20) {
+            /* sharing without a busy reading: assume the GPU is
+             * saturated (conservative) and take our residency ratio */
+            uint64_t denom = h->occ_ema + h->oth_ema;
+            uint32_t c = denom ? (uint32_t)(1000ull * h->occ_ema /
+                                            denom)
+                               : 0;
+            *cont_permille = c ? c : 1;
+            *dev_permille = 1000;
+            h->attrib_mode = 1;
+            return true;
+        }
+        if (h->evt_mean_ns > 0) {
+            uint64_t cyc_ns = (uint64_t)WATCHER_CYCLE_MS * 1000000ull;
+            uint64_t duty = h->evt_mean_ns * dl * 1000ull / cyc_ns;
+            if (duty > 1000) duty = 1000;
+            *cont_permille = (uint32_t)duty ? (uint32_t)duty : 1;
+            *dev_permille = (uint32_t)duty;
+            h->attrib_mode = 0;
+            return true;
+        }
+    }
     return false;
 }
 
