@@ -240,6 +240,9 @@ print("B-OK")
 def test_torch_throttle_slows_matmul():
     """A PyTorch matmul loop under a 20% CU limit must run markedly
     slower than unthrottled (real-framework throttle evidence)."""
+    # rate over a fixed window: big-tile GEMMs are token-cheap, so
+    # the throttle binds only after the utilization loop converges
+    # (~1s); the window includes an untimed convergence phase
     code = """
 import time, torch
 assert torch.cuda.is_available()
@@ -248,12 +251,16 @@ b = torch.randn(4096, 4096, device="cuda")
 for _ in range(3):
     (a @ b).sum().item()        # warmup + shim init
 t0 = time.perf_counter()
-for i in range(600):
+while time.perf_counter() - t0 < 3.0:   # convergence phase
     c = a @ b
-    if i % 20 == 19:
-        torch.cuda.synchronize()   # keep the queue shallow
-torch.cuda.synchronize()
-print("ELAPSED", time.perf_counter() - t0)
+    torch.cuda.synchronize()
+n = 0
+t0 = time.perf_counter()
+while time.perf_counter() - t0 < 4.0:   # measured phase
+    c = a @ b
+    torch.cuda.synchronize()
+    n += 1
+print("RATE", n / (time.perf_counter() - t0))
 """
     r0 = run_py(code, {"VGPU_PIDS_SELF_ONLY": "1"}, timeout=600)
     assert r0.returncode == 0, r0.stdout + r0.stderr
@@ -262,5 +269,5 @@ print("ELAPSED", time.perf_counter() - t0)
                        "VGPU_PIDS_SELF_ONLY": "1"}, timeout=600)
     assert r1.returncode == 0, r1.stdout + r1.stderr
     lim = float(r1.stdout.strip().splitlines()[-1].split()[-1])
-    assert lim > base * 1.5, \
-        f"torch throttle ineffective: {base:.3f}s -> {lim:.3f}s"
+    assert lim < base * 0.6, \
+        f"torch throttle ineffective: {base:.1f}/s -> {lim:.1f}/s"
