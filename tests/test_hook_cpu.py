@@ -148,3 +148,44 @@ def test_pid_set_from_host_proc(built_library, tmp_path):
     assert r.returncode == 0, r.stdout + r.stderr
     # both spellings of the pod uid matched; foreign pids excluded
     assert "host-view: 2 pids" in r.stderr, r.stderr
+
+
+def test_runtime_limit_mutation_under_load(built_library, tmp_path):
+    """Cross-language seqlock LIVE: the Python control plane mutates
+    the core limit thousands of times while the C shim's launch path
+    and watcher read snapshots — no torn read may crash or wedge the
+    workload (reference resource_data_seqlock_versioning_design)."""
+    import subprocess as sp
+    import time as _t
+
+    from vgpu_manager_amd.config.regions import (
+        DeviceLimit,
+        VgpuConfigWriter,
+    )
+    p = str(tmp_path / "vgpu.config")
+    w = VgpuConfigWriter(p)
+    w.write(pod_uid="u", pod_name="p", pod_namespace="n",
+            container_name="c",
+            limits=[DeviceLimit(uuid="GPU-x", host_index=0,
+                                memory_bytes=1 << 30, core_limit=50)])
+
+    build = os.path.join(LIB_DIR, "build")
+    env = dict(os.environ)
+    env.update({
+        "VGPU_CONFIG_PATH_OVERRIDE": p,
+        "LD_PRELOAD": os.path.join(build, "libvgpu-control.so"),
+        "LD_LIBRARY_PATH": os.path.join(build, "stub"),
+        "VGPU_REAL_HIP_PATH": os.path.join(build, "stub",
+                                           "libamdhip64.so.7"),
+    })
+    proc = sp.Popen([os.path.join(build, "test_hook_cpu"), "throttle"],
+                    env=env, stdout=sp.PIPE, stderr=sp.PIPE, text=True)
+    # churn the limit while the C side runs
+    n = 0
+    while proc.poll() is None and n < 200000:
+        w.modify_device(0, core_limit=30 + (n % 5) * 10)
+        n += 1
+    out, err = proc.communicate(timeout=120)
+    w.close()
+    assert n > 1000, f"churn loop barely ran ({n})"
+    assert proc.returncode == 0, out + err
