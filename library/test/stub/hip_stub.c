@@ -287,3 +287,40 @@ EXPORT hipError_t hipGraphLaunch(hipGraphExec_t exec, hipStream_t s) {
     __atomic_fetch_add(&c_launch, (uint64_t)g->n_nodes, __ATOMIC_RELAXED);
     return hipSuccess;
 }
+
+EXPORT hipError_t hipMallocPitch(void **ptr, size_t *pitch, size_t width,
+                                 size_t height) {
+    *pitch = (width + 255u) & ~(size_t)255u;
+    *ptr = malloc(*pitch * height);
+    if (*ptr) __atomic_fetch_add(&c_malloc, 1, __ATOMIC_RELAXED);
+    return *ptr ? hipSuccess : hipErrorOutOfMemory;
+}
+
+EXPORT hipError_t hipMalloc3D(hipPitchedPtr *p, hipExtent extent) {
+    size_t pitch = (extent.width + 255u) & ~(size_t)255u;
+    void *mem = malloc(pitch * extent.height * extent.depth);
+    if (!mem) return hipErrorOutOfMemory;
+    __atomic_fetch_add(&c_malloc, 1, __ATOMIC_RELAXED);
+    p->ptr = mem;
+    p->pitch = pitch;
+    p->xsize = extent.width;
+    p->ysize = extent.height;
+    return hipSuccess;
+}
+
+EXPORT hipError_t hipMallocArray(hipArray_t *array,
+                                 const hipChannelFormatDesc *desc,
+                                 size_t width, size_t height,
+                                 unsigned int flags) {
+    (void)desc; (void)flags;
+    *array = (hipArray_t)malloc(width * (height ? height : 1) * 16);
+    if (*array) __atomic_fetch_add(&c_malloc, 1, __ATOMIC_RELAXED);
+    return *array ? hipSuccess : hipErrorOutOfMemory;
+}
+
+EXPORT hipError_t hipFreeArray(hipArray_t array) {
+    free(array);
+    __atomic_fetch_add(&c_free, 1, __ATOMIC_RELAXED);
+    return hipSuccess;
+}
+

@@ -260,6 +260,54 @@ static int scenario_graph(void) {
     return 0;
 }
 
+static int scenario_variants(void) {
+    /* env: VGPU_MEM_LIMIT_0=1m.  Every alloc variant must charge the
+     * quota and every matching free must retire it exactly (reference
+     * test_alloc_pitch / arrays / async accounting).                 */
+    size_t freeb = 0, total = 0;
+    CHECK(hipMemGetInfo(&freeb, &total) == hipSuccess);
+    CHECK(freeb == total);
+
+    void *p = NULL;
+    size_t pitch = 0;
+    CHECK(hipMallocPitch(&p, &pitch, 1000, 200) == hipSuccess);
+    CHECK(pitch >= 1000);
+    size_t after_pitch = 0;
+    CHECK(hipMemGetInfo(&after_pitch, &total) == hipSuccess);
+    CHECK(total - after_pitch >= pitch * 200);
+
+    hipPitchedPtr pp;
+    hipExtent ext = {512, 16, 2};
+    CHECK(hipMalloc3D(&pp, ext) == hipSuccess);
+
+    hipArray_t arr = NULL;
+    hipChannelFormatDesc desc;
+    memset(&desc, 0, sizeof(desc));
+    desc.x = 32;
+    CHECK(hipMallocArray(&arr, &desc, 1024, 32, 0) == hipSuccess);
+
+    void *ap = NULL;
+    CHECK(hipMallocAsync(&ap, 100 * 1024, NULL) == hipSuccess);
+
+    size_t mid = 0;
+    CHECK(hipMemGetInfo(&mid, &total) == hipSuccess);
+    CHECK(mid < after_pitch); /* every variant charged               */
+
+    /* over-quota pitch alloc must OOM (not silently pass)           */
+    void *big = NULL;
+    size_t bp = 0;
+    CHECK(hipMallocPitch(&big, &bp, 1 << 20, 2) == hipErrorOutOfMemory);
+
+    CHECK(hipFree(p) == hipSuccess);
+    CHECK(hipFree(pp.ptr) == hipSuccess);
+    CHECK(hipFreeArray(arr) == hipSuccess);
+    CHECK(hipFreeAsync(ap, NULL) == hipSuccess);
+    CHECK(hipMemGetInfo(&freeb, &total) == hipSuccess);
+    CHECK(freeb == total); /* all charges retired                    */
+    printf("PASS variants\n");
+    return 0;
+}
+
 int main(int argc, char **argv) {
     if (argc < 2) {
         fprintf(stderr, "usage: %s quota|oversold|launch|nolimit\n", argv[0]);
@@ -273,5 +321,6 @@ int main(int argc, char **argv) {
     if (strcmp(argv[1], "fork") == 0) return scenario_fork();
     if (strcmp(argv[1], "cleanup") == 0) return scenario_cleanup();
     if (strcmp(argv[1], "graph") == 0) return scenario_graph();
+    if (strcmp(argv[1], "variants") == 0) return scenario_variants();
     return 2;
 }

@@ -38,6 +38,9 @@ def run_scenario(scenario, env_extra):
     ("fork", {"VGPU_CORE_LIMIT_0": "50"}),
     # graph launches debit the sum of their kernel nodes' grids
     ("graph", {"VGPU_CORE_LIMIT_0": "50"}),
+    # pitch/3D/array/async variants all charge and retire the quota
+    ("variants", {"VGPU_MEM_LIMIT_0": "1m",
+                  "VGPU_MEM_ACCOUNT_MODE": "ledger"}),
 ])
 def test_hook_scenario(built_library, scenario, env):
     run_scenario(scenario, env)
