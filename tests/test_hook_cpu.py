@@ -7,6 +7,7 @@ the token-bucket throttle without a GPU.
 """
 import os
 import subprocess
+import sys
 
 import pytest
 
@@ -192,3 +193,39 @@ def test_runtime_limit_mutation_under_load(built_library, tmp_path):
     w.close()
     assert n > 1000, f"churn loop barely ran ({n})"
     assert proc.returncode == 0, out + err
+
+
+def test_dlsym_route_ctypes_path(built_library, tmp_path):
+    """ctypes users resolve hip symbols via dlopen+dlsym, not link-time
+    binding — the shim's exported dlsym hook must route them to the
+    hooks (reference test_dlsym_hijack / getproc routing)."""
+    import subprocess as sp
+    build = os.path.join(LIB_DIR, "build")
+    code = """
+import ctypes
+lib = ctypes.CDLL("libamdhip64.so.7")   # dlopen -> our dlsym hook
+lib.hipMalloc.restype = ctypes.c_int
+lib.hipMalloc.argtypes = [ctypes.c_void_p, ctypes.c_size_t]
+lib.hipMemGetInfo.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+p = ctypes.c_void_p()
+rc = lib.hipMalloc(ctypes.byref(p), 512 * 1024)
+assert rc == 0, rc
+rc = lib.hipMalloc(ctypes.byref(p), 2 * 1024 * 1024)  # over 1m quota
+assert rc == 2, f"expected hipErrorOutOfMemory via dlsym route, got {rc}"
+free = ctypes.c_size_t(); total = ctypes.c_size_t()
+assert lib.hipMemGetInfo(ctypes.byref(free), ctypes.byref(total)) == 0
+assert total.value == 1024 * 1024, total.value   # spoofed view
+print("DLSYM-OK")
+"""
+    env = dict(os.environ)
+    env.update({
+        "VGPU_MEM_LIMIT_0": "1m", "VGPU_MEM_ACCOUNT_MODE": "ledger",
+        "LD_PRELOAD": os.path.join(build, "libvgpu-control.so"),
+        "LD_LIBRARY_PATH": os.path.join(build, "stub"),
+        "VGPU_REAL_HIP_PATH": os.path.join(build, "stub",
+                                           "libamdhip64.so.7"),
+    })
+    r = sp.run([sys.executable, "-c", code], env=env,
+               capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "DLSYM-OK" in r.stdout
