@@ -640,11 +640,24 @@ void vgpu_hook_fork_child(void) {
     g_self_probe_tries = 0; /* the child is a NEW host pid            */
     __atomic_store_n(&g_watcher_parked, 0, __ATOMIC_RELEASE);
     __atomic_store_n(&g_shutdown, 0, __ATOMIC_RELEASE);
+    g_occ_sum = g_oth_sum = 0;
+    g_occ_n = 0;
     for (int i = 0; i < MAX_DEVICE_COUNT; i++) {
         g_state.dev[i].pool = 0;
         g_state.dev[i].waiting = 0;
         g_state.dev[i].throttled = 0;
+        g_state.dev[i].occ_ema = 0;
+        g_state.dev[i].oth_ema = 0;
+        g_state.dev[i].evt_mean_ns = 0;
+        g_state.dev[i].evt_prev_launches = 0;
+        /* the parent's hipEvent handles are not valid in the child   */
+        for (int j = 0; j < EVT_SLOTS; j++) {
+            g_evt[i][j].start = g_evt[i][j].stop = NULL;
+            g_evt[i][j].pending = 0;
+        }
+        g_evt_ctr[i] = 0;
     }
+    pthread_mutex_init(&g_evt_mu, NULL);
 }
 
 static void rate_limiter(int dev, int64_t grids) {
