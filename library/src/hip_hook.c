@@ -49,8 +49,16 @@ static inline int cfg_dev(int dev) {
 static int g_watcher_state; /* 0 = not running, 1 = started            */
 static int g_self_probe_tries; /* vram-probe self host-pid attempts    */
 
-/* sampled self-timing (defined with the launch gate below)            */
+/* sampled self-timing (logic with the launch gate below)              */
 #define EVT_SAMPLE_MASK 15u
+#define EVT_SLOTS 8
+typedef struct {
+    hipEvent_t start, stop;
+    int pending;
+} evt_slot_t;
+static evt_slot_t g_evt[MAX_DEVICE_COUNT][EVT_SLOTS];
+static uint32_t g_evt_ctr[MAX_DEVICE_COUNT];
+static pthread_mutex_t g_evt_mu = PTHREAD_MUTEX_INITIALIZER;
 static uint64_t evt_harvest(int dev, uint32_t *n_out);
 
 /* CU-occupancy sub-sampling: the watcher samples KFD every ~10ms
@@ -760,15 +768,6 @@ static void gap_end(int dev, hipStream_t stream, uint32_t core_limit) {
  * harvested asynchronously by the watcher (hipEventQuery, non-
  * blocking); the 1/16 sampling keeps the launch-path cost ~2 event
  * records per 16 launches only while a core limit is active.        */
-#define EVT_SLOTS 8
-typedef struct {
-    hipEvent_t start, stop;
-    int pending;
-} evt_slot_t;
-static evt_slot_t g_evt[MAX_DEVICE_COUNT][EVT_SLOTS];
-static uint32_t g_evt_ctr[MAX_DEVICE_COUNT];
-static pthread_mutex_t g_evt_mu = PTHREAD_MUTEX_INITIALIZER;
-
 static int evt_begin(int dev, hipStream_t stream) {
     if ((__atomic_add_fetch(&g_evt_ctr[dev], 1, __ATOMIC_RELAXED) &
          EVT_SAMPLE_MASK) != 0)
