@@ -308,6 +308,46 @@ static int scenario_variants(void) {
     return 0;
 }
 
+static int scenario_sharedbucket(void) {
+    /* env: VGPU_CORE_LIMIT_0=50 VGPU_SM_NODE_PATH_OVERRIDE=<tmp>.
+     * Two processes of one "container" share ONE token bucket: the
+     * refill election must pick a single owner per cycle and both
+     * processes' storms must drain the SAME bucket (the multi-
+     * process over-supply problem the sm_node design solves).       */
+    struct timespec t0, t1;
+    clock_gettime(CLOCK_MONOTONIC, &t0);
+    dim3 grid = {16384, 1, 1}, block = {256, 1, 1};
+    /* parent primes the region + watcher                             */
+    for (int i = 0; i < 5; i++)
+        CHECK(hipLaunchKernel((void *)scenario_sharedbucket, grid, block,
+                              NULL, 0, NULL) == hipSuccess);
+    pid_t kids[2];
+    for (int k = 0; k < 2; k++) {
+        kids[k] = fork();
+        CHECK(kids[k] >= 0);
+        if (kids[k] == 0) {
+            alarm(60);
+            for (int i = 0; i < 100; i++)
+                if (hipLaunchKernel((void *)scenario_sharedbucket, grid,
+                                    block, NULL, 0, NULL) != hipSuccess)
+                    _exit(1);
+            _exit(0);
+        }
+    }
+    for (int k = 0; k < 2; k++) {
+        int st = 0;
+        CHECK(waitpid(kids[k], &st, 0) == kids[k]);
+        CHECK(WIFEXITED(st) && WEXITSTATUS(st) == 0);
+    }
+    clock_gettime(CLOCK_MONOTONIC, &t1);
+    double el = (double)(t1.tv_sec - t0.tv_sec) +
+                (double)(t1.tv_nsec - t0.tv_nsec) / 1e9;
+    /* 205 x 16384 grids through one bucket needs several refills     */
+    CHECK(el >= 0.15);
+    printf("PASS sharedbucket (%.2fs)\n", el);
+    return 0;
+}
+
 int main(int argc, char **argv) {
     if (argc < 2) {
         fprintf(stderr, "usage: %s quota|oversold|launch|nolimit\n", argv[0]);
@@ -322,5 +362,7 @@ int main(int argc, char **argv) {
     if (strcmp(argv[1], "cleanup") == 0) return scenario_cleanup();
     if (strcmp(argv[1], "graph") == 0) return scenario_graph();
     if (strcmp(argv[1], "variants") == 0) return scenario_variants();
+    if (strcmp(argv[1], "sharedbucket") == 0)
+        return scenario_sharedbucket();
     return 2;
 }

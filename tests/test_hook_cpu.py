@@ -47,6 +47,15 @@ def test_hook_scenario(built_library, scenario, env):
     run_scenario(scenario, env)
 
 
+def test_shared_bucket_two_processes(built_library, tmp_path):
+    # one container bucket shared by two forked processes (refill
+    # election + common drain), reference sm_node design
+    run_scenario("sharedbucket", {
+        "VGPU_CORE_LIMIT_0": "50",
+        "VGPU_SM_NODE_PATH_OVERRIDE": str(tmp_path / "sm_node.config"),
+    })
+
+
 def test_hook_cleanup_and_sweep(built_library, tmp_path):
     # shared vmem region: normal exit retires charges; _exit leaks a
     # spill record that the ledger-full sweep reclaims (dead pid)
