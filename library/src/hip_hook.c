@@ -131,6 +131,9 @@ static bool sample_util(int dev, uint32_t *cont_permille,
                     other_cus += u->procs[i].cu_occupancy;
                 }
             if (!seq_read_valid(&u->seq, s0)) continue;
+            if (mono_ns() - ts >= 1000000000ull)
+                break; /* stale region: do NOT touch the EMAs — the
+                        * local source owns them this cycle          */
             /* same sole-tenant/co-tenant policy as the local source */
             if (cont == 0 && h->cu_count > 0) {
                 uint32_t inst = cont_cus * 1000u /
@@ -152,12 +155,9 @@ static bool sample_util(int dev, uint32_t *cont_permille,
                     cont = c ? c : 1;
                 }
             }
-            if (mono_ns() - ts < 1000000000ull) { /* fresh within 1 s */
-                *cont_permille = cont > 1000 ? 1000 : cont;
-                *dev_permille = busy;
-                return true;
-            }
-            break;
+            *cont_permille = cont > 1000 ? 1000 : cont;
+            *dev_permille = busy;
+            return true;
         }
     }
 
