@@ -786,7 +786,7 @@ static int evt_begin(int dev, hipStream_t stream) {
     int slot = -1;
     for (int i = 0; i < EVT_SLOTS; i++) {
         evt_slot_t *e = &g_evt[dev][i];
-        if (e->pending) continue;
+        if (e->pending) continue; /* 1 = awaiting harvest, 2 = claimed */
         if (!e->start &&
             (real_hip.hipEventCreateWithFlags(&e->start, 0) !=
                  hipSuccess ||
@@ -795,8 +795,12 @@ static int evt_begin(int dev, hipStream_t stream) {
             e->start = e->stop = NULL;
             break;
         }
-        if (real_hip.hipEventRecord(e->start, stream) == hipSuccess)
+        if (real_hip.hipEventRecord(e->start, stream) == hipSuccess) {
+            /* claim until evt_end records the stop event — another
+             * thread between begin and end must not reuse the slot  */
+            e->pending = 2;
             slot = i;
+        }
         break;
     }
     pthread_mutex_unlock(&g_evt_mu);
@@ -808,6 +812,8 @@ static void evt_end(int dev, hipStream_t stream, int slot) {
     evt_slot_t *e = &g_evt[dev][slot];
     if (real_hip.hipEventRecord(e->stop, stream) == hipSuccess)
         __atomic_store_n(&e->pending, 1, __ATOMIC_RELEASE);
+    else
+        __atomic_store_n(&e->pending, 0, __ATOMIC_RELEASE);
 }
 
 /* harvest completed samples; returns summed kernel ns this call      */
@@ -820,7 +826,8 @@ static uint64_t evt_harvest(int dev, uint32_t *n_out) {
     }
     for (int i = 0; i < EVT_SLOTS; i++) {
         evt_slot_t *e = &g_evt[dev][i];
-        if (!__atomic_load_n(&e->pending, __ATOMIC_ACQUIRE)) continue;
+        if (__atomic_load_n(&e->pending, __ATOMIC_ACQUIRE) != 1)
+            continue; /* 0 free, 2 claimed (stop not recorded yet)   */
         if (real_hip.hipEventQuery(e->stop) != hipSuccess) continue;
         float ms = 0.f;
         if (real_hip.hipEventElapsedTime(&ms, e->start, e->stop) ==
