@@ -324,3 +324,20 @@ EXPORT hipError_t hipFreeArray(hipArray_t array) {
     return hipSuccess;
 }
 
+
+/* minimal GetProcAddress: returns the STUB's own entry points.  The
+ * shim's hook must REPLACE these with hook pointers — the getproc
+ * routing test proves that by observing quota enforcement through the
+ * returned pointer (the raw stub pointer would not enforce).         */
+EXPORT hipError_t hipGetProcAddress(const char *symbol, void **pfn,
+                                    int hipVersion, uint64_t flags,
+                                    hipDriverProcAddressQueryResult *res) {
+    (void)hipVersion; (void)flags; (void)res;
+    if (strcmp(symbol, "hipMalloc") == 0)
+        *pfn = (void *)&hipMalloc;
+    else if (strcmp(symbol, "hipMemGetInfo") == 0)
+        *pfn = (void *)&hipMemGetInfo;
+    else
+        *pfn = NULL;
+    return *pfn ? hipSuccess : hipErrorNotSupported;
+}

@@ -348,6 +348,30 @@ static int scenario_sharedbucket(void) {
     return 0;
 }
 
+static int scenario_getproc(void) {
+    /* env: VGPU_MEM_LIMIT_0=1m.  hipGetProcAddress must hand back the
+     * HOOK, not the raw runtime entry — the pointer we get must
+     * enforce the quota (reference cuGetProcAddress routing).        */
+    typedef hipError_t (*malloc_fn)(void **, size_t);
+    typedef hipError_t (*meminfo_fn)(size_t *, size_t *);
+    void *pfn = NULL;
+    CHECK(hipGetProcAddress("hipMalloc", &pfn, 0, 0, NULL) ==
+          hipSuccess);
+    CHECK(pfn != NULL);
+    malloc_fn mal = (malloc_fn)pfn;
+    void *a = NULL;
+    CHECK(mal(&a, 512 * 1024) == hipSuccess);
+    CHECK(mal(&a, 2 * 1024 * 1024) == hipErrorOutOfMemory);
+    CHECK(hipGetProcAddress("hipMemGetInfo", &pfn, 0, 0, NULL) ==
+          hipSuccess);
+    meminfo_fn mi = (meminfo_fn)pfn;
+    size_t freeb = 0, total = 0;
+    CHECK(mi(&freeb, &total) == hipSuccess);
+    CHECK(total == 1024 * 1024); /* spoofed view through getproc      */
+    printf("PASS getproc\n");
+    return 0;
+}
+
 int main(int argc, char **argv) {
     if (argc < 2) {
         fprintf(stderr, "usage: %s quota|oversold|launch|nolimit\n", argv[0]);
@@ -364,5 +388,6 @@ int main(int argc, char **argv) {
     if (strcmp(argv[1], "variants") == 0) return scenario_variants();
     if (strcmp(argv[1], "sharedbucket") == 0)
         return scenario_sharedbucket();
+    if (strcmp(argv[1], "getproc") == 0) return scenario_getproc();
     return 2;
 }

@@ -42,6 +42,10 @@ def run_scenario(scenario, env_extra):
     # pitch/3D/array/async variants all charge and retire the quota
     ("variants", {"VGPU_MEM_LIMIT_0": "1m",
                   "VGPU_MEM_ACCOUNT_MODE": "ledger"}),
+    # hipGetProcAddress must return hook pointers (quota enforced
+    # through the returned function)
+    ("getproc", {"VGPU_MEM_LIMIT_0": "1m",
+                 "VGPU_MEM_ACCOUNT_MODE": "ledger"}),
 ])
 def test_hook_scenario(built_library, scenario, env):
     run_scenario(scenario, env)
