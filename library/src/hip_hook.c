@@ -1006,6 +1006,13 @@ EXPORT hipError_t hipExtMallocWithFlags(void **ptr, size_t size,
     if (route < 0) return hipErrorOutOfMemory;
     if (route == 1) return managed_spill(dev, ptr, size, VMEM_KIND_SYNC);
     hipError_t rc = real_hip.hipExtMallocWithFlags(ptr, size, flags);
+    if (rc == hipErrorOutOfMemory &&
+        (g_state.cfg->oversold ||
+         (vgpu_device_flags(dev) & DEV_FLAG_OVERSOLD))) {
+        malloc_done(lockfd);
+        if (real_hip.hipGetLastError) real_hip.hipGetLastError();
+        return managed_spill(dev, ptr, size, VMEM_KIND_SYNC);
+    }
     if (rc == hipSuccess) {
         dev_hooked_add(dev, (int64_t)size);
         alloc_registry_add(*ptr, size, ALLOC_KIND_DEVICE, dev, -1, NULL);
@@ -1054,6 +1061,14 @@ EXPORT hipError_t hipMallocAsync(void **ptr, size_t size,
         return managed_spill(dev, ptr, size, vkind);
     }
     hipError_t rc = real_hip.hipMallocAsync(ptr, size, stream);
+    if (rc == hipErrorOutOfMemory &&
+        (g_state.cfg->oversold ||
+         (vgpu_device_flags(dev) & DEV_FLAG_OVERSOLD))) {
+        /* real HBM exhausted under oversold: spill like hipMalloc   */
+        malloc_done(lockfd);
+        if (real_hip.hipGetLastError) real_hip.hipGetLastError();
+        return managed_spill(dev, ptr, size, vkind);
+    }
     if (rc == hipSuccess) {
         dev_hooked_add(dev, (int64_t)size);
         alloc_registry_add(*ptr, size, kind, dev, -1, NULL);
