@@ -444,7 +444,11 @@ static int64_t control_cycle(int dev) {
                   snap.soft_core_limit > snap.core_limit &&
                   g_state.cfg->compute_policy == COMPUTE_POLICY_BALANCE;
     if (soft_on) {
-        uint32_t others = busy > obs ? busy - obs : 0;
+        /* others from the TRUE attribution (cont), never from obs:
+         * the busy-when-active fallback substitutes whole-device
+         * busy into obs, which would make a co-tenant's load look
+         * like ours and grant exclusivity wrongly                   */
+        uint32_t others = busy > cont ? busy - cont : 0;
         int exclusive = others < (uint32_t)c->auto_ext_util_threshold;
         if (exclusive != (int)h->excl_state) {
             if (++h->debounce >= c->auto_debounce_cycles) {
@@ -477,6 +481,16 @@ static int64_t control_cycle(int dev) {
     int ctl = c->controller == 3 ? 2 : c->controller; /* auto -> aimd  */
     share = ctl == 1 ? ctl_delta(c, h, share, eff_target, obs)
                      : ctl_aimd(c, h, share, eff_target, obs);
+    /* elastic soft ramp (reference SOFT_ADJUST elastic design): once
+     * the exclusivity FSM raises the target, approach the soft share
+     * geometrically — the AI trickle alone would take minutes.  The
+     * controllers still pull DOWN the moment a co-tenant appears.    */
+    if (soft_on && h->excl_state) {
+        int64_t soft_share = h->pool * (int64_t)eff_target / 1000 /
+                             (1000 / WATCHER_CYCLE_MS);
+        if (share < soft_share)
+            share += (soft_share - share) / 4 + 1;
+    }
     /* feedforward bounds: the utilization loop trims INSIDE a band
      * around the limit-proportional grant.  An attribution failure
      * can then neither starve a pod below half its share nor let it

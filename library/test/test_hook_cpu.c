@@ -372,6 +372,21 @@ static int scenario_getproc(void) {
     return 0;
 }
 
+static int scenario_storm(void) {
+    /* neutral storm: run and report elapsed; callers compare regimes */
+    struct timespec t0, t1;
+    clock_gettime(CLOCK_MONOTONIC, &t0);
+    dim3 grid = {16384, 1, 1}, block = {256, 1, 1};
+    for (int i = 0; i < 200; i++)
+        CHECK(hipLaunchKernel((void *)scenario_storm, grid, block, NULL,
+                              0, NULL) == hipSuccess);
+    clock_gettime(CLOCK_MONOTONIC, &t1);
+    double el = (double)(t1.tv_sec - t0.tv_sec) +
+                (double)(t1.tv_nsec - t0.tv_nsec) / 1e9;
+    printf("PASS storm elapsed=%.3f\n", el);
+    return 0;
+}
+
 int main(int argc, char **argv) {
     if (argc < 2) {
         fprintf(stderr, "usage: %s quota|oversold|launch|nolimit\n", argv[0]);
@@ -389,5 +404,6 @@ int main(int argc, char **argv) {
     if (strcmp(argv[1], "sharedbucket") == 0)
         return scenario_sharedbucket();
     if (strcmp(argv[1], "getproc") == 0) return scenario_getproc();
+    if (strcmp(argv[1], "storm") == 0) return scenario_storm();
     return 2;
 }

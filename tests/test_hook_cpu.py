@@ -242,3 +242,62 @@ print("DLSYM-OK")
                capture_output=True, text=True, timeout=120)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "DLSYM-OK" in r.stdout
+
+
+def test_soft_limit_exclusive_burst(built_library, tmp_path):
+    """compute-policy `balance`: when the external watcher shows the
+    GPU otherwise idle, the exclusivity FSM raises the target to the
+    soft limit and the same storm completes much faster than when a
+    co-tenant is visible (reference soft-limit elastic design)."""
+    import subprocess as sp
+    import threading
+    import time as _t
+
+    from vgpu_manager_amd.config.regions import UtilRegionWriter
+
+    region = str(tmp_path / "sm_util.config")
+    writer = UtilRegionWriter(region, device_count=1)
+    busy_box = {"v": 0}
+    stop = threading.Event()
+
+    def feed():
+        while not stop.is_set():
+            writer.publish(0, dev_busy_permille=busy_box["v"],
+                           vram_used_bytes=0, procs=[])
+            stop.wait(0.04)
+
+    t = threading.Thread(target=feed, daemon=True)
+    t.start()
+    try:
+        def run_storm(extra):
+            build = os.path.join(LIB_DIR, "build")
+            env = dict(os.environ)
+            env.update({
+                "VGPU_CORE_LIMIT_0": "20",
+                "VGPU_CORE_SOFT_LIMIT_0": "100",
+                "VGPU_COMPUTE_POLICY": "balance",
+                "VGPU_UTIL_PATH_OVERRIDE": region,
+                "LD_PRELOAD": os.path.join(build,
+                                           "libvgpu-control.so"),
+                "LD_LIBRARY_PATH": os.path.join(build, "stub"),
+                "VGPU_REAL_HIP_PATH": os.path.join(
+                    build, "stub", "libamdhip64.so.7"),
+            })
+            env.update(extra)
+            r = sp.run([os.path.join(build, "test_hook_cpu"), "storm"],
+                       env=env, capture_output=True, text=True,
+                       timeout=180)
+            assert r.returncode == 0, r.stdout + r.stderr
+            return float(r.stdout.split("elapsed=")[1].split()[0])
+
+        busy_box["v"] = 600          # a co-tenant is burning the GPU
+        t_shared = run_storm({})
+        busy_box["v"] = 0            # exclusive: burst to soft limit
+        t_exclusive = run_storm({})
+        assert t_exclusive * 2 < t_shared, (
+            f"exclusive burst ineffective: shared={t_shared:.2f}s "
+            f"exclusive={t_exclusive:.2f}s")
+    finally:
+        stop.set()
+        t.join(timeout=2)
+        writer.close()
