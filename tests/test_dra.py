@@ -457,3 +457,66 @@ def test_dynamic_cpx_busy_gpu_fails_claim(tmp_path):
     assert state.prepared_claims() == []
     assert be.modes[0] == "SPX"
     assert state.partition_manager.holders(0) == set()
+
+
+def test_dra_grpc_prepare_with_sharing_strategy(state, tmp_path):
+    """claim-wide time-slicing config flows through resolve ->
+    prepare -> written core limits (the full gRPC-path shape)."""
+    import grpc as _grpc
+    from vgpu_manager_amd.client.kube import FakeKubeClient
+    from vgpu_manager_amd.dra import api as dapi
+    from vgpu_manager_amd.dra.driver import DraDriver, DraDriverServer
+
+    client = FakeKubeClient()
+    claim = {
+        "metadata": {"name": "c2", "namespace": "default",
+                     "uid": "uid-ts"},
+        "status": {"allocation": {"devices": {
+            "results": [
+                {"request": "a", "driver": "manager.amd.com",
+                 "pool": "node-a", "device": "GPU-fake-0000"},
+                {"request": "b", "driver": "manager.amd.com",
+                 "pool": "node-a", "device": "GPU-fake-0000"},
+            ],
+            "config": [
+                {"requests": ["a"], "opaque": {"parameters": {
+                    "partitionKey": "cont-a"}}},
+                {"requests": ["b"], "opaque": {"parameters": {
+                    "partitionKey": "cont-b"}}},
+                {"opaque": {"parameters": {
+                    "strategy": "time-slicing",
+                    "percents": [70, 30]}}},
+            ],
+        }}},
+    }
+    client.add_resource_claim(claim)
+    endpoint = str(tmp_path / "p" / "dra.sock")
+    driver = DraDriver(state, client, endpoint=endpoint)
+    server = DraDriverServer(driver,
+                             plugins_dir=str(tmp_path / "p"),
+                             plugins_registry=str(tmp_path / "r"))
+    server.start()
+    try:
+        ch = _grpc.insecure_channel(f"unix://{endpoint}")
+        prep = ch.unary_unary(
+            f"/{dapi.DRA_SERVICE}/NodePrepareResources",
+            request_serializer=lambda m: m.encode(),
+            response_deserializer=dapi.NodePrepareResourcesResponse
+            .decode)
+        resp = prep(dapi.NodePrepareResourcesRequest(claims=[
+            dapi.Claim(namespace="default", uid="uid-ts", name="c2")]),
+            timeout=10)
+        assert resp.claims[0].value.error == ""
+        base = state.checkpoint.claims["uid-ts"]["container_dir"]
+        from vgpu_manager_amd.config.regions import VgpuConfigReader
+        snap_a = VgpuConfigReader(
+            os.path.join(base, "cont-a", "config",
+                         "vgpu.config")).snapshot()
+        snap_b = VgpuConfigReader(
+            os.path.join(base, "cont-b", "config",
+                         "vgpu.config")).snapshot()
+        assert snap_a["devices"][0]["core_limit"] == 70
+        assert snap_b["devices"][0]["core_limit"] == 30
+        ch.close()
+    finally:
+        server.stop()
