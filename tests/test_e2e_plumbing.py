@@ -135,3 +135,39 @@ def test_second_pod_rejected_when_full(tmp_path):
     res = f.filter({"Pod": p2, "NodeNames": ["gpu-node-1"]})
     assert res["NodeNames"] == []
     assert "gpu-node-1" in res["FailedNodes"]
+
+
+def test_topology_annotation_roundtrip_through_scheduler():
+    """Node registers its xGMI/NUMA topology as annotations; the
+    filter decodes it and honors numa-strict placement (the wire
+    format between the node agent and the scheduler)."""
+    client = FakeKubeClient()
+    manager = FakeDeviceManager("gpu-node-1", n_devices=8)
+    client.add_node({"metadata": {"name": "gpu-node-1",
+                                  "annotations": {}}})
+    manager.register(client)
+    ann = client.get_node("gpu-node-1")["metadata"]["annotations"]
+    assert consts.node_topology_ann() in ann
+
+    pod = make_pod(number=2, name="numa-pod", ann={
+        consts.topology_mode_ann(): consts.TOPO_NUMA_STRICT})
+    client.add_pod(pod)
+    res = GpuFilter(client).filter(
+        {"Pod": pod, "NodeNames": ["gpu-node-1"]})
+    assert res["NodeNames"] == ["gpu-node-1"], res
+    pre = client.get_pod("default", "numa-pod")["metadata"][
+        "annotations"][consts.pre_alloc_ann()]
+    # claim text "main[i_GPU-fake-xxxx_c_m,...]": device ids -> NUMA
+    ids = [int(c.split("_")[0]) for c in
+           pre.split("[")[1].rstrip("]").split(",")]
+    numas = {manager.devices[i].numa for i in ids}
+    assert len(numas) == 1, f"numa-strict crossed domains: {ids}"
+
+    # a 6-GPU numa-strict pod cannot fit a 4-GPU domain
+    big = make_pod(number=6, name="too-wide", ann={
+        consts.topology_mode_ann(): consts.TOPO_NUMA_STRICT})
+    client.add_pod(big)
+    res = GpuFilter(client).filter(
+        {"Pod": big, "NodeNames": ["gpu-node-1"]})
+    assert res["NodeNames"] == []
+    assert "Topology" in res["FailedNodes"]["gpu-node-1"]
