@@ -159,3 +159,23 @@ def test_uuid_include_exclude():
     cdcs = Allocator(node).allocate(
         build_allocation_request(make_pod(ann=ann)))
     assert cdcs[0].claims[0].id == 2
+
+
+def test_claim_codec_malformed_inputs():
+    """Malformed annotation text must raise ValueError, never crash
+    or silently mis-parse (annotations are user-influencable)."""
+    from vgpu_manager_amd.device.types import unmarshal_pod_claim
+    import pytest as _pytest
+    for bad in [
+        "main[",                 # unterminated
+        "main[0_GPU_x]",         # too few fields
+        "main[a_GPU-1_b_c]",     # non-numeric id/cores/memory
+        "main[0_GPU-1_50_10,]",  # trailing comma
+        "[0_GPU-1_50_10]",       # missing container name
+        "main[0_GPU-1_50_10]x",  # trailing junk
+    ]:
+        with _pytest.raises(ValueError):
+            unmarshal_pod_claim(bad)
+    # empty containers are legal ("cont[]")
+    out = unmarshal_pod_claim("init[]")
+    assert out[0].name == "init" and out[0].claims == []

@@ -90,11 +90,16 @@ class ContainerDeviceClaim:
     @staticmethod
     def unmarshal(text: str) -> "ContainerDeviceClaim":
         text = text.replace(" ", "")
-        m = re.fullmatch(r"([^\[\]]*)\[(.*)\]", text)
+        m = re.fullmatch(r"([^\[\]]+)\[(.*)\]", text)
         if not m:
             raise ValueError(f"container claim format error: {text!r}")
         name, inner = m.group(1), m.group(2)
-        claims = [DeviceClaim.unmarshal(t) for t in inner.split(",") if t]
+        if not inner:
+            return ContainerDeviceClaim(name=name, claims=[])
+        parts = inner.split(",")
+        if any(not t for t in parts):
+            raise ValueError(f"empty device claim in: {text!r}")
+        claims = [DeviceClaim.unmarshal(t) for t in parts]
         return ContainerDeviceClaim(name=name, claims=claims)
 
 
