@@ -279,3 +279,28 @@ def test_dra_health_republish():
         healthy = [x["basic"]["attributes"]["healthy"]["bool"]
                    for x in devs]
         assert healthy.count(False) == 1
+
+
+def test_pbwire_decode_garbage_is_safe():
+    """Truncated/garbage protobuf buffers (kubelet is a remote peer)
+    must raise cleanly — no hangs, no partial-state crashes."""
+    import random
+
+    from vgpu_manager_amd.deviceplugin import api
+    from vgpu_manager_amd.util.pbwire import Message
+
+    rng = random.Random(1234)
+    classes = [api.AllocateRequest, api.PreferredAllocationRequest,
+               api.RegisterRequest, api.ListAndWatchResponse]
+    for cls in classes:
+        for n in (0, 1, 3, 17, 64, 257):
+            for _ in range(20):
+                buf = bytes(rng.randrange(256) for _ in range(n))
+                try:
+                    cls.decode(buf)
+                except (ValueError, IndexError):
+                    pass  # clean rejection is fine
+    # roundtrip still intact after the fuzz pass
+    req = api.RegisterRequest(version="v1beta1", endpoint="e",
+                              resource_name="amd.com/vgpu-number")
+    assert api.RegisterRequest.decode(req.encode()).endpoint == "e"
