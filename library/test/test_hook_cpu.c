@@ -377,7 +377,7 @@ static int scenario_storm(void) {
     struct timespec t0, t1;
     clock_gettime(CLOCK_MONOTONIC, &t0);
     dim3 grid = {16384, 1, 1}, block = {256, 1, 1};
-    for (int i = 0; i < 200; i++)
+    for (int i = 0; i < 60; i++)
         CHECK(hipLaunchKernel((void *)scenario_storm, grid, block, NULL,
                               0, NULL) == hipSuccess);
     clock_gettime(CLOCK_MONOTONIC, &t1);
