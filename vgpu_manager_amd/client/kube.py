@@ -282,12 +282,28 @@ class RestKubeClient(KubeClient):
         headers = {}
         if content_type:
             headers["Content-Type"] = content_type
-        r = self.session.request(method, f"{self.base}{path}",
-                                 json=body, headers=headers,
-                                 verify=self.verify, timeout=30)
-        if r.status_code >= 300:
-            raise KubeError(f"{method} {path}: {r.status_code} {r.text[:200]}")
-        return r.json() if r.text else {}
+        # retry transient failures for idempotent reads only; writes
+        # surface immediately (callers own their retry semantics —
+        # e.g. lease updates MUST NOT blind-retry a conflict)
+        attempts = 3 if method == "GET" else 1
+        last = None
+        for i in range(attempts):
+            try:
+                r = self.session.request(method, f"{self.base}{path}",
+                                         json=body, headers=headers,
+                                         verify=self.verify, timeout=30)
+            except OSError as e:
+                last = KubeError(f"{method} {path}: {e}")
+                time.sleep(0.2 * (i + 1))
+                continue
+            if r.status_code in (429, 500, 502, 503, 504) and                     i + 1 < attempts:
+                time.sleep(0.2 * (i + 1))
+                continue
+            if r.status_code >= 300:
+                raise KubeError(
+                    f"{method} {path}: {r.status_code} {r.text[:200]}")
+            return r.json() if r.text else {}
+        raise last or KubeError(f"{method} {path}: retries exhausted")
 
     def get_node(self, name):
         return self._req("GET", f"/api/v1/nodes/{name}")
