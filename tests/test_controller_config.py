@@ -108,3 +108,19 @@ def test_fake_id_store(tmp_path):
     assert order2 == ["GPU-a", "GPU-b", "GPU-c"]
     # removed device drops without disturbing order
     assert store2.stable_order(["GPU-c", "GPU-a"]) == ["GPU-a", "GPU-c"]
+
+
+def test_cmd_entrypoints_argparse_wiring():
+    """Every binary's argparse wiring stays importable and parses
+    --help (catches renamed/missing flag regressions cheaply)."""
+    import subprocess, sys
+    mods = ["vgpu_manager_amd.cmd.device_plugin",
+            "vgpu_manager_amd.cmd.device_scheduler",
+            "vgpu_manager_amd.cmd.device_monitor",
+            "vgpu_manager_amd.cmd.device_webhook",
+            "vgpu_manager_amd.cmd.kubelet_plugin"]
+    for m in mods:
+        r = subprocess.run([sys.executable, "-m", m, "--help"],
+                           capture_output=True, text=True, timeout=60)
+        assert r.returncode == 0, f"{m}: {r.stderr[-300:]}"
+        assert "usage:" in r.stdout
