@@ -520,3 +520,33 @@ def test_dra_grpc_prepare_with_sharing_strategy(state, tmp_path):
         ch.close()
     finally:
         server.stop()
+
+
+def test_dynamic_cpx_holders_survive_restart(tmp_path):
+    """Driver restart: holder refcounts rebuilt from the checkpoint so
+    unpreparing ONE of two cpx claims does not revert the shared GPU."""
+    from vgpu_manager_amd.device.partition import (
+        FakePartitionBackend,
+        PartitionManager,
+    )
+    be = FakePartitionBackend(n_gpus=1)
+    devices = [fake_device(0)]
+    cp = str(tmp_path / "cp.json")
+    s1 = DeviceState("n", devices, claims_dir=str(tmp_path / "c"),
+                     checkpoint_path=cp,
+                     partition_manager=PartitionManager(be))
+    s1.prepare("cpx-a", [VgpuClaimParams(uuid="GPU-fake-0000",
+                                         cpx_partitions=[0])])
+    s1.prepare("cpx-b", [VgpuClaimParams(uuid="GPU-fake-0000",
+                                         cpx_partitions=[1],
+                                         partition_key="k2")])
+    assert be.modes[0] == "CPX"
+
+    # restart: fresh DeviceState + fresh manager over the same backend
+    s2 = DeviceState("n", devices, claims_dir=str(tmp_path / "c"),
+                     checkpoint_path=cp,
+                     partition_manager=PartitionManager(be))
+    s2.unprepare("cpx-a")
+    assert be.modes[0] == "CPX", "reverted while cpx-b still holds it"
+    s2.unprepare("cpx-b")
+    assert be.modes[0] == "SPX"

@@ -161,5 +161,12 @@ class PartitionManager:
                 log.warning("GPU %d: revert to SPX failed: %s",
                             gpu_index, e)
 
+    def restore(self, gpu_index: int, claim_uid: str) -> None:
+        """Re-register a holder from a checkpoint after restart —
+        no mode switch (the GPU is already in whatever mode it is);
+        only the refcount is rebuilt so a later unprepare of ANOTHER
+        claim cannot revert a still-shared GPU to SPX."""
+        self._cpx_claims.setdefault(gpu_index, set()).add(claim_uid)
+
     def holders(self, gpu_index: int) -> set:
         return set(self._cpx_claims.get(gpu_index, set()))

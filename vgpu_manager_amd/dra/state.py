@@ -195,6 +195,13 @@ class DeviceState:
         # static partitioning (modes set by the operator)
         self.partition_manager = partition_manager
         self._mu = threading.Lock()
+        # rebuild CPX holder refcounts from the checkpoint: without
+        # this, the first unprepare after a driver restart would
+        # revert a GPU to SPX while other claims still hold it
+        if partition_manager is not None:
+            for uid, entry in self.checkpoint.claims.items():
+                for gpu in entry.get("cpx_gpus", []):
+                    partition_manager.restore(gpu, uid)
 
     # ---- prepare ----
     def prepare(self, claim_uid: str, params: List[VgpuClaimParams],
