@@ -33,6 +33,8 @@ extern "C" {
 #include <atomic>
 #include <chrono>
 #include <cstring>
+#include <dirent.h>
+#include <functional>
 #include <map>
 #include <string>
 #include <thread>
@@ -70,7 +72,8 @@ struct AgentSampler {
 
     bool init(rocprofiler_agent_id_t a);
     bool sample(uint32_t window_ms, double *gpu_util_pct,
-                double *sq_waves);
+                double *sq_waves,
+                const std::function<void()> &tick = nullptr);
 };
 
 static std::vector<rocprofiler_agent_v0_t> g_agents;
@@ -144,10 +147,17 @@ bool AgentSampler::init(rocprofiler_agent_id_t a) {
 }
 
 bool AgentSampler::sample(uint32_t window_ms, double *gpu_util_pct,
-                          double *sq_waves) {
+                          double *sq_waves,
+                          const std::function<void()> &tick) {
     std::vector<rocprofiler_counter_record_t> recs(record_cap);
     RP_CHECK(rocprofiler_start_context(ctx));
-    std::this_thread::sleep_for(std::chrono::milliseconds(window_ms));
+    /* ~10ms ticks during the counting window: callers sub-sample
+     * per-process KFD occupancy for an unbiased duty mean            */
+    for (uint32_t t = 0; t < window_ms; t += 10) {
+        std::this_thread::sleep_for(std::chrono::milliseconds(
+            window_ms - t > 10 ? 10 : window_ms - t));
+        if (tick) tick();
+    }
     size_t n = recs.size();
     rocprofiler_status_t st = rocprofiler_sample_device_counting_service(
         ctx, {}, ROCPROFILER_COUNTER_FLAG_NONE, recs.data(), &n);
@@ -206,6 +216,54 @@ extern "C" rocprofiler_tool_configure_result_t *rocprofiler_configure(
     return &result;
 }
 
+/* ---- per-pid KFD occupancy sub-sampling --------------------------- */
+/* kfd proc stats cu_occupancy files: CUs the
+ * process occupies RIGHT NOW.  Point samples -> accumulate over the
+ * window, publish the MEAN (same estimator the in-container shim
+ * uses; full-or-zero point samples whipsaw controllers).            */
+struct KfdOccWindow {
+    std::map<int, std::pair<uint64_t, uint32_t>> acc; /* pid -> sum,n */
+
+    void tick() {
+        DIR *d = opendir("/sys/class/kfd/kfd/proc");
+        if (!d) return;
+        struct dirent *e;
+        while ((e = readdir(d)) != nullptr) {
+            if (e->d_name[0] < '0' || e->d_name[0] > '9') continue;
+            int pid = atoi(e->d_name);
+            char pdir[320];
+            snprintf(pdir, sizeof(pdir),
+                     "/sys/class/kfd/kfd/proc/%s", e->d_name);
+            DIR *pd = opendir(pdir);
+            if (!pd) continue;
+            uint64_t occ = 0;
+            struct dirent *se;
+            while ((se = readdir(pd)) != nullptr) {
+                if (strncmp(se->d_name, "stats_", 6) != 0) continue;
+                char fp[640];
+                snprintf(fp, sizeof(fp), "%s/%s/cu_occupancy", pdir,
+                         se->d_name);
+                FILE *f = fopen(fp, "re");
+                if (!f) continue;
+                unsigned v = 0;
+                if (fscanf(f, "%u", &v) == 1) occ += v;
+                fclose(f);
+            }
+            closedir(pd);
+            auto &a = acc[pid];
+            a.first += occ;
+            a.second++;
+        }
+        closedir(d);
+    }
+
+    uint32_t mean(int pid) const {
+        auto it = acc.find(pid);
+        if (it == acc.end() || it->second.second == 0) return 0;
+        return (uint32_t)(it->second.first / it->second.second);
+    }
+};
+
 /* ---- amd-smi per-process view ------------------------------------- */
 struct SmiState {
     bool ok = false;
@@ -215,7 +273,8 @@ struct SmiState {
     uint64_t prev_ns = 0;
 
     bool init();
-    void fill(int dev, device_util_t *out, uint64_t now);
+    void fill(int dev, device_util_t *out, uint64_t now,
+              const KfdOccWindow *occ = nullptr);
 };
 
 bool SmiState::init() {
@@ -241,7 +300,8 @@ bool SmiState::init() {
     return ok;
 }
 
-void SmiState::fill(int dev, device_util_t *out, uint64_t now) {
+void SmiState::fill(int dev, device_util_t *out, uint64_t now,
+                    const KfdOccWindow *occ) {
     if (!ok || dev >= (int)handles.size()) return;
     amdsmi_processor_handle h = handles[dev];
     uint64_t vram = 0;
@@ -269,7 +329,11 @@ void SmiState::fill(int dev, device_util_t *out, uint64_t now) {
         out->procs[count].gfx_busy_permille =
             permille > 1000 ? 1000 : permille;
         out->procs[count].vram_bytes = procs[i].memory_usage.vram_mem;
-        out->procs[count].cu_occupancy = procs[i].cu_occupancy;
+        /* window-mean occupancy when available (unbiased duty),
+         * else amd-smi's point sample                               */
+        uint32_t m = occ ? occ->mean(pid) : 0;
+        out->procs[count].cu_occupancy =
+            m ? m : procs[i].cu_occupancy;
         count++;
     }
     out->proc_count = count;
@@ -327,15 +391,18 @@ int main(int argc, char **argv) {
         uint64_t now = mono_ns_now();
         for (int d = 0; d < ndev; d++) {
             double util = 0, waves = 0;
+            KfdOccWindow occ;
             /* the counting window IS the cadence: GPU_UTIL =
-             * GRBM_GUI_ACTIVE/GRBM_COUNT over the window */
-            if (!g_samplers[d].sample(interval_ms, &util, &waves))
+             * GRBM_GUI_ACTIVE/GRBM_COUNT over the window; per-pid
+             * occupancy is sub-sampled every ~10ms inside it */
+            if (!g_samplers[d].sample(interval_ms, &util, &waves,
+                                      [&occ] { occ.tick(); }))
                 continue;
             device_util_t *u = &region->devices[d];
             __atomic_fetch_add(&u->seq, 1, __ATOMIC_ACQ_REL); /* odd */
             u->dev_busy_permille = (uint32_t)(util * 10.0);
             u->sample_ns = now;
-            smi.fill(d, u, now);
+            smi.fill(d, u, now, &occ);
             __atomic_fetch_add(&u->seq, 1, __ATOMIC_RELEASE); /* even */
         }
         smi.prev_ns = now;
