@@ -136,6 +136,14 @@ class PluginSet:
             if register:
                 s.register(kubelet_socket)
 
+    def register_all(self, kubelet_socket: str = api.KUBELET_SOCKET
+                     ) -> None:
+        """Re-announce to a restarted kubelet WITHOUT recreating the
+        gRPC servers (they still serve on their sockets; recreating
+        them would leak the old grpc.Server instances)."""
+        for s in self.servers:
+            s.register(kubelet_socket)
+
     def stop_all(self) -> None:
         for s in self.servers:
             s.stop()
@@ -158,7 +166,7 @@ def watch_kubelet_restart(plugin_set: PluginSet,
                     ino != last_ino:
                 log.warning("kubelet restart detected; re-registering")
                 try:
-                    plugin_set.start_all(kubelet_socket)
+                    plugin_set.register_all(kubelet_socket)
                 except Exception as e:
                     log.error("re-register failed: %s", e)
             last_ino = ino
