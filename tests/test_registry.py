@@ -57,11 +57,16 @@ def test_register_unknown_allocation_rejected(registry):
 def test_spoofed_pid_rejected(registry):
     sock, base, server = registry
     # claim pid 1 (different cgroup from the test process in most
-    # environments; if not, force the verifier)
+    # environments; if not, force the verifier).  Unverifiable pids
+    # are SKIPPED, not fatal: a pid-namespaced container legitimately
+    # sends pid numbers that mean nothing on the host — the security
+    # property is that the spoofed pid is never persisted.
     server.verify = lambda peer, claimed: peer == claimed
     out = register_via_socket(sock, "uid-1", "main", pids=[1])
-    assert not out["ok"]
-    assert "not in caller" in out["error"]
+    assert out["ok"]
+    recorded = read_pids(base)
+    assert 1 not in recorded, "spoofed pid persisted"
+    assert os.getpid() in recorded  # peercred caller
 
 
 def test_path_traversal_rejected(registry):

@@ -41,6 +41,19 @@ def peer_credentials(conn: socket.socket):
     return pid, uid, gid
 
 
+def _host_ppid(pid: int) -> int:
+    """Parent pid in HOST namespace via /proc (the registry runs on
+    the host): for the fork/exec'd device-client this is the workload
+    process that triggered registration."""
+    try:
+        for ln in open(f"/proc/{pid}/status"):
+            if ln.startswith("PPid:"):
+                return int(ln.split()[1])
+    except (OSError, ValueError):
+        pass
+    return 0
+
+
 def cgroup_of(pid: int) -> str:
     try:
         return open(f"/proc/{pid}/cgroup").read()
@@ -96,20 +109,26 @@ class _Handler(socketserver.StreamRequestHandler):
             req = json.loads(line)
             pod_uid = str(req.get("pod_uid", ""))
             container = str(req.get("container_name", ""))
-            pids = [int(p) for p in req.get("pids", [])] or [peer_pid]
+            claimed = [int(p) for p in req.get("pids", [])]
             if not pod_uid or not container:
                 raise ValueError("pod_uid and container_name required")
             if "/" in pod_uid or "/" in container or \
                     ".." in pod_uid or ".." in container:
                 raise ValueError("invalid identifier")
-            for p in pids:
-                if not server.verify(peer_pid, p):
-                    raise PermissionError(
-                        f"pid {p} not in caller's container "
-                        f"(peer {peer_pid})")
-            # always include the actual caller
+            # A pid-namespaced container sends ITS pid numbers, which
+            # mean nothing here: verify each against the peer's
+            # cgroup and silently SKIP mismatches (they are either ns
+            # pids or spoofs; either way they must not be persisted).
+            pids = [p for p in claimed if server.verify(peer_pid, p)]
+            # the caller (the exec'd device-client) and — crucially —
+            # its parent, the actual workload process, in HOST pid
+            # terms from the kernel, immune to namespace games
             if peer_pid not in pids:
                 pids.append(peer_pid)
+            ppid = _host_ppid(peer_pid)
+            if ppid > 1 and ppid not in pids and \
+                    server.verify(peer_pid, ppid):
+                pids.append(ppid)
             n = server.state.register(pod_uid, container, pids)
             self.wfile.write(json.dumps(
                 {"ok": True, "count": n}).encode() + b"\n")
