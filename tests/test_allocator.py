@@ -179,3 +179,45 @@ def test_claim_codec_malformed_inputs():
     # empty containers are legal ("cont[]")
     out = unmarshal_pod_claim("init[]")
     assert out[0].name == "init" and out[0].claims == []
+
+
+def test_claim_codec_property_roundtrip():
+    """Property test: every well-formed claim survives
+    marshal->unmarshal byte-identically (hypothesis)."""
+    from hypothesis import given, settings, strategies as st
+
+    from vgpu_manager_amd.device.types import (
+        ContainerDeviceClaim,
+        DeviceClaim,
+        marshal_pod_claim,
+        unmarshal_pod_claim,
+    )
+
+    name_st = st.text(
+        alphabet=st.characters(whitelist_categories=("Ll", "Lu", "Nd"),
+                               whitelist_characters="-."),
+        min_size=1, max_size=20)
+    uuid_st = st.text(
+        alphabet=st.characters(whitelist_categories=("Ll", "Lu", "Nd"),
+                               whitelist_characters="-"),
+        min_size=1, max_size=40)
+    claim_st = st.builds(
+        DeviceClaim,
+        id=st.integers(0, 15),
+        uuid=uuid_st,
+        cores=st.integers(0, 100),
+        memory=st.integers(0, 1 << 20))
+    cdc_st = st.builds(
+        ContainerDeviceClaim,
+        name=name_st,
+        claims=st.lists(claim_st, max_size=4))
+
+    @settings(max_examples=200, deadline=None)
+    @given(st.lists(cdc_st, min_size=1, max_size=3))
+    def roundtrip(cdcs):
+        text = marshal_pod_claim(cdcs)
+        back = unmarshal_pod_claim(text)
+        assert back == cdcs
+        assert marshal_pod_claim(back) == text
+
+    roundtrip()
