@@ -304,3 +304,36 @@ def test_pbwire_decode_garbage_is_safe():
     req = api.RegisterRequest(version="v1beta1", endpoint="e",
                               resource_name="amd.com/vgpu-number")
     assert api.RegisterRequest.decode(req.encode()).endpoint == "e"
+
+
+def test_pbwire_property_roundtrip():
+    """Property: typed messages survive encode->decode (hypothesis),
+    across strings, ints, bools, repeated and nested fields."""
+    from hypothesis import given, settings, strategies as st
+
+    from vgpu_manager_amd.deviceplugin import api
+
+    dev_st = st.builds(
+        api.Device,
+        ID=st.text(min_size=0, max_size=40),
+        health=st.sampled_from([api.HEALTHY, api.UNHEALTHY, ""]),
+        topology=st.one_of(
+            st.none(),
+            st.builds(api.TopologyInfo,
+                      nodes=st.lists(st.builds(api.NUMANode,
+                                               ID=st.integers(0, 7)),
+                                     max_size=2))))
+
+    @settings(max_examples=150, deadline=None)
+    @given(st.lists(dev_st, max_size=5))
+    def roundtrip(devices):
+        msg = api.ListAndWatchResponse(devices=devices)
+        back = api.ListAndWatchResponse.decode(msg.encode())
+        assert len(back.devices) == len(devices)
+        for a, b in zip(devices, back.devices):
+            assert a.ID == b.ID and a.health == b.health
+            if a.topology is not None and a.topology.nodes:
+                assert [n.ID for n in a.topology.nodes] == \
+                    [n.ID for n in b.topology.nodes]
+
+    roundtrip()
