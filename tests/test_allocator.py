@@ -221,3 +221,37 @@ def test_claim_codec_property_roundtrip():
         assert marshal_pod_claim(back) == text
 
     roundtrip()
+
+
+def test_allocator_capacity_invariant_property():
+    """Property: however pods arrive, accepted allocations never
+    oversubscribe any device's slots, cores or memory."""
+    from hypothesis import given, settings, strategies as st
+
+    from vgpu_manager_amd.device.types import fake_node
+
+    pod_st = st.tuples(st.integers(1, 4),          # number
+                       st.sampled_from([0, 25, 50, 100]),  # cores/dev
+                       st.sampled_from([0, 1024, 8192, 65536]))
+
+    @settings(max_examples=60, deadline=None)
+    @given(st.lists(pod_st, min_size=1, max_size=30))
+    def run(pods):
+        node = fake_node("n", 8)
+        for i, (num, cores, mem) in enumerate(pods):
+            pod = make_pod(number=num, cores=cores * num,
+                           memory=mem * num, name=f"p{i}")
+            req = build_allocation_request(pod)
+            try:
+                # allocate() itself accumulates usage into `node`
+                # (the scheduler rebuilds a fresh snapshot per filter
+                # call and applies claims from pod annotations)
+                Allocator(node).allocate(req)
+            except AllocationError:
+                continue
+        for d in node.devices.values():
+            assert d.used_number <= d.info.number
+            assert d.used_cores <= d.info.core
+            assert d.used_memory <= d.info.memory
+
+    run()
