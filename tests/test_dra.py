@@ -550,3 +550,40 @@ def test_dynamic_cpx_holders_survive_restart(tmp_path):
     assert be.modes[0] == "CPX", "reverted while cpx-b still holds it"
     s2.unprepare("cpx-b")
     assert be.modes[0] == "SPX"
+
+
+def test_sharing_property_invariants():
+    """Property: any accepted cu-partition decision hands out DISJOINT
+    partitions within the 8 XCDs; any accepted time-slicing decision
+    never grants more than 100% total."""
+    from hypothesis import given, settings, strategies as st
+
+    from vgpu_manager_amd.dra.sharing import (
+        SharingError,
+        apply_sharing_config,
+    )
+
+    @settings(max_examples=150, deadline=None)
+    @given(st.integers(1, 10),
+           st.one_of(st.none(), st.integers(0, 12)),
+           st.sampled_from(["cu-partition", "time-slicing"]))
+    def run(n_consumers, per, strategy):
+        params = [VgpuClaimParams(uuid="GPU-fake-0000")
+                  for _ in range(n_consumers)]
+        cfg = {"strategy": strategy}
+        if strategy == "cu-partition" and per is not None:
+            cfg["partitionsPerConsumer"] = per
+        try:
+            dec = apply_sharing_config(params, cfg)
+        except SharingError:
+            return  # rejection is always acceptable
+        if strategy == "cu-partition":
+            seen = []
+            for idx, parts in dec.partitions.items():
+                assert all(0 <= p < 8 for p in parts)
+                seen += parts
+            assert len(seen) == len(set(seen)), "partitions overlap"
+        else:
+            assert sum(dec.core_limits.values()) <= 100
+
+    run()
