@@ -255,3 +255,10 @@ def test_allocator_capacity_invariant_property():
             assert d.used_memory <= d.info.memory
 
     run()
+
+
+def test_claim_marshal_rejects_underscore_uuid():
+    from vgpu_manager_amd.device.types import DeviceClaim
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        DeviceClaim(id=0, uuid="GPU_bad", cores=0, memory=0).marshal()

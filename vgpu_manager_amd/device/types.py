@@ -66,6 +66,12 @@ class DeviceClaim:
     memory: int  # MiB
 
     def marshal(self) -> str:
+        # "_" is the field separator of the claim text (reference
+        # format "id_uuid_cores_memory"); a uuid containing one would
+        # produce an unparseable annotation — fail at WRITE time
+        if "_" in self.uuid:
+            raise ValueError(f"device uuid may not contain '_': "
+                             f"{self.uuid!r}")
         return f"{self.id}_{self.uuid}_{self.cores}_{self.memory}"
 
     @staticmethod
