@@ -587,3 +587,36 @@ def test_sharing_property_invariants():
             assert sum(dec.core_limits.values()) <= 100
 
     run()
+
+
+def test_dra_grpc_uid_mismatch_rejected(state, tmp_path):
+    """A claim fetched by name whose uid differs from the request's
+    (stale/recreated claim) must fail per-claim, not prepare."""
+    import grpc as _grpc
+    from vgpu_manager_amd.client.kube import FakeKubeClient
+    from vgpu_manager_amd.dra import api as dapi
+    from vgpu_manager_amd.dra.driver import DraDriver, DraDriverServer
+
+    client = FakeKubeClient()
+    client.add_resource_claim(_alloc_claim(uid="uid-NEW"))
+    endpoint = str(tmp_path / "p" / "dra.sock")
+    server = DraDriverServer(
+        DraDriver(state, client, endpoint=endpoint),
+        plugins_dir=str(tmp_path / "p"),
+        plugins_registry=str(tmp_path / "r"))
+    server.start()
+    try:
+        ch = _grpc.insecure_channel(f"unix://{endpoint}")
+        prep = ch.unary_unary(
+            f"/{dapi.DRA_SERVICE}/NodePrepareResources",
+            request_serializer=lambda m: m.encode(),
+            response_deserializer=dapi.NodePrepareResourcesResponse
+            .decode)
+        resp = prep(dapi.NodePrepareResourcesRequest(claims=[
+            dapi.Claim(namespace="default", uid="uid-OLD", name="c")]),
+            timeout=10)
+        assert "mismatch" in resp.claims[0].value.error
+        assert state.prepared_claims() == []
+        ch.close()
+    finally:
+        server.stop()
