@@ -301,3 +301,24 @@ def test_soft_limit_exclusive_burst(built_library, tmp_path):
         stop.set()
         t.join(timeout=2)
         writer.close()
+
+
+def test_vmem_region_cross_language_read(built_library, tmp_path):
+    """The monitor-side Python reader consumes a vmem region the C
+    shim created and used (header, counters layout)."""
+    from vgpu_manager_amd.config.regions import VmemRegionReader
+    vmem = str(tmp_path / "vmem_node.config")
+    run_scenario("oversold", {
+        "VGPU_MEM_LIMIT_0": "1m", "VGPU_MEM_OVERSOLD": "1",
+        "VGPU_MEM_ACCOUNT_MODE": "ledger",
+        "VGPU_VMEM_PATH_OVERRIDE": vmem,
+    })
+    r = VmemRegionReader(vmem)
+    usage = r.device_usage()
+    assert len(usage) == 16
+    # the scenario frees everything before exiting; charges retired
+    assert usage[0]["vmem_used"] == 0
+    # dev 0 still holds 900K of device memory at scenario end? no —
+    # the PROCESS exited: exit cleanup retired its hooked bytes too
+    assert usage[0]["dev_hooked_used"] == 0
+    r.close()
