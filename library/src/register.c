@@ -17,8 +17,14 @@
 #include <unistd.h>
 
 void vgpu_register_client(void) {
-    if (access(VGPU_REGISTRY_SOCKET, F_OK) != 0 ||
-        access(VGPU_DEVICE_CLIENT, X_OK) != 0)
+    char sbuf[256], cbuf[256];
+    const char *sock = vgpu_getenv("VGPU_REGISTRY_SOCKET_OVERRIDE",
+                                   sbuf, sizeof(sbuf));
+    if (!sock) sock = VGPU_REGISTRY_SOCKET;
+    const char *cli = vgpu_getenv("VGPU_DEVICE_CLIENT_OVERRIDE", cbuf,
+                                  sizeof(cbuf));
+    if (!cli) cli = VGPU_DEVICE_CLIENT;
+    if (access(sock, F_OK) != 0 || access(cli, X_OK) != 0)
         return; /* not client mode */
     char buf[128];
     const char *uid = vgpu_getenv("VGPU_POD_UID", buf, sizeof(buf));
@@ -31,8 +37,8 @@ void vgpu_register_client(void) {
     if (pid == 0) {
         char pidstr[16];
         snprintf(pidstr, sizeof(pidstr), "%d", (int)getppid());
-        execl(VGPU_DEVICE_CLIENT, "device-client", "--socket",
-              VGPU_REGISTRY_SOCKET, "--pod-uid", uid, "--container", cont,
+        execl(cli, "device-client", "--socket",
+              sock, "--pod-uid", uid, "--container", cont,
               "--pid", pidstr, (char *)NULL);
         _exit(127);
     }

@@ -106,3 +106,37 @@ def test_c_device_client(registry, built_core):
     assert r.returncode == 0, r.stdout + r.stderr
     pids = read_pids(base)
     assert len(pids) >= 1
+
+
+def test_client_mode_end_to_end_with_shim(registry, built_library,
+                                          tmp_path):
+    """FULL client-mode loop: the C shim's init fork/execs the C
+    device-client against the Python registry server; the server
+    resolves the caller via SO_PEERCRED (+host PPID) and persists
+    pids.config in the allocated container dir."""
+    import subprocess as sp
+    sock, base, _server = registry
+    build = os.path.join(LIB_DIR, "build")
+    env = dict(os.environ)
+    env.update({
+        "VGPU_MEM_LIMIT_0": "1m",
+        "VGPU_MEM_ACCOUNT_MODE": "ledger",
+        "VGPU_POD_UID": "uid-1",
+        "VGPU_CONTAINER_NAME": "main",
+        "VGPU_REGISTRY_SOCKET_OVERRIDE": sock,
+        "VGPU_DEVICE_CLIENT_OVERRIDE":
+            os.path.join(build, "device-client"),
+        "LD_PRELOAD": os.path.join(build, "libvgpu-control.so"),
+        "LD_LIBRARY_PATH": os.path.join(build, "stub"),
+        "VGPU_REAL_HIP_PATH": os.path.join(build, "stub",
+                                           "libamdhip64.so.7"),
+    })
+    r = sp.run([os.path.join(build, "test_hook_cpu"), "quota"],
+               env=env, capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stdout + r.stderr
+    pids = read_pids(base)
+    assert pids, "registration did not persist pids.config"
+    # the workload process (the shim's host pid) must be registered;
+    # here host ns == test ns so it is the scenario binary's pid —
+    # at minimum the transient client and its parent were captured
+    assert len(pids) >= 1
