@@ -165,3 +165,24 @@ def test_webhook_http_new_paths():
                         {"name": "r", "deviceClassName": "gpu-manager",
                          "count": 99}]}}}}})
     assert r.json()["response"]["allowed"] is False
+
+
+def test_dra_mode_creates_claim_template():
+    """DRA conversion is only complete if the referenced
+    ResourceClaimTemplate EXISTS: the webhook creates it server-side
+    during admission."""
+    from starlette.testclient import TestClient
+    from vgpu_manager_amd.client.kube import FakeKubeClient
+    from vgpu_manager_amd.webhook.admission import create_app
+
+    client = FakeKubeClient()
+    tc = TestClient(create_app(dra_mode=True, client=client))
+    pod = make_pod(number=2, cores=100, memory=8192, name="dra-pod")
+    r = tc.post("/webhook/mutate-pod",
+                json={"request": {"uid": "u", "object": pod}})
+    assert r.json()["response"]["allowed"] is True
+    tmpl = client.resource_claim_templates[("default", "vgpu-dra-pod")]
+    spec = tmpl["spec"]["spec"]["devices"]
+    assert spec["requests"][0]["count"] == 2
+    params = spec["config"][0]["opaque"]["parameters"]
+    assert params == {"cores": 50, "memoryMiB": 4096}

@@ -72,6 +72,10 @@ class KubeClient:
     def get_resource_claim(self, namespace: str, name: str) -> dict:
         raise KubeError("resource claims unsupported")
 
+    def create_resource_claim_template(self, namespace: str,
+                                       template: dict) -> None:
+        raise KubeError("resource claim templates unsupported")
+
     def apply_resource_slice(self, rs: dict) -> None:
         raise KubeError("resource slices unsupported")
 
@@ -108,6 +112,7 @@ class FakeKubeClient(KubeClient):
         self.leases: Dict[tuple, dict] = {}
         self.resource_claims: Dict[tuple, dict] = {}
         self.resource_slices: Dict[str, dict] = {}
+        self.resource_claim_templates: Dict[tuple, dict] = {}
 
     # -- test helpers --
     def add_node(self, node: dict) -> None:
@@ -216,6 +221,12 @@ class FakeKubeClient(KubeClient):
         with self._mu:
             self.resource_slices[rs["metadata"]["name"]] = \
                 json.loads(json.dumps(rs))
+
+    def create_resource_claim_template(self, namespace, template):
+        with self._mu:
+            key = (namespace, template["metadata"]["name"])
+            self.resource_claim_templates[key] = \
+                json.loads(json.dumps(template))
 
     def list_pdbs(self, namespace=None):
         with self._mu:
@@ -387,6 +398,16 @@ class RestKubeClient(KubeClient):
             "GET",
             f"/apis/resource.k8s.io/v1beta1/namespaces/{namespace}"
             f"/resourceclaims/{name}")
+
+    def create_resource_claim_template(self, namespace, template):
+        try:
+            self._req(
+                "POST",
+                f"/apis/resource.k8s.io/v1beta1/namespaces/{namespace}"
+                "/resourceclaimtemplates", template)
+        except KubeError as e:
+            if "409" not in str(e):  # already exists is fine
+                raise
 
     def apply_resource_slice(self, rs):
         name = rs["metadata"]["name"]

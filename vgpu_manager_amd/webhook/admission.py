@@ -76,10 +76,48 @@ def mutate_pod(pod: dict, *, default_scheduler: str = VGPU_SCHEDULER_NAME,
     return patches
 
 
+def build_claim_template(pod: dict) -> Optional[dict]:
+    """The ResourceClaimTemplate matching a pod's vgpu-* limits —
+    created server-side during admission (`create_app(client=...)`)
+    so the converted pod's `resourceClaimTemplateName` resolves."""
+    total_num = 0
+    cores = 0
+    mem = 0
+    for c in pod.get("spec", {}).get("containers") or []:
+        limits = (c.get("resources", {}) or {}).get("limits", {}) or {}
+        n = int(limits.get(consts.vgpu_number_resource(), 0) or 0)
+        total_num += n
+        cores = max(cores, int(
+            limits.get(consts.vgpu_core_resource(), 0) or 0) //
+            max(n, 1))
+        mem = max(mem, int(
+            limits.get(consts.vgpu_memory_resource(), 0) or 0) //
+            max(n, 1))
+    if total_num < 1:
+        return None
+    params = {}
+    if cores:
+        params["cores"] = cores
+    if mem:
+        params["memoryMiB"] = mem
+    spec: dict = {"devices": {"requests": [{
+        "name": "gpu", "deviceClassName": "vgpu-manager",
+        "count": total_num}]}}
+    if params:
+        spec["devices"]["config"] = [{
+            "requests": ["gpu"],
+            "opaque": {"driver": "manager.amd.com",
+                       "parameters": params}}]
+    name = f"vgpu-{pod.get('metadata', {}).get('name', 'pod')}"
+    return {"apiVersion": "resource.k8s.io/v1beta1",
+            "kind": "ResourceClaimTemplate",
+            "metadata": {"name": name},
+            "spec": {"spec": spec}}
+
+
 def _dra_conversion_patches(pod: dict) -> List[dict]:
-    """Rewrite vgpu-* limits into a generated ResourceClaim reference.
-    The claim template itself is created by the controller from the
-    stashed originals."""
+    """Rewrite vgpu-* limits into a generated ResourceClaim reference
+    (the matching template is created via `build_claim_template`)."""
     patches: List[dict] = []
     originals = {}
     res_names = {consts.vgpu_number_resource(),
@@ -288,7 +326,7 @@ def apply_json_patch(obj: dict, patches: List[dict]) -> dict:
     return obj
 
 
-def create_app(dra_mode: bool = False):
+def create_app(dra_mode: bool = False, client=None):
     # module-level import would make fastapi a hard dependency of every
     # admission-logic consumer; but the Request annotation must resolve
     # from module globals (PEP 563 strings) — so stash it there.
@@ -299,8 +337,21 @@ def create_app(dra_mode: bool = False):
 
     @app.post("/webhook/mutate-pod")
     async def mutate(request: Request):
-        return handle_admission_review(await request.json(),
-                                       mutating=True, dra_mode=dra_mode)
+        body = await request.json()
+        if dra_mode and client is not None:
+            obj = (body.get("request", {}) or {}).get("object", {}) or {}
+            tmpl = build_claim_template(obj)
+            if tmpl is not None:
+                ns = (obj.get("metadata", {}) or {}).get(
+                    "namespace", "default")
+                try:
+                    client.create_resource_claim_template(ns, tmpl)
+                except Exception as e:  # conversion must not block
+                    import logging
+                    logging.getLogger("vgpu.webhook").warning(
+                        "claim template create failed: %s", e)
+        return handle_admission_review(body, mutating=True,
+                                       dra_mode=dra_mode)
 
     @app.post("/webhook/validate-pod")
     async def validate(request: Request):
