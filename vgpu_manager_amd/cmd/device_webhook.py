@@ -16,13 +16,18 @@ def main(argv=None):
     import uvicorn
     from ..webhook.admission import create_app
 
+    client = None
+    if args.dra_mode:
+        # DRA conversion creates ResourceClaimTemplates server-side
+        from ..client.kube import RestKubeClient
+        client = RestKubeClient()
     host, port = args.bind.rsplit(":", 1)
     kwargs = {}
     if args.tls_cert and args.tls_key:
         kwargs = dict(ssl_certfile=args.tls_cert,
                       ssl_keyfile=args.tls_key)
-    uvicorn.run(create_app(dra_mode=args.dra_mode), host=host,
-                port=int(port), **kwargs)
+    uvicorn.run(create_app(dra_mode=args.dra_mode, client=client),
+                host=host, port=int(port), **kwargs)
     return 0
 
 
