@@ -186,3 +186,50 @@ def test_dra_mode_creates_claim_template():
     assert spec["requests"][0]["count"] == 2
     params = spec["config"][0]["opaque"]["parameters"]
     assert params == {"cores": 50, "memoryMiB": 4096}
+
+def test_dra_per_container_conversion():
+    """Per-container DRA mode: each vgpu container gets its OWN claim
+    and template (reference supports combined and per-container)."""
+    from starlette.testclient import TestClient
+    from vgpu_manager_amd.client.kube import FakeKubeClient
+    from vgpu_manager_amd.webhook.admission import create_app, mutate_pod
+
+    pod = {"metadata": {"name": "multi", "namespace": "ns1"},
+           "spec": {"containers": [
+               {"name": "a", "resources": {"limits": {
+                   consts.vgpu_number_resource(): "1",
+                   consts.vgpu_core_resource(): "30"}}},
+               {"name": "plain", "resources": {}},
+               {"name": "b", "resources": {"limits": {
+                   consts.vgpu_number_resource(): "2",
+                   consts.vgpu_memory_resource(): "8192"}}},
+           ]}}
+    out = apply_json_patch(
+        pod, mutate_pod(pod, dra_mode=True, dra_per_container=True))
+    claims = {c["name"]: c["resourceClaimTemplateName"]
+              for c in out["spec"]["resourceClaims"]}
+    assert claims == {"vgpu-a": "vgpu-multi-a",
+                      "vgpu-b": "vgpu-multi-b"}
+    assert out["spec"]["containers"][0]["resources"]["claims"] == \
+        [{"name": "vgpu-a"}]
+    assert out["spec"]["containers"][2]["resources"]["claims"] == \
+        [{"name": "vgpu-b"}]
+    assert "claims" not in out["spec"]["containers"][1].get(
+        "resources", {})
+
+    client = FakeKubeClient()
+    tc = TestClient(create_app(dra_mode=True, client=client,
+                               dra_per_container=True))
+    r = tc.post("/webhook/mutate-pod",
+                json={"request": {"uid": "u", "object": pod}})
+    assert r.json()["response"]["allowed"] is True
+    ta = client.resource_claim_templates[("ns1", "vgpu-multi-a")]
+    tb = client.resource_claim_templates[("ns1", "vgpu-multi-b")]
+    assert ta["spec"]["spec"]["devices"]["requests"][0]["count"] == 1
+    assert tb["spec"]["spec"]["devices"]["requests"][0]["count"] == 2
+    pa = ta["spec"]["spec"]["devices"]["config"][0]["opaque"][
+        "parameters"]
+    assert pa["cores"] == 30 and pa["partitionKey"] == "a"
+    pb = tb["spec"]["spec"]["devices"]["config"][0]["opaque"][
+        "parameters"]
+    assert pb["memoryMiB"] == 4096 and "cores" not in pb

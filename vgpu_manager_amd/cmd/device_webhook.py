@@ -11,6 +11,9 @@ def main(argv=None):
     ap.add_argument("--tls-cert", default=None)
     ap.add_argument("--tls-key", default=None)
     ap.add_argument("--dra-mode", action="store_true")
+    ap.add_argument("--dra-per-container", action="store_true",
+                    help="emit one ResourceClaim per container instead "
+                         "of one combined pod claim")
     args = ap.parse_args(argv)
 
     import uvicorn
@@ -26,7 +29,8 @@ def main(argv=None):
     if args.tls_cert and args.tls_key:
         kwargs = dict(ssl_certfile=args.tls_cert,
                       ssl_keyfile=args.tls_key)
-    uvicorn.run(create_app(dra_mode=args.dra_mode, client=client),
+    uvicorn.run(create_app(dra_mode=args.dra_mode, client=client,
+                           dra_per_container=args.dra_per_container),
                 host=host, port=int(port), **kwargs)
     return 0
 

@@ -40,7 +40,8 @@ def _is_vgpu_pod(pod: dict) -> bool:
 
 
 def mutate_pod(pod: dict, *, default_scheduler: str = VGPU_SCHEDULER_NAME,
-               dra_mode: bool = False) -> List[dict]:
+               dra_mode: bool = False,
+               dra_per_container: bool = False) -> List[dict]:
     """Returns a JSONPatch list."""
     patches: List[dict] = []
     ann = pod.get("metadata", {}).get("annotations", {}) or {}
@@ -72,7 +73,8 @@ def mutate_pod(pod: dict, *, default_scheduler: str = VGPU_SCHEDULER_NAME,
     default_ann(consts.compute_policy_ann(), consts.COMPUTE_FIXED)
 
     if dra_mode:
-        patches.extend(_dra_conversion_patches(pod))
+        patches.extend(_dra_conversion_patches(
+            pod, per_container=dra_per_container))
     return patches
 
 
@@ -115,14 +117,19 @@ def build_claim_template(pod: dict) -> Optional[dict]:
             "spec": {"spec": spec}}
 
 
-def _dra_conversion_patches(pod: dict) -> List[dict]:
-    """Rewrite vgpu-* limits into a generated ResourceClaim reference
-    (the matching template is created via `build_claim_template`)."""
+def _dra_conversion_patches(pod: dict,
+                            per_container: bool = False) -> List[dict]:
+    """Rewrite vgpu-* limits into generated ResourceClaim references
+    — one combined claim (default) or one per container (reference
+    pod_mutate.go supports both shapes).  Matching templates are
+    created via `build_claim_template(s)`."""
     patches: List[dict] = []
     originals = {}
     res_names = {consts.vgpu_number_resource(),
                  consts.vgpu_core_resource(),
                  consts.vgpu_memory_resource()}
+    pod_name = pod.get('metadata', {}).get('name', 'pod')
+    claim_entries = []
     for ci, c in enumerate(pod.get("spec", {}).get("containers") or []):
         limits = (c.get("resources", {}) or {}).get("limits", {}) or {}
         mine = {k: str(v) for k, v in limits.items() if k in res_names}
@@ -134,23 +141,65 @@ def _dra_conversion_patches(pod: dict) -> List[dict]:
                 "op": "remove",
                 "path": f"/spec/containers/{ci}/resources/limits/" +
                         k.replace("/", "~1")})
+        claim_name = f"vgpu-{c['name']}" if per_container \
+            else "vgpu-claim"
         patches.append({
             "op": "add",
             "path": f"/spec/containers/{ci}/resources/claims",
-            "value": [{"name": "vgpu-claim"}]})
+            "value": [{"name": claim_name}]})
+        if per_container:
+            claim_entries.append({
+                "name": claim_name,
+                "resourceClaimTemplateName":
+                    f"vgpu-{pod_name}-{c['name']}"})
     if originals:
         patches.append({
             "op": "add",
             "path": "/metadata/annotations/" +
                     consts.DRA_ORIGINAL_RESOURCES_ANN.replace("/", "~1"),
             "value": json.dumps(originals, separators=(",", ":"))})
+        if not per_container:
+            claim_entries = [{"name": "vgpu-claim",
+                              "resourceClaimTemplateName":
+                                  f"vgpu-{pod_name}"}]
         patches.append({
             "op": "add",
             "path": "/spec/resourceClaims",
-            "value": [{"name": "vgpu-claim",
-                       "resourceClaimTemplateName":
-                           f"vgpu-{pod.get('metadata', {}).get('name', 'pod')}"}]})
+            "value": claim_entries})
     return patches
+
+
+def build_claim_templates_per_container(pod: dict) -> List[dict]:
+    """One ResourceClaimTemplate per vgpu container (per-container
+    conversion mode); partitionKey pins each container to its own
+    config partition."""
+    out = []
+    pod_name = pod.get('metadata', {}).get('name', 'pod')
+    for c in pod.get("spec", {}).get("containers") or []:
+        limits = (c.get("resources", {}) or {}).get("limits", {}) or {}
+        n = int(limits.get(consts.vgpu_number_resource(), 0) or 0)
+        if n < 1:
+            continue
+        params = {"partitionKey": c["name"]}
+        cores = int(limits.get(consts.vgpu_core_resource(), 0) or 0)
+        mem = int(limits.get(consts.vgpu_memory_resource(), 0) or 0)
+        if cores:
+            params["cores"] = cores // n
+        if mem:
+            params["memoryMiB"] = mem // n
+        spec = {"devices": {
+            "requests": [{"name": "gpu",
+                          "deviceClassName": "vgpu-manager",
+                          "count": n}],
+            "config": [{"requests": ["gpu"],
+                        "opaque": {"driver": "manager.amd.com",
+                                   "parameters": params}}]}}
+        out.append({"apiVersion": "resource.k8s.io/v1beta1",
+                    "kind": "ResourceClaimTemplate",
+                    "metadata": {"name":
+                                 f"vgpu-{pod_name}-{c['name']}"},
+                    "spec": {"spec": spec}})
+    return out
 
 
 def validate_pod(pod: dict) -> Tuple[bool, str]:
@@ -272,6 +321,7 @@ def validate_resource_claim(claim: dict) -> Tuple[bool, str]:
 
 def handle_admission_review(body: dict, *, mutating: bool,
                             dra_mode: bool = False,
+                            dra_per_container: bool = False,
                             kind: str = "Pod") -> dict:
     req = body.get("request", {}) or {}
     uid = req.get("uid", "")
@@ -281,7 +331,8 @@ def handle_admission_review(body: dict, *, mutating: bool,
         if kind == "VolcanoJob":
             patches = mutate_volcano_job(obj)
         else:
-            patches = mutate_pod(obj, dra_mode=dra_mode)
+            patches = mutate_pod(obj, dra_mode=dra_mode,
+                                 dra_per_container=dra_per_container)
         if patches:
             response["patchType"] = "JSONPatch"
             response["patch"] = base64.b64encode(
@@ -326,7 +377,8 @@ def apply_json_patch(obj: dict, patches: List[dict]) -> dict:
     return obj
 
 
-def create_app(dra_mode: bool = False, client=None):
+def create_app(dra_mode: bool = False, client=None,
+               dra_per_container: bool = False):
     # module-level import would make fastapi a hard dependency of every
     # admission-logic consumer; but the Request annotation must resolve
     # from module globals (PEP 563 strings) — so stash it there.
@@ -340,18 +392,23 @@ def create_app(dra_mode: bool = False, client=None):
         body = await request.json()
         if dra_mode and client is not None:
             obj = (body.get("request", {}) or {}).get("object", {}) or {}
-            tmpl = build_claim_template(obj)
-            if tmpl is not None:
-                ns = (obj.get("metadata", {}) or {}).get(
-                    "namespace", "default")
+            if dra_per_container:
+                tmpls = build_claim_templates_per_container(obj)
+            else:
+                t = build_claim_template(obj)
+                tmpls = [t] if t is not None else []
+            ns = (obj.get("metadata", {}) or {}).get(
+                "namespace", "default")
+            for tmpl in tmpls:
                 try:
                     client.create_resource_claim_template(ns, tmpl)
                 except Exception as e:  # conversion must not block
                     import logging
                     logging.getLogger("vgpu.webhook").warning(
                         "claim template create failed: %s", e)
-        return handle_admission_review(body, mutating=True,
-                                       dra_mode=dra_mode)
+        return handle_admission_review(
+            body, mutating=True, dra_mode=dra_mode,
+            dra_per_container=dra_per_container)
 
     @app.post("/webhook/validate-pod")
     async def validate(request: Request):
