@@ -233,3 +233,41 @@ def test_dra_per_container_conversion():
     pb = tb["spec"]["spec"]["devices"]["config"][0]["opaque"][
         "parameters"]
     assert pb["memoryMiB"] == 4096 and "cores" not in pb
+
+def test_per_container_template_flows_through_dra_resolve():
+    """Cross-component: the webhook's per-container template passes
+    claim validation and, once allocated, resolves to per-container
+    VgpuClaimParams with the container name as partitionKey."""
+    from vgpu_manager_amd.webhook.admission import (
+        build_claim_templates_per_container,
+        validate_resource_claim,
+    )
+    from vgpu_manager_amd.dra.resolve import resolve_claim
+
+    pod = {"metadata": {"name": "flow"},
+           "spec": {"containers": [
+               {"name": "train", "resources": {"limits": {
+                   consts.vgpu_number_resource(): "2",
+                   consts.vgpu_core_resource(): "60",
+                   consts.vgpu_memory_resource(): "16384"}}}]}}
+    (tmpl,) = build_claim_templates_per_container(pod)
+    spec = tmpl["spec"]["spec"]
+    ok, msg = validate_resource_claim({"spec": spec})
+    assert ok, msg
+
+    # simulate the scheduler allocating the template's request
+    claim = {"metadata": {"uid": "u-flow"},
+             "spec": spec,
+             "status": {"allocation": {"devices": {
+                 "results": [
+                     {"request": "gpu", "device": "GPU-aaaa"},
+                     {"request": "gpu", "device": "GPU-bbbb"},
+                 ],
+                 "config": spec["devices"]["config"],
+             }}}}
+    params, sharing = resolve_claim(claim)
+    assert sharing is None
+    assert [p.uuid for p in params] == ["GPU-aaaa", "GPU-bbbb"]
+    for p in params:
+        assert p.partition_key == "train"
+        assert p.cores == 30 and p.memory_mib == 8192
