@@ -1,0 +1,57 @@
+"""Deploy manifest lint: YAML parses, every container command maps to
+a real cmd module, and every flag the manifests pass actually exists
+in that binary's argparse surface (catches manifest/CLI drift)."""
+import glob
+import os
+import re
+import subprocess
+import sys
+
+import pytest
+import yaml
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+MANIFESTS = sorted(glob.glob(os.path.join(REPO, "deploy", "**", "*.yaml"),
+                             recursive=True))
+
+
+def iter_containers():
+    for path in MANIFESTS:
+        if "/chart/" in path:
+            continue  # helm templates are not plain yaml
+        for doc in yaml.safe_load_all(open(path)):
+            if not isinstance(doc, dict):
+                continue
+            spec = doc.get("spec", {}) or {}
+            pod = (spec.get("template", {}) or {}).get("spec") or \
+                (spec if doc.get("kind") == "Pod" else None)
+            if not pod:
+                continue
+            for c in pod.get("containers") or []:
+                yield path, c
+
+
+def test_all_manifests_parse():
+    assert MANIFESTS
+    for path in MANIFESTS:
+        if "/chart/" in path:
+            continue
+        list(yaml.safe_load_all(open(path)))
+
+
+@pytest.mark.parametrize("path,container", [
+    (p, c) for p, c in iter_containers()
+    if (c.get("command") or [""])[:2] == ["python", "-m"]],
+    ids=lambda v: v.split("/")[-1] if isinstance(v, str) else
+    v.get("name", "c"))
+def test_manifest_flags_exist(path, container):
+    mod = container["command"][2]
+    help_text = subprocess.run(
+        [sys.executable, "-m", mod, "--help"],
+        capture_output=True, text=True, timeout=60).stdout
+    assert help_text, f"{mod} --help produced nothing"
+    for arg in container.get("args") or []:
+        m = re.match(r"(--[a-z0-9-]+)", str(arg))
+        if m:
+            assert m.group(1) in help_text, \
+                f"{path}: {mod} does not accept {m.group(1)}"
