@@ -55,3 +55,23 @@ def test_manifest_flags_exist(path, container):
         if m:
             assert m.group(1) in help_text, \
                 f"{path}: {mod} does not accept {m.group(1)}"
+
+
+def test_chart_values_references_resolve():
+    """Every `.Values.x.y` referenced in chart templates exists in
+    values.yaml (helm is not installed here; this catches the common
+    rename drift statically)."""
+    chart = os.path.join(REPO, "deploy", "chart", "vgpu-manager")
+    values = yaml.safe_load(open(os.path.join(chart, "values.yaml")))
+    refs = set()
+    for tpl in glob.glob(os.path.join(chart, "templates", "*")):
+        for m in re.finditer(r"\.Values\.([A-Za-z0-9_.]+)",
+                             open(tpl).read()):
+            refs.add(m.group(1))
+    assert refs, "no .Values references found — chart gutted?"
+    for ref in sorted(refs):
+        node = values
+        for part in ref.split("."):
+            assert isinstance(node, dict) and part in node, \
+                f".Values.{ref} not present in values.yaml"
+            node = node[part]
