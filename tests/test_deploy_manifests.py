@@ -75,3 +75,38 @@ def test_chart_values_references_resolve():
             assert isinstance(node, dict) and part in node, \
                 f".Values.{ref} not present in values.yaml"
             node = node[part]
+
+
+def test_rbac_covers_every_rest_resource():
+    """Every API (group, resource) the REST client can touch is
+    granted by at least one ClusterRole in deploy/rbac.yaml — catches
+    'client grew a new resource, rbac never updated' drift."""
+    src = open(os.path.join(
+        REPO, "vgpu_manager_amd", "client", "kube.py")).read()
+    # join implicitly-concatenated string literals so split paths
+    # ("…{namespace}"\n  "/resourceclaimtemplates") regex as one
+    src = re.sub(r'"\s*\n\s*f?"', "", src)
+    used = set()
+    for m in re.finditer(
+            r'/api/v1/(?:namespaces/\{[^}]*\}/)?([a-z]+)', src):
+        used.add(("", m.group(1)))
+    for m in re.finditer(
+            r'/apis/([a-z0-9.]+)/v[0-9a-z]+/'
+            r'(?:namespaces/\{[^}]*\}/)?([a-z]+)', src):
+        used.add((m.group(1), m.group(2)))
+    for m in re.finditer(r'/pods/\{[^}]*\}/([a-z]+)', src):
+        used.add(("", "pods/" + m.group(1)))
+    used.discard(("", "namespaces"))  # path component, not a verb target
+    assert ("resource.k8s.io", "resourceclaimtemplates") in used
+
+    granted = set()
+    for doc in yaml.safe_load_all(
+            open(os.path.join(REPO, "deploy", "rbac.yaml"))):
+        if not doc or doc.get("kind") != "ClusterRole":
+            continue
+        for rule in doc.get("rules") or []:
+            for g in rule.get("apiGroups") or []:
+                for r in rule.get("resources") or []:
+                    granted.add((g, r))
+    missing = used - granted
+    assert not missing, f"REST resources with no RBAC grant: {missing}"
