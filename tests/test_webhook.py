@@ -271,3 +271,17 @@ def test_per_container_template_flows_through_dra_resolve():
     for p in params:
         assert p.partition_key == "train"
         assert p.cores == 30 and p.memory_mib == 8192
+
+def test_mutation_is_idempotent():
+    """Re-admitting an already-mutated pod (controller resync, retry)
+    must not stack claims or flip annotations."""
+    from vgpu_manager_amd.webhook.admission import mutate_pod
+
+    pod = make_pod(number=1, cores=50, memory=4096, name="idem")
+    once = apply_json_patch(pod, mutate_pod(pod, dra_mode=True))
+    again = mutate_pod(once, dra_mode=True)
+    assert again == []
+    # non-DRA path: defaults already set, second pass adds nothing new
+    pod2 = make_pod(number=1)
+    m1 = apply_json_patch(pod2, mutate_pod(pod2))
+    assert mutate_pod(m1) == []
