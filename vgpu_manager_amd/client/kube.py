@@ -17,7 +17,7 @@ import json
 import os
 import threading
 import time
-from typing import Callable, Dict, List, Optional
+from typing import Dict, List, Optional
 
 
 class KubeError(Exception):

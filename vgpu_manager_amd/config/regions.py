@@ -17,7 +17,7 @@ import ctypes
 import mmap
 import os
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Iterable, Optional
 
 from .abi import (

@@ -11,7 +11,7 @@ import logging
 import os
 import threading
 import time
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from ..client.kube import KubeClient, KubeError
 from ..util import consts

@@ -9,12 +9,11 @@ pattern).
 """
 from __future__ import annotations
 
-import json
 import logging
 import os
 import threading
 import time
-from typing import Callable, Dict, List, Optional
+from typing import Callable, List, Optional
 
 from ..client.kube import KubeClient
 from ..util import consts

@@ -17,7 +17,6 @@ import logging
 import os
 import threading
 from concurrent import futures
-from typing import Optional
 
 import grpc
 

@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import logging
 import os
-from typing import List, Optional
+from typing import Optional
 
 log = logging.getLogger("vgpu.dra.vfio")
 

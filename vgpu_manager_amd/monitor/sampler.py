@@ -13,7 +13,6 @@ from __future__ import annotations
 import logging
 import threading
 import time
-from typing import List, Optional
 
 from ..config.regions import UtilRegionWriter
 

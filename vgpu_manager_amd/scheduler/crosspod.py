@@ -8,7 +8,7 @@ as the siblings' devices (on a fully-connected xGMI node the NUMA
 domain is the remaining locality axis — SURVEY §5.8)."""
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, Optional, Tuple
 
 from ..client.kube import KubeClient, KubeError
 from ..device.types import decode_node_devices, unmarshal_pod_claim

@@ -19,7 +19,6 @@ from ..device.allocator import (
     Allocator,
     build_allocation_request,
 )
-from ..util import consts
 from . import metrics
 from .snapshot import build_node_info, pod_claim_annotation
 

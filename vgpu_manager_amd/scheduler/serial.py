@@ -4,7 +4,6 @@ from __future__ import annotations
 
 import threading
 import time
-from typing import Dict
 
 
 class KeyedLocker:

@@ -9,7 +9,7 @@ uses.  Field numbers mirror the upstream api.proto exactly.
 """
 from __future__ import annotations
 
-from typing import Any, Dict, List, Tuple
+from typing import Any, Dict, Tuple
 
 WIRE_VARINT = 0
 WIRE_I64 = 1
