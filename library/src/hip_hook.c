@@ -1630,6 +1630,8 @@ static const hook_entry_t g_hooks[] = {
     {"hipExtLaunchKernel", (void *)hipExtLaunchKernel},
     {"hipModuleLaunchKernel", (void *)hipModuleLaunchKernel},
     {"hipExtModuleLaunchKernel", (void *)hipExtModuleLaunchKernel},
+    {"hipLaunchKernelExC", (void *)hipLaunchKernelExC},
+    {"hipDrvLaunchKernelEx", (void *)hipDrvLaunchKernelEx},
     {"hipLaunchCooperativeKernel", (void *)hipLaunchCooperativeKernel},
     {"hipModuleLaunchCooperativeKernel",
      (void *)hipModuleLaunchCooperativeKernel},

@@ -94,3 +94,48 @@ def test_shim_exports_exactly_match_exports_map():
     got = {line.split()[-1] for line in out.splitlines() if line.strip()}
     assert got == want, (f"exported-symbol drift: extra={got - want} "
                          f"missing={want - got}")
+
+
+def test_hook_table_consistency():
+    """hack/-style CI guard (reference hack/check_cuda_hook_consistency):
+    every EXPORTed hip/smi hook must appear in exports.map AND in the
+    dlsym/GetProcAddress routing table, and every routed hip hook must
+    have a real-table LOAD entry — a symbol missing from any of the
+    three silently bypasses the gate for one resolution path (this
+    caught hipLaunchKernelExC/hipDrvLaunchKernelEx missing from the
+    routing table)."""
+    import re
+
+    from tests.conftest import LIB_DIR
+    src_dir = os.path.join(LIB_DIR, "src")
+    hook_src = open(os.path.join(src_dir, "hip_hook.c")).read()
+    smi_src = open(os.path.join(src_dir, "smi_hook.c")).read()
+    loader_src = open(os.path.join(src_dir, "loader.c")).read()
+
+    exported_hip = set(re.findall(
+        r"^EXPORT\s+\w+\s+(hip\w+)\s*\(", hook_src, re.M))
+    exported_smi = set(re.findall(
+        r"^EXPORT\s+\w+\s+((?:amdsmi|rsmi)_\w+)\s*\(", smi_src, re.M))
+    assert exported_hip and exported_smi
+
+    routed_hip = set(re.findall(r'\{"(hip\w+)"', hook_src))
+    routed_smi = set(re.findall(r'\{"((?:amdsmi|rsmi)_\w+)"', smi_src))
+    loaded = set(re.findall(r"LOAD\((hip\w+)\)", loader_src))
+
+    exports_map = set(re.findall(
+        r"^\s+([A-Za-z_][A-Za-z0-9_]*);", open(
+            os.path.join(LIB_DIR, "exports.map")).read(), re.M))
+
+    missing_route = exported_hip - routed_hip
+    assert not missing_route, \
+        f"exported hip hooks absent from routing table: {missing_route}"
+    assert not (exported_smi - routed_smi), \
+        f"smi hooks absent from routing: {exported_smi - routed_smi}"
+    assert not (exported_hip - exports_map), \
+        f"hip hooks absent from exports.map: {exported_hip - exports_map}"
+    assert not (exported_smi - exports_map), \
+        f"smi hooks absent from exports.map: {exported_smi - exports_map}"
+    # alias: plain hipGetDeviceProperties routes to the R0600 hook
+    assert not (routed_hip - loaded - {"hipGetDeviceProperties"}), \
+        f"routed hooks with no real-table LOAD: " \
+        f"{routed_hip - loaded - {'hipGetDeviceProperties'}}"
