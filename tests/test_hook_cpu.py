@@ -228,8 +228,17 @@ assert rc == 2, f"expected hipErrorOutOfMemory via dlsym route, got {rc}"
 free = ctypes.c_size_t(); total = ctypes.c_size_t()
 assert lib.hipMemGetInfo(ctypes.byref(free), ctypes.byref(total)) == 0
 assert total.value == 1024 * 1024, total.value   # spoofed view
+# pointer identity: dlsym on the REAL lib handle must hand back the
+# SHIM's launch hooks (the gap this guards: a launch entry missing
+# from the routing table resolves to the real fn and skips the gate)
+shim = ctypes.CDLL(%r)
+for sym in ("hipLaunchKernel", "hipLaunchKernelExC",
+            "hipDrvLaunchKernelEx", "hipGraphLaunch"):
+    got = ctypes.cast(getattr(lib, sym), ctypes.c_void_p).value
+    want = ctypes.cast(getattr(shim, sym), ctypes.c_void_p).value
+    assert got == want, f"{sym}: dlsym bypassed the shim"
 print("DLSYM-OK")
-"""
+"""  % (os.path.join(build, "libvgpu-control.so"),)
     env = dict(os.environ)
     env.update({
         "VGPU_MEM_LIMIT_0": "1m", "VGPU_MEM_ACCOUNT_MODE": "ledger",
