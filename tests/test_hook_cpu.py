@@ -331,3 +331,21 @@ def test_vmem_region_cross_language_read(built_library, tmp_path):
     # the PROCESS exited: exit cleanup retired its hooked bytes too
     assert usage[0]["dev_hooked_used"] == 0
     r.close()
+
+
+def test_metrics_counters_emitted(built_library):
+    """metrics.c power-of-two counter lines reach stderr (reference
+    metrics.c observability): the quota scenario's rejected alloc
+    must log `metric oom=1` at info level."""
+    r = run_scenario("quota", {
+        "VGPU_MEM_LIMIT_0": "1m",
+        "VGPU_MEM_ACCOUNT_MODE": "ledger",
+        "VGPU_LOGGER_LEVEL": "3",        # LOG_INFO
+    })
+    assert "metric oom=1" in r.stderr, r.stderr[-2000:]
+    # throttle scenario exercises the rate limiter counter
+    r2 = run_scenario("throttle", {
+        "VGPU_CORE_LIMIT_0": "10",
+        "VGPU_LOGGER_LEVEL": "3",
+    })
+    assert "metric rate_limit_sleep=1" in r2.stderr, r2.stderr[-2000:]
