@@ -1,0 +1,50 @@
+"""bench.py driver-contract tests: exactly one JSON line on rank 0
+with the BASELINE.json metric fields, single-process and under
+torchrun world-size 2 (the way the driver launches N>1)."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+REQUIRED = {"metric", "value", "unit", "n_gpus", "steps", "warmup",
+            "ms_per_step", "higher_is_better", "scaling",
+            "vs_baseline", "dtype", "data", "config"}
+
+
+def last_json_line(text):
+    lines = [ln for ln in text.splitlines()
+             if ln.startswith("{") and ln.rstrip().endswith("}")]
+    assert len(lines) == 1, f"want exactly one JSON line, got:\n{text}"
+    return json.loads(lines[0])
+
+
+def test_bench_single_process():
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"),
+         "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=280, cwd=REPO)
+    assert r.returncode == 0, r.stdout + r.stderr
+    out = last_json_line(r.stdout)
+    assert REQUIRED <= set(out)
+    base = json.load(open(os.path.join(REPO, "BASELINE.json")))
+    assert "hook overhead" in base["metric"]
+    assert out["metric"] == "hook_overhead_pct_vs_bare_hip"
+    assert out["n_gpus"] == 1 and out["steps"] == 2
+    assert out["higher_is_better"] is False
+
+
+@pytest.mark.timeout(280)
+def test_bench_torchrun_world2():
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29713", "bench.py",
+         "--gpus", "2", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=260, cwd=REPO)
+    assert r.returncode == 0, r.stdout + r.stderr
+    out = last_json_line(r.stdout)   # rank 0 only prints
+    assert out["n_gpus"] == 2
+    assert REQUIRED <= set(out)
