@@ -40,8 +40,8 @@ def test_bench_single_process():
 def test_bench_torchrun_world2():
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29713", "bench.py",
+         "--nproc-per-node", "2", "--standalone",
+         "--local-addr", "127.0.0.1", "bench.py",
          "--gpus", "2", "--steps", "2", "--warmup", "1"],
         capture_output=True, text=True, timeout=260, cwd=REPO)
     assert r.returncode == 0, r.stdout + r.stderr
