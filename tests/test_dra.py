@@ -620,3 +620,46 @@ def test_dra_grpc_uid_mismatch_rejected(state, tmp_path):
         ch.close()
     finally:
         server.stop()
+
+
+def test_dra_grpc_garbage_payload_does_not_kill_server(state, tmp_path):
+    """A kubelet bug / version skew can deliver an undecodable
+    payload: the handler must surface a gRPC error for THAT call and
+    keep serving subsequent valid ones."""
+    import grpc as _grpc
+    from vgpu_manager_amd.client.kube import FakeKubeClient
+    from vgpu_manager_amd.dra import api as dapi
+    from vgpu_manager_amd.dra.driver import DraDriver, DraDriverServer
+
+    client = FakeKubeClient()
+    client.add_resource_claim(_alloc_claim(uid="uid-1", configs=[]))
+    endpoint = str(tmp_path / "plugins" / "drv" / "dra.sock")
+    driver = DraDriver(state, client, endpoint=endpoint)
+    server = DraDriverServer(
+        driver, plugins_dir=str(tmp_path / "plugins"),
+        plugins_registry=str(tmp_path / "registry"))
+    server.start()
+    try:
+        ch = _grpc.insecure_channel(f"unix://{endpoint}")
+        raw = ch.unary_unary(
+            f"/{dapi.DRA_SERVICE}/NodePrepareResources",
+            request_serializer=lambda b: b,
+            response_deserializer=lambda b: b)
+        for payload in (b"\xff" * 64, b"\x0a", os.urandom(256)):
+            try:
+                raw(payload, timeout=10)
+            except _grpc.RpcError:
+                pass  # an error status is the acceptable outcome
+        # server must still answer a well-formed request
+        prep = ch.unary_unary(
+            f"/{dapi.DRA_SERVICE}/NodePrepareResources",
+            request_serializer=lambda m: m.encode(),
+            response_deserializer=dapi.NodePrepareResourcesResponse
+            .decode)
+        resp = prep(dapi.NodePrepareResourcesRequest(claims=[
+            dapi.Claim(namespace="default", uid="uid-1", name="c")]),
+            timeout=10)
+        assert resp.claims[0].key == "uid-1"
+        ch.close()
+    finally:
+        server.stop()
