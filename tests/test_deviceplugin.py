@@ -337,3 +337,32 @@ def test_pbwire_property_roundtrip():
                     [n.ID for n in b.topology.nodes]
 
     roundtrip()
+
+
+def test_grpc_garbage_payload_keeps_serving(plugin_env, tmp_path):
+    """Undecodable kubelet payloads must error per-call, not kill the
+    plugin server (symmetric with the DRA-side robustness test)."""
+    client, mgr, plugin = plugin_env
+    server = PluginServer(plugin, "test-vgpu.sock",
+                          plugins_dir=str(tmp_path / "plugins"))
+    server.start()
+    try:
+        ch = grpc.insecure_channel(f"unix://{server.socket_path}")
+        raw = ch.unary_unary(
+            "/v1beta1.DevicePlugin/GetDevicePluginOptions",
+            request_serializer=lambda b: b,
+            response_deserializer=lambda b: b)
+        for payload in (b"\xff" * 32, os.urandom(128)):
+            try:
+                raw(payload, timeout=5)
+            except grpc.RpcError:
+                pass
+        opts = ch.unary_unary(
+            "/v1beta1.DevicePlugin/GetDevicePluginOptions",
+            request_serializer=lambda m: m.encode(),
+            response_deserializer=api.DevicePluginOptions.decode)(
+                api.Empty(), timeout=5)
+        assert opts.pre_start_required is True
+        ch.close()
+    finally:
+        server.stop()
