@@ -140,3 +140,27 @@ def test_client_mode_end_to_end_with_shim(registry, built_library,
     # here host ns == test ns so it is the scenario binary's pid —
     # at minimum the transient client and its parent were captured
     assert len(pids) >= 1
+
+
+def test_garbage_bytes_keep_server_alive(registry):
+    """A confused client (or port scan) sending non-JSON must get an
+    error/close for that connection only; the next legitimate
+    registration succeeds."""
+    import socket
+
+    sock, base, _ = registry
+    for payload in (b"\xff" * 64, b"not json\n", b"{truncated",
+                    b"\x00" * 1024):
+        s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        s.settimeout(5)
+        s.connect(sock)
+        s.sendall(payload)
+        s.shutdown(socket.SHUT_WR)
+        try:
+            s.recv(4096)  # whatever it answers (or close) is fine
+        except OSError:
+            pass
+        s.close()
+    out = register_via_socket(sock, "uid-1", "main")
+    assert out["ok"], out
+    assert os.getpid() in read_pids(base)
