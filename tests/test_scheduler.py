@@ -308,3 +308,26 @@ def test_filter_scale_correctness():
     assert all(v for v in failed.values())
     # a 300-node filter pass must fit inside an extender HTTP timeout
     assert elapsed < 10.0, f"filter took {elapsed:.1f}s for 300 nodes"
+
+
+def test_extender_undecodable_body_returns_structured_error():
+    """Non-JSON bytes on any verb answer with the verb's Error field
+    (kube-scheduler logs it) rather than a bare 500."""
+    import warnings
+    warnings.filterwarnings("ignore")
+    from starlette.testclient import TestClient
+
+    from vgpu_manager_amd.client.kube import FakeKubeClient
+    from vgpu_manager_amd.scheduler.http import create_app
+
+    tc = TestClient(create_app(FakeKubeClient()),
+                    raise_server_exceptions=False)
+    for path in ("/scheduler/filter", "/scheduler/bind",
+                 "/scheduler/preempt"):
+        r = tc.post(path, content=b"\xff\xfe not json",
+                    headers={"content-type": "application/json"})
+        assert r.status_code == 200, (path, r.status_code)
+        assert "undecodable" in r.json()["Error"]
+    # degenerate-but-valid JSON keeps working
+    r = tc.post("/scheduler/filter", json={})
+    assert r.status_code == 200 and r.json()["Error"] == ""

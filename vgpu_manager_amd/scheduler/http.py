@@ -21,24 +21,42 @@ def create_app(client: KubeClient) -> FastAPI:
     binder = NodeBinder(client)
     preempter = VgpuPreempter(client)
 
+    async def _args(request: Request):
+        # an unparseable body answers with the verb's structured Error
+        # (kube-scheduler logs it) instead of a bare 500
+        try:
+            return await request.json(), None
+        except (json.JSONDecodeError, UnicodeDecodeError) as e:
+            return None, f"extender: undecodable request body: {e}"
+
     @app.post("/scheduler/filter")
     async def filter_(request: Request):
-        args = await request.json()
+        args, err = await _args(request)
+        if err:
+            return {"Nodes": None, "NodeNames": [], "FailedNodes": {},
+                    "Error": err}
         return gpu_filter.filter(args, dry_run=False)
 
     @app.post("/scheduler/filter-dryrun")
     async def filter_dryrun(request: Request):
-        args = await request.json()
+        args, err = await _args(request)
+        if err:
+            return {"Nodes": None, "NodeNames": [], "FailedNodes": {},
+                    "Error": err}
         return gpu_filter.filter(args, dry_run=True)
 
     @app.post("/scheduler/bind")
     async def bind(request: Request):
-        args = await request.json()
+        args, err = await _args(request)
+        if err:
+            return {"Error": err}
         return binder.bind(args)
 
     @app.post("/scheduler/preempt")
     async def preempt(request: Request):
-        args = await request.json()
+        args, err = await _args(request)
+        if err:
+            return {"NodeNameToMetaVictims": {}, "Error": err}
         return preempter.preempt(args)
 
     @app.get("/healthz")
