@@ -110,3 +110,38 @@ def test_rbac_covers_every_rest_resource():
                     granted.add((g, r))
     missing = used - granted
     assert not missing, f"REST resources with no RBAC grant: {missing}"
+
+
+def test_every_shim_env_is_documented():
+    """Every env var the C shim reads (getenv/vgpu_getenv/env_* call
+    sites) appears in docs/api_reference.md — the doc uses
+    `VGPU_X_{A,B}` brace shorthand, so expand it before comparing."""
+    src_dir = os.path.join(REPO, "library", "src")
+    used = set()
+    for f in glob.glob(os.path.join(src_dir, "*.c")):
+        s = open(f).read()
+        for m in re.finditer(
+                r'(?:vgpu_getenv|getenv|env_(?:int|bool|str|u64|size))'
+                r'\(\s*"([A-Z][A-Z0-9_]+)"', s):
+            used.add(m.group(1))
+        for m in re.finditer(r'"((?:VGPU|CUDA)_[A-Z_]+)_%d"', s):
+            used.add(m.group(1) + "_<i>")
+    assert len(used) > 20, used  # sanity: extraction still works
+
+    doc = re.sub(r"\s+", "", open(os.path.join(
+        REPO, "docs", "api_reference.md")).read())
+    # expand {A,B,C} groups (possibly several per token) into full names
+    names = set()
+    for tok in re.findall(r"[A-Z][A-Z0-9_{},<>i]*", doc):
+        variants = [""]
+        for part in re.split(r"(\{[A-Z0-9_,<>i]+\})", tok):
+            if part.startswith("{"):
+                opts = part[1:-1].split(",")
+                variants = [v + o for v in variants for o in opts]
+            else:
+                variants = [v + part for v in variants]
+        names.update(variants)
+    blob = doc  # plain occurrences (incl. inside longer words)
+    missing = {e for e in used
+               if e not in names and e not in blob}
+    assert not missing, f"shim envs missing from api_reference: {missing}"
