@@ -145,3 +145,23 @@ def test_every_shim_env_is_documented():
     missing = {e for e in used
                if e not in names and e not in blob}
     assert not missing, f"shim envs missing from api_reference: {missing}"
+
+
+def test_webhook_config_paths_are_served():
+    """Every webhook path registered in deploy/webhook.yaml must be an
+    actual route of the admission app (and the reverse: every POST
+    route should be reachable from some webhook configuration)."""
+    from vgpu_manager_amd.webhook.admission import create_app
+
+    text = open(os.path.join(REPO, "deploy", "webhook.yaml")).read()
+    cfg_paths = set(re.findall(r"path:\s*(/webhook/[a-z-]+)", text))
+    assert cfg_paths, "no webhook paths found in deploy/webhook.yaml"
+    app = create_app()
+    served = {r.path for r in app.routes if r.path.startswith("/webhook")}
+    assert cfg_paths <= served, \
+        f"configured but unserved: {cfg_paths - served}"
+    # unreferenced POST routes are fine only if deliberate; today the
+    # one exception is validate-volcanojob (volcano validates via its
+    # own admission by default)
+    unref = served - cfg_paths
+    assert unref <= {"/webhook/validate-volcanojob"}, unref
